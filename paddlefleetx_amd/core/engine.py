@@ -1,0 +1,387 @@
+"""EagerEngine: the trainer.
+
+Reference: ppfleetx/core/engine/eager_engine.py (fit :422, _fit_impl :479,
+micro-batch accumulation :522-561, optim update :563-578, save/load
+:717-825 with the epoch_X_step_Y/mp_XX_sharding_XX_pp_XX layout).
+
+MI355X-native differences: AMP-O2 = bf16 weights + fp32 flat master/grad
+buffers inside FusedAdamW (one HIP kernel per bucket); DP gradient
+allreduce runs directly on the fused fp32 buffers over RCCL; fp16 path
+keeps a dynamic loss scale with a cross-rank found_inf allreduce
+(reference amp.py:193-234); pipeline parallel delegates to the native
+1F1B scheduler.
+"""
+
+from __future__ import annotations
+
+import os
+import time
+from typing import Any, Dict, Optional
+
+import torch
+import torch.distributed as dist
+
+from paddlefleetx_amd.optims import build_lr_scheduler, build_optimizer
+from paddlefleetx_amd.optims.optimizer import FusedAdamW
+from paddlefleetx_amd.parallel import sp as sp_ops
+from paddlefleetx_amd.parallel.env import get_hcg
+from paddlefleetx_amd.utils.log import logger
+
+
+class BasicEngine:
+    def fit(self, *a, **k):
+        raise NotImplementedError
+
+    def evaluate(self, *a, **k):
+        raise NotImplementedError
+
+
+def _split_micro(batch, n: int):
+    """Split each tensor of the batch along dim0 into n micro-batches."""
+    if n == 1:
+        return [batch]
+    parts = [torch.chunk(t, n, dim=0) if torch.is_tensor(t) else [t] * n
+             for t in batch]
+    return [tuple(p[i] for p in parts) for i in range(n)]
+
+
+class EagerEngine(BasicEngine):
+    def __init__(self, configs, module, mode: str = "train"):
+        self.configs = configs
+        self.module = module
+        self.mode = mode
+        self.hcg = get_hcg()
+
+        e = configs["Engine"]
+        self.accumulate_steps = int(e.get("accumulate_steps", 1))
+        self.max_steps = e.get("max_steps")
+        self.logging_freq = int(e.get("logging_freq", 10))
+        self.eval_freq = e.get("eval_freq")
+        self.eval_iters = int(e.get("eval_iters", 10))
+        self.num_train_epochs = int(e.get("num_train_epochs", 1))
+        sl = e.get("save_load", {})
+        self.save_steps = sl.get("save_steps")
+        self.output_dir = sl.get("output_dir", "./output")
+
+        mpc = e.get("mix_precision", {})
+        self.amp_enable = bool(mpc.get("enable", True))
+        self.amp_dtype = {"bfloat16": torch.bfloat16, "float16": torch.float16,
+                          "float32": torch.float32}[mpc.get("dtype", "bfloat16")]
+        self.loss_scale = float(mpc.get("scale_loss", 32768.0)) \
+            if self.amp_dtype == torch.float16 else 1.0
+        self._scale_growth_interval = 1000
+        self._good_steps = 0
+
+        self.device = torch.device("cuda") if torch.cuda.is_available() \
+            else torch.device("cpu")
+        self.module.to(self.device)
+
+        self.is_pipeline = self.hcg.get_pipe_parallel_world_size() > 1
+        if self.is_pipeline:
+            assert hasattr(self.module.model, "forward_backward_pipeline"), \
+                "pp_degree>1 requires a pipeline model"
+
+        # ZeRO sharding
+        sh = configs.get("Distributed", {}).get("sharding", {})
+        self.sharding_stage = int(sh.get("sharding_stage", 1))
+        self.sharding_degree = self.hcg.get_sharding_parallel_world_size()
+
+        opt_cfg = configs.get("Optimizer", {})
+        self.lr_scheduler = build_lr_scheduler(opt_cfg.get("lr", {}))
+        self.grad_clip_norm = None
+        gc = opt_cfg.get("grad_clip", None)
+        if isinstance(gc, dict):
+            self.grad_clip_norm = gc.get("clip_norm", 1.0)
+        elif isinstance(gc, (int, float)):
+            self.grad_clip_norm = float(gc)
+
+        if mode == "train":
+            if self.sharding_degree > 1 and self.sharding_stage >= 2:
+                from paddlefleetx_amd.parallel.zero import ShardedOptimizer
+                self.optimizer = ShardedOptimizer(
+                    self.module.model, opt_cfg, self.hcg,
+                    stage=self.sharding_stage,
+                    lr_value=self.lr_scheduler.get_lr())
+            else:
+                self.optimizer = build_optimizer(
+                    opt_cfg, self.module.model,
+                    lr_value=self.lr_scheduler.get_lr())
+            # broadcast initial params across dp so replicas agree
+            self._sync_params()
+        else:
+            self.optimizer = None
+
+        self._load_recovery = {"step": 0, "epoch": 0}
+        ckpt_dir = sl.get("ckpt_dir") if isinstance(sl, dict) else None
+        if ckpt_dir:
+            self.load(ckpt_dir)
+
+    # ------------------------------------------------------------------
+    def _sync_params(self):
+        """Broadcast params from dp-rank-0 (reference strategy.py:43 sync_params_buffers)."""
+        dp = self.hcg.get_data_parallel_group()
+        groups = [dp]
+        sd = self.hcg.get_sharding_parallel_group()
+        if self.sharding_stage == 1:
+            groups.append(sd)
+        for g in groups:
+            if g.world_size <= 1 or not dist.is_initialized():
+                continue
+            if isinstance(self.optimizer, FusedAdamW):
+                for b in self.optimizer.buckets:
+                    dist.broadcast(b.model_flat, src=g.ranks[0], group=g.group)
+                    b.master.copy_(b.model_flat.float())
+            else:
+                for p in self.module.model.parameters():
+                    dist.broadcast(p.data, src=g.ranks[0], group=g.group)
+
+    # ------------------------------------------------------------------
+    def fit(self, train_data_loader=None, valid_data_loader=None, epoch=None):
+        epochs = epoch if epoch is not None else self.num_train_epochs
+        start_epoch = self._load_recovery["epoch"]
+        for ep in range(start_epoch, epochs):
+            done = self._train_one_epoch(ep, train_data_loader, valid_data_loader)
+            if done:
+                break
+
+    def _train_one_epoch(self, epoch: int, train_loader, valid_loader) -> bool:
+        self.module.model.train()
+        skip_until = self._load_recovery["step"] if \
+            epoch == self._load_recovery["epoch"] else 0
+        t_start = time.time()
+        interval_cost = 0.0
+        for step, batch in enumerate(train_loader):
+            if step < skip_until:
+                continue
+            loss = self._fit_impl(batch)
+            self.module.global_step += 1
+            gstep = self.module.global_step
+            interval_cost = time.time() - t_start
+            if gstep % self.logging_freq == 0:
+                self.module.training_step_end({
+                    "epoch": epoch, "batch": step,
+                    "loss": float(loss),
+                    "train_cost": interval_cost / self.logging_freq,
+                    "lr": self.lr_scheduler.get_lr(),
+                    "found_inf": float(getattr(self, "_found_inf", 0.0)),
+                })
+                t_start = time.time()
+            if self.eval_freq and gstep % self.eval_freq == 0 and valid_loader:
+                self._evaluate_impl(epoch, valid_loader)
+                self.module.model.train()
+            if self.save_steps and gstep % self.save_steps == 0:
+                if self.device.type == "cuda":
+                    torch.cuda.synchronize()
+                self.save(epoch, gstep)
+            if self.max_steps and gstep >= self.max_steps:
+                return True
+        return False
+
+    # ------------------------------------------------------------------
+    def _fit_impl(self, batch) -> torch.Tensor:
+        batch = self.module.pretreating_batch(batch)
+        batch = tuple(t.to(self.device, non_blocking=True) if torch.is_tensor(t)
+                      else t for t in batch)
+        if self.is_pipeline:
+            loss = self.module.model.forward_backward_pipeline(
+                batch, self.module.loss_fn, self.accumulate_steps,
+                scale=self.loss_scale)
+        else:
+            loss = self._model_forward_backward(batch)
+        self._optim_update_params()
+        return loss
+
+    def _model_forward_backward(self, batch) -> torch.Tensor:
+        micros = _split_micro(batch, self.accumulate_steps)
+        total = 0.0
+        for mb in micros:
+            loss = self.module.training_step(mb)
+            scaled = loss * (self.loss_scale / self.accumulate_steps)
+            self.module.backward(scaled)
+            total += float(loss.detach())
+        return torch.tensor(total / self.accumulate_steps)
+
+    def _optim_update_params(self):
+        model = self.module.model
+        # SP: LayerNorm/bias grads over mp group (sequence_parallel_utils.py:166-185)
+        if getattr(model, "sequence_parallel", False) or any(
+                getattr(p, "sequence_parallel", False)
+                for p in model.parameters()):
+            self._allreduce_sp_main_grads()
+        self._found_inf = 0.0
+        if isinstance(self.optimizer, FusedAdamW):
+            # DP (+ZeRO-1 sharding) reduce on fused buffers
+            dp = self.hcg.get_data_parallel_group()
+            n_replicas = dp.world_size
+            sd = self.hcg.get_sharding_parallel_group()
+            if self.sharding_degree > 1 and self.sharding_stage == 1:
+                n_replicas *= sd.world_size
+                self.optimizer.reduce_gradients(
+                    self.hcg.get_data_world_group(), avg_factor=1.0)
+            elif dp.world_size > 1:
+                self.optimizer.reduce_gradients(dp, avg_factor=1.0)
+            inv = 1.0 / (n_replicas * self.loss_scale)
+            if inv != 1.0:
+                for b in self.optimizer.buckets:
+                    b.main_grad.mul_(inv)
+            if self.loss_scale != 1.0:
+                self._found_inf = self._check_found_inf()
+                if self._found_inf:
+                    self.loss_scale = max(1.0, self.loss_scale / 2)
+                    self._good_steps = 0
+                    self.optimizer.zero_grad()
+                    self.lr_scheduler.step()
+                    return
+                self._good_steps += 1
+                if self._good_steps >= self._scale_growth_interval:
+                    self.loss_scale *= 2
+                    self._good_steps = 0
+            if self.grad_clip_norm:
+                self.optimizer.clip_grads(
+                    self.grad_clip_norm,
+                    mp_group=self.hcg.get_model_parallel_group(),
+                    pp_group=self.hcg.get_pipe_parallel_group())
+            self.lr_scheduler.step()
+            self.optimizer.step(lr=self.lr_scheduler.get_lr())
+            self.optimizer.zero_grad()
+        else:
+            # plain torch optimizer path (incl. ShardedOptimizer which mimics it)
+            if hasattr(self.optimizer, "reduce_and_step"):
+                self.lr_scheduler.step()
+                self.optimizer.reduce_and_step(
+                    lr=self.lr_scheduler.get_lr(),
+                    grad_clip=self.grad_clip_norm,
+                    loss_scale=self.loss_scale)
+            else:
+                self._allreduce_plain_grads()
+                if self.grad_clip_norm:
+                    torch.nn.utils.clip_grad_norm_(
+                        self.module.model.parameters(), self.grad_clip_norm)
+                self.lr_scheduler.step()
+                for g in self.optimizer.param_groups:
+                    g["lr"] = self.lr_scheduler.get_lr()
+                self.optimizer.step()
+                self.optimizer.zero_grad(set_to_none=True)
+
+    def _check_found_inf(self) -> float:
+        found = 0.0
+        for b in self.optimizer.buckets:
+            if not torch.isfinite(b.main_grad).all():
+                found = 1.0
+                break
+        if dist.is_initialized():
+            t = torch.tensor(found, device=self.device)
+            dist.all_reduce(t, op=dist.ReduceOp.MAX)
+            found = float(t)
+        return found
+
+    def _allreduce_sp_main_grads(self):
+        g = self.hcg.get_model_parallel_group()
+        if g.world_size == 1:
+            return
+        for p in self.module.model.parameters():
+            if getattr(p, "sequence_parallel", False):
+                buf = p.main_grad if hasattr(p, "main_grad") else p.grad
+                if buf is not None:
+                    dist.all_reduce(buf, group=g.group)
+
+    def _allreduce_plain_grads(self):
+        dp = self.hcg.get_data_parallel_group()
+        if dp.world_size == 1:
+            return
+        for p in self.module.model.parameters():
+            if p.grad is not None:
+                dist.all_reduce(p.grad, group=dp.group)
+                p.grad.div_(dp.world_size)
+
+    # ------------------------------------------------------------------
+    @torch.no_grad()
+    def _evaluate_impl(self, epoch: int, loader):
+        self.module.model.eval()
+        t0 = time.time()
+        for i, batch in enumerate(loader):
+            if i >= self.eval_iters:
+                break
+            batch = tuple(t.to(self.device) if torch.is_tensor(t) else t
+                          for t in batch)
+            loss = self.module.validation_step(batch)
+            self.module.validation_step_end({
+                "epoch": epoch, "batch": i, "loss": float(loss),
+                "eval_cost": (time.time() - t0) / (i + 1)})
+
+    def evaluate(self, valid_data_loader=None, epoch: int = 0):
+        self._evaluate_impl(epoch, valid_data_loader)
+
+    @torch.no_grad()
+    def predict(self, data_loader):
+        self.module.model.eval()
+        outs = []
+        for batch in data_loader:
+            batch = tuple(t.to(self.device) if torch.is_tensor(t) else t
+                          for t in batch)
+            outs.append(self.module(*batch))
+        return outs
+
+    # ------------------------------------------------------------------
+    # checkpoint: epoch_X_step_Y/mp_XX_sharding_XX_pp_XX/{model.pdparams,
+    #             model_state.pdopt, meta_state.pdopt}  (eager_engine.py:717-755)
+    def _ckpt_subdir(self) -> str:
+        return "mp_{:02d}_sharding_{:02d}_pp_{:02d}".format(
+            self.hcg.get_model_parallel_rank(),
+            self.hcg.get_sharding_parallel_rank(),
+            self.hcg.get_pipe_parallel_rank())
+
+    def save(self, epoch: int = 0, step: int = 0):
+        if self.hcg.get_data_parallel_rank() != 0:
+            return  # only dp rank 0 saves (io.py:44-46)
+        out = os.path.join(self.output_dir, f"epoch_{epoch}_step_{step}",
+                           self._ckpt_subdir())
+        os.makedirs(out, exist_ok=True)
+        torch.save(self.module.model.state_dict(),
+                   os.path.join(out, "model.pdparams"))
+        if self.optimizer is not None:
+            torch.save(self.optimizer.state_dict(),
+                       os.path.join(out, "model_state.pdopt"))
+        meta = {"epoch": epoch, "step": step,
+                "global_step": self.module.global_step,
+                "lr_scheduler": self.lr_scheduler.state_dict(),
+                "cpu_rng_state": torch.get_rng_state()}
+        if torch.cuda.is_available():
+            meta["cuda_rng_state"] = torch.cuda.get_rng_state()
+        torch.save(meta, os.path.join(out, "meta_state.pdopt"))
+        logger.info(f"saved checkpoint to {out}")
+
+    def load(self, ckpt_dir: str):
+        path = os.path.join(ckpt_dir, self._ckpt_subdir())
+        if not os.path.isdir(path):
+            path = ckpt_dir
+        model_path = os.path.join(path, "model.pdparams")
+        sd = torch.load(model_path, map_location=self.device,
+                        weights_only=False)
+        missing, unexpected = self.module.model.load_state_dict(sd, strict=False)
+        if missing or unexpected:
+            logger.warning(f"ckpt load: missing={missing} unexpected={unexpected}")
+        if isinstance(self.optimizer, FusedAdamW):
+            # re-pointing: state_dict load above replaced p.data tensors; refuse
+            for b in self.optimizer.buckets:
+                for p, off in zip(b.params, b.offsets):
+                    b.model_flat[off:off + p.numel()].copy_(p.data.reshape(-1))
+                    p.data = b.model_flat[off:off + p.numel()].view(p.shape)
+                b.master.copy_(b.model_flat.float())
+        opt_path = os.path.join(path, "model_state.pdopt")
+        if self.optimizer is not None and os.path.exists(opt_path):
+            self.optimizer.load_state_dict(
+                torch.load(opt_path, map_location=self.device,
+                           weights_only=False))
+        meta_path = os.path.join(path, "meta_state.pdopt")
+        if os.path.exists(meta_path):
+            meta = torch.load(meta_path, map_location="cpu", weights_only=False)
+            self._load_recovery = {"step": meta["step"], "epoch": meta["epoch"]}
+            self.module.global_step = meta.get("global_step", meta["step"])
+            if "lr_scheduler" in meta:
+                self.lr_scheduler.load_state_dict(meta["lr_scheduler"])
+            if "cpu_rng_state" in meta:
+                torch.set_rng_state(meta["cpu_rng_state"])
+            if "cuda_rng_state" in meta and torch.cuda.is_available():
+                torch.cuda.set_rng_state(meta["cuda_rng_state"])
+        logger.info(f"loaded checkpoint from {path}")

@@ -1,0 +1,80 @@
+"""build_dataloader (reference ppfleetx/data/__init__.py:28-119)."""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import numpy as np
+import torch
+from torch.utils.data import DataLoader
+
+from paddlefleetx_amd.data.gpt_dataset import GPTDataset, GPTSyntheticDataset
+from paddlefleetx_amd.data.sampler import GPTBatchSampler
+from paddlefleetx_amd.parallel.env import (get_data_world_rank,
+                                           get_data_world_size)
+from paddlefleetx_amd.utils.log import logger
+
+_DATASETS = {
+    "GPTDataset": GPTDataset,
+    "GPTSyntheticDataset": GPTSyntheticDataset,
+}
+
+
+def register_dataset(name, cls):
+    _DATASETS[name] = cls
+
+
+def gpt_collate_fn(samples):
+    tokens = torch.stack([s[0] for s in samples])
+    position_ids = torch.stack([s[1] for s in samples])
+    labels = torch.stack([s[2] for s in samples])
+    loss_mask = torch.stack([s[3] for s in samples])
+    return tokens, position_ids, labels, loss_mask
+
+
+def _worker_init(worker_id: int):
+    seed = torch.initial_seed() % (2 ** 31)
+    np.random.seed(seed + worker_id)
+
+
+def build_dataloader(data_cfg, mode: str = "Train", consumed_samples: int = 0,
+                     batch_size: Optional[int] = None) -> Optional[DataLoader]:
+    """data_cfg may be the full config (with Data/Global sections) or just Data."""
+    if "Data" in data_cfg:
+        if batch_size is None:
+            batch_size = data_cfg.get("Global", {}).get("local_batch_size")
+        data_cfg = data_cfg["Data"]
+    section = data_cfg.get(mode) if mode in data_cfg else data_cfg
+    if section is None:
+        return None
+    ds_cfg = dict(section.get("dataset", {}))
+    name = ds_cfg.pop("name", "GPTSyntheticDataset")
+    if name not in _DATASETS:
+        raise ValueError(f"unknown dataset {name}")
+    ds_cfg.setdefault("mode", mode)
+    try:
+        dataset = _DATASETS[name](**ds_cfg)
+    except TypeError:
+        ds_cfg.pop("mode", None)
+        dataset = _DATASETS[name](**ds_cfg)
+
+    loader_cfg = dict(section.get("loader", {}))
+    sampler_cfg = dict(section.get("sampler", {}))
+    if batch_size is None:
+        batch_size = sampler_cfg.get("batch_size", loader_cfg.get("batch_size", 1))
+    batch_size = int(batch_size)
+    sampler = GPTBatchSampler(
+        dataset, batch_size,
+        shuffle=bool(sampler_cfg.get("shuffle", False)),
+        drop_last=bool(sampler_cfg.get("drop_last", True)),
+        rank=get_data_world_rank(), num_replicas=get_data_world_size(),
+        consumed_samples=consumed_samples)
+    loader = DataLoader(
+        dataset, batch_sampler=sampler,
+        num_workers=int(loader_cfg.get("num_workers", 0)),
+        pin_memory=bool(loader_cfg.get("use_shared_memory", False)),
+        collate_fn=gpt_collate_fn, worker_init_fn=_worker_init,
+        persistent_workers=int(loader_cfg.get("num_workers", 0)) > 0)
+    logger.info(f"dataloader[{mode}]: dataset={name} len={len(dataset)} "
+                f"batch_size={batch_size} replicas={get_data_world_size()}")
+    return loader

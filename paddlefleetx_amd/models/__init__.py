@@ -1,0 +1,30 @@
+"""Module factory (reference ppfleetx/models/__init__.py:31-35 build_module)."""
+
+from __future__ import annotations
+
+from paddlefleetx_amd.utils.log import logger
+
+
+def build_module(config):
+    name = config["Model"]["name"]
+    from paddlefleetx_amd.models.language_module import (GPTGenerationModule,
+                                                         GPTModule)
+    table = {
+        "GPTModule": GPTModule,
+        "GPTGenerationModule": GPTGenerationModule,
+    }
+    # late registrations to avoid importing every family eagerly
+    if name == "MoEModule":
+        from paddlefleetx_amd.models.moe_module import MoEModule
+        table["MoEModule"] = MoEModule
+    if name == "GPTEvalModule":
+        from paddlefleetx_amd.models.eval_module import GPTEvalModule
+        table["GPTEvalModule"] = GPTEvalModule
+    if name == "ViTModule" or name == "GeneralClsModule":
+        from paddlefleetx_amd.models.vit_module import GeneralClsModule
+        table["ViTModule"] = GeneralClsModule
+        table["GeneralClsModule"] = GeneralClsModule
+    if name not in table:
+        raise ValueError(f"unknown module {name}")
+    logger.info(f"building module {name}")
+    return table[name](config)

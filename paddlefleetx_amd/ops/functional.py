@@ -1,0 +1,317 @@
+"""Autograd wrappers over the HIP kernels (CUDA) / torch references (CPU).
+
+Kernel parity map (reference -> here):
+  paddle flash_attention (hybrid_model.py:284-301)  -> flash_attention
+  incubate.softmax_mask_fuse_upper_triangle (:325)  -> fused_softmax_causal
+  fused LayerNorm (paddle nn.LayerNorm fused path)  -> layernorm
+  ParallelCrossEntropy inner kernel (:951)          -> cross_entropy(+parallel in parallel/tp.py)
+  FusedAdamW (optims/optimizer.py:31)               -> fused_adamw_flat
+  fused_gemm_epilogue gelu path                     -> bias_gelu
+  ppfleetx/ops/topp_sampling.cu:377                 -> topp_sampling
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional, Tuple
+
+import torch
+
+from paddlefleetx_amd.ops import _reference as ref
+from paddlefleetx_amd.ops import hip_ext, use_hip
+
+__all__ = [
+    "layernorm", "rmsnorm", "FusedLayerNorm", "FusedRMSNorm", "bias_gelu",
+    "flash_attention", "fused_softmax_causal", "cross_entropy",
+    "fused_adamw_flat", "rope", "topp_sampling",
+]
+
+
+# ---------------------------------------------------------------------------
+# LayerNorm / RMSNorm
+# ---------------------------------------------------------------------------
+
+class _LayerNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, eps):
+        shape = x.shape
+        x2 = x.contiguous().view(-1, shape[-1])
+        if use_hip(x2):
+            y, mean, rstd = hip_ext().layernorm_fwd(x2, weight, bias, eps)
+        else:
+            y, mean, rstd = ref.layernorm_fwd(x2, weight, bias, eps)
+        ctx.save_for_backward(x2, weight, mean, rstd)
+        return y.view(shape)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x2, weight, mean, rstd = ctx.saved_tensors
+        dy2 = dy.contiguous().view(-1, dy.shape[-1])
+        if use_hip(x2):
+            dx, dw, db = hip_ext().layernorm_bwd(dy2, x2, weight, mean, rstd)
+        else:
+            dx, dw, db = ref.layernorm_bwd(dy2, x2, weight, mean, rstd)
+        return dx.view(dy.shape), dw.to(weight.dtype), db.to(weight.dtype), None
+
+
+def layernorm(x, weight, bias, eps: float = 1e-5):
+    return _LayerNormFn.apply(x, weight, bias, eps)
+
+
+class _RMSNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, eps):
+        shape = x.shape
+        x2 = x.contiguous().view(-1, shape[-1])
+        if use_hip(x2):
+            y, rstd = hip_ext().rmsnorm_fwd(x2, weight, eps)
+        else:
+            y, rstd = ref.rmsnorm_fwd(x2, weight, eps)
+        ctx.save_for_backward(x2, weight, rstd)
+        return y.view(shape)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x2, weight, rstd = ctx.saved_tensors
+        dy2 = dy.contiguous().view(-1, dy.shape[-1])
+        if use_hip(x2):
+            dx, dw = hip_ext().rmsnorm_bwd(dy2, x2, weight, rstd)
+        else:
+            dx, dw = ref.rmsnorm_bwd(dy2, x2, weight, rstd)
+        return dx.view(dy.shape), dw.to(weight.dtype), None
+
+
+def rmsnorm(x, weight, eps: float = 1e-6):
+    return _RMSNormFn.apply(x, weight, eps)
+
+
+class FusedLayerNorm(torch.nn.Module):
+    def __init__(self, hidden_size: int, eps: float = 1e-5,
+                 dtype: Optional[torch.dtype] = None):
+        super().__init__()
+        self.eps = eps
+        self.weight = torch.nn.Parameter(torch.ones(hidden_size, dtype=dtype))
+        self.bias = torch.nn.Parameter(torch.zeros(hidden_size, dtype=dtype))
+
+    def forward(self, x):
+        return layernorm(x, self.weight, self.bias, self.eps)
+
+
+class FusedRMSNorm(torch.nn.Module):
+    def __init__(self, hidden_size: int, eps: float = 1e-6,
+                 dtype: Optional[torch.dtype] = None):
+        super().__init__()
+        self.eps = eps
+        self.weight = torch.nn.Parameter(torch.ones(hidden_size, dtype=dtype))
+
+    def forward(self, x):
+        return rmsnorm(x, self.weight, self.eps)
+
+
+# ---------------------------------------------------------------------------
+# Bias + GeLU (tanh approx) fusion
+# ---------------------------------------------------------------------------
+
+class _BiasGeluFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, bias):
+        xc = x.contiguous()
+        if use_hip(xc):
+            y = hip_ext().bias_gelu_fwd(xc.view(-1, xc.shape[-1]),
+                                        bias if bias is not None else torch.Tensor())
+            y = y.view(xc.shape)
+        else:
+            y = ref.bias_gelu_fwd(xc, bias)
+        ctx.save_for_backward(xc, bias if bias is not None else None)
+        ctx.has_bias = bias is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        if ctx.has_bias:
+            x, bias = ctx.saved_tensors
+        else:
+            (x,) = ctx.saved_tensors
+            bias = None
+        dyc = dy.contiguous()
+        if use_hip(x):
+            dx, db = hip_ext().bias_gelu_bwd(dyc.view(-1, x.shape[-1]),
+                                             x.view(-1, x.shape[-1]),
+                                             bias if bias is not None else torch.Tensor())
+            dx = dx.view(x.shape)
+            db = db if ctx.has_bias else None
+        else:
+            dx, db = ref.bias_gelu_bwd(dyc, x, bias)
+        if db is not None and bias is not None:
+            db = db.to(bias.dtype)
+        return dx, db
+
+
+def bias_gelu(x, bias=None):
+    return _BiasGeluFn.apply(x, bias)
+
+
+# ---------------------------------------------------------------------------
+# Flash attention (causal), bf16, [B, H, S, D]
+# ---------------------------------------------------------------------------
+
+class _FlashAttnFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, causal, scale):
+        scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
+        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        if use_hip(q):
+            o, lse = hip_ext().attn_fwd(q, k, v, causal, scale)
+        else:
+            o, lse = ref.attention_fwd(q, k, v, causal, scale)
+        ctx.save_for_backward(q, k, v, o, lse)
+        ctx.causal, ctx.scale = causal, scale
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        q, k, v, o, lse = ctx.saved_tensors
+        do = do.contiguous()
+        if use_hip(q):
+            dq, dk, dv = hip_ext().attn_bwd(do, q, k, v, o, lse,
+                                            ctx.causal, ctx.scale)
+        else:
+            dq, dk, dv = ref.attention_bwd(do, q, k, v, o, lse,
+                                           ctx.causal, ctx.scale)
+        return dq, dk, dv, None, None
+
+
+def flash_attention(q, k, v, causal: bool = True, scale: Optional[float] = None):
+    """q,k,v: [B, H, S, D] -> o: [B, H, S, D]. O(S) memory, online softmax."""
+    return _FlashAttnFn.apply(q, k, v, causal, scale)
+
+
+# ---------------------------------------------------------------------------
+# Fused scale + causal mask + softmax (non-flash core_attn path)
+# ---------------------------------------------------------------------------
+
+class _SoftmaxCausalFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, scores, scale):
+        sc = scores.contiguous()
+        if use_hip(sc):
+            y = hip_ext().softmax_causal_fwd(sc, scale)
+        else:
+            y = ref.softmax_causal_fwd(sc, scale)
+        ctx.save_for_backward(y)
+        ctx.scale = scale
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (y,) = ctx.saved_tensors
+        dy = dy.contiguous()
+        if use_hip(y):
+            ds = hip_ext().softmax_causal_bwd(dy, y, ctx.scale)
+        else:
+            ds = ref.softmax_causal_bwd(dy, y, ctx.scale)
+        return ds, None
+
+
+def fused_softmax_causal(scores, scale: float = 1.0):
+    """[B,H,Sq,Sk] scores -> causal softmax(scale*scores). Fused mask, no mask tensor."""
+    return _SoftmaxCausalFn.apply(scores, scale)
+
+
+# ---------------------------------------------------------------------------
+# Cross entropy (single-rank). Vocab-parallel version lives in parallel/tp.py
+# and reuses these kernels plus RCCL allreduce.
+# ---------------------------------------------------------------------------
+
+class _CrossEntropyFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, labels, ignore_index):
+        lc = logits.contiguous()
+        if use_hip(lc):
+            loss, lse = hip_ext().cross_entropy_fwd(lc, labels, ignore_index)
+        else:
+            loss, lse = ref.cross_entropy_fwd(lc, labels, ignore_index)
+        ctx.save_for_backward(lc, labels, lse)
+        ctx.ignore_index = ignore_index
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        logits, labels, lse = ctx.saved_tensors
+        if use_hip(logits):
+            dl = hip_ext().cross_entropy_bwd(dloss.contiguous(), logits, labels,
+                                             lse, ctx.ignore_index)
+        else:
+            dl = ref.cross_entropy_bwd(dloss, logits, labels, lse, ctx.ignore_index)
+        return dl, None, None
+
+
+def cross_entropy(logits, labels, ignore_index: int = -100):
+    """logits [N, V], labels [N] -> per-token loss [N] (fp32)."""
+    return _CrossEntropyFn.apply(logits, labels, ignore_index)
+
+
+# ---------------------------------------------------------------------------
+# Fused AdamW over flat fp32 master buffers
+# ---------------------------------------------------------------------------
+
+def fused_adamw_flat(master: torch.Tensor, grad: torch.Tensor,
+                     exp_avg: torch.Tensor, exp_avg_sq: torch.Tensor,
+                     model_out: Optional[torch.Tensor], lr: float,
+                     beta1: float, beta2: float, eps: float,
+                     weight_decay: float, step: int) -> None:
+    """In-place AdamW on a flat fp32 master buffer; writes bf16 model copy."""
+    if use_hip(master):
+        hip_ext().adamw_flat(master, grad, exp_avg, exp_avg_sq,
+                             model_out if model_out is not None else torch.Tensor(),
+                             lr, beta1, beta2, eps, weight_decay, step)
+    else:
+        ref.adamw_step(master, grad, exp_avg, exp_avg_sq, model_out,
+                       lr, beta1, beta2, eps, weight_decay, step)
+
+
+# ---------------------------------------------------------------------------
+# RoPE
+# ---------------------------------------------------------------------------
+
+class _RopeFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, cos, sin):
+        xc = x.contiguous()
+        if use_hip(xc):
+            y = hip_ext().rope_fwd(xc, cos, sin)
+        else:
+            y = ref.rope_fwd(xc, cos, sin)
+        ctx.save_for_backward(cos, sin)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        cos, sin = ctx.saved_tensors
+        dyc = dy.contiguous()
+        if use_hip(dyc):
+            dx = hip_ext().rope_fwd(dyc, cos, -sin)
+        else:
+            dx = ref.rope_bwd(dyc, cos, sin)
+        return dx, None, None
+
+
+def rope(x, cos, sin):
+    """Rotary embedding, interleaved pairs. x [B,H,S,D], cos/sin [S,D/2] fp32."""
+    return _RopeFn.apply(x, cos, sin)
+
+
+# ---------------------------------------------------------------------------
+# Top-p (nucleus) sampling
+# ---------------------------------------------------------------------------
+
+def topp_sampling(probs: torch.Tensor, top_p: torch.Tensor, seed: int = -1
+                  ) -> Tuple[torch.Tensor, torch.Tensor]:
+    """probs [B, V] rows sum to 1; top_p [B] -> (ids [B,1], prob [B,1]).
+
+    Replaces ppfleetx/ops/topp_sampling.cu (CUB radix sort + prefix scan) with
+    a gfx950 kernel.
+    """
+    if use_hip(probs):
+        return hip_ext().topp_sampling(probs.contiguous(), top_p, seed)
+    return ref.topp_sampling(probs, top_p, seed)

@@ -1,0 +1,32 @@
+"""Optimizer/LR builders (reference ppfleetx/optims/__init__.py:29-74)."""
+
+from __future__ import annotations
+
+from paddlefleetx_amd.optims.lr_scheduler import (ConstantLR,
+                                                  CosineAnnealingWithWarmupDecay,
+                                                  LinearDecayWithWarmup,
+                                                  build_lr_scheduler)
+from paddlefleetx_amd.optims.optimizer import AdamW, FusedAdamW
+
+__all__ = ["build_optimizer", "build_lr_scheduler", "FusedAdamW", "AdamW",
+           "CosineAnnealingWithWarmupDecay", "LinearDecayWithWarmup", "ConstantLR"]
+
+_OPTIMIZERS = {"FusedAdamW": FusedAdamW, "AdamW": AdamW}
+
+
+def build_optimizer(cfg, model, lr_value: float = None):
+    cfg = dict(cfg or {})
+    name = cfg.pop("name", "FusedAdamW")
+    cfg.pop("lr", None)
+    grad_clip = cfg.pop("grad_clip", None)
+    if isinstance(grad_clip, dict):
+        grad_clip = grad_clip.get("clip_norm", 1.0)
+    if name not in _OPTIMIZERS:
+        raise ValueError(f"unknown optimizer {name}")
+    kwargs = {k: v for k, v in cfg.items()
+              if k in ("weight_decay", "beta1", "beta2", "epsilon",
+                       "multi_precision", "tensor_fusion")}
+    return _OPTIMIZERS[name](model.named_parameters(),
+                             lr=lr_value or 1e-4, grad_clip=grad_clip,
+                             **kwargs) if name == "FusedAdamW" else \
+        _OPTIMIZERS[name](model.named_parameters(), lr=lr_value or 1e-4, **kwargs)

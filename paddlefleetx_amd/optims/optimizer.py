@@ -1,0 +1,225 @@
+"""FusedAdamW over fused flat parameter/gradient storages.
+
+Design (MI355X-native): every parameter's data is re-pointed into a flat
+per-group buffer (<=256 MB buckets, 256-B aligned — reference
+utils/tensor_fusion_helper.py:30-106 semantics); gradients accumulate into
+a flat fp32 main-grad buffer via post-accumulate hooks (reference
+distributed/apis/amp.py:30-68 MixPrecisionLayer); the AdamW update is ONE
+hand-written HIP kernel launch per bucket on the flat fp32 master /
+exp_avg / exp_avg_sq buffers which also writes back the bf16 model copy
+(reference optims/optimizer.py:31 FusedAdamW + paddle fused adam kernel).
+
+DP/sharding allreduce operates directly on the flat grad buffers
+(engine calls `reduce_gradients`).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Dict, Iterable, List, Optional, Tuple
+
+import torch
+import torch.distributed as dist
+
+from paddlefleetx_amd.ops import fused_adamw_flat
+from paddlefleetx_amd.utils.log import logger
+
+__all__ = ["FusedAdamW", "AdamW"]
+
+ALIGN_ELEMS = 64  # 256 B / 4 B
+BUCKET_BYTES = 256 * 1024 * 1024
+
+
+def _no_decay(name: str) -> bool:
+    # reference decay filter (optims/optimizer.py:44-47)
+    return ("bias" in name) or ("norm" in name) or name.endswith("b_0") \
+        or ("ln" in name.split(".")[-2:][0] if "." in name else False)
+
+
+class _Bucket:
+    """One fused storage: flat model copy + fp32 master/m/v/main_grad."""
+
+    def __init__(self, params: List[torch.nn.Parameter], dtype: torch.dtype,
+                 device: torch.device, weight_decay: float):
+        self.params = params
+        self.dtype = dtype
+        self.weight_decay = weight_decay
+        offs = []
+        total = 0
+        for p in params:
+            offs.append(total)
+            n = p.numel()
+            total += (n + ALIGN_ELEMS - 1) // ALIGN_ELEMS * ALIGN_ELEMS
+        self.numel = total
+        self.offsets = offs
+        self.model_flat = torch.zeros(total, dtype=dtype, device=device)
+        # re-point param data into the flat buffer
+        for p, off in zip(params, offs):
+            n = p.numel()
+            self.model_flat[off:off + n].copy_(p.data.reshape(-1).to(dtype))
+            p.data = self.model_flat[off:off + n].view(p.shape)
+        self.master = self.model_flat.float()
+        self.exp_avg = torch.zeros(total, dtype=torch.float32, device=device)
+        self.exp_avg_sq = torch.zeros(total, dtype=torch.float32, device=device)
+        self.main_grad = torch.zeros(total, dtype=torch.float32, device=device)
+        # expose per-param main_grad views (reference amp.py main_grad)
+        for p, off in zip(params, offs):
+            p.main_grad = self.main_grad[off:off + p.numel()].view(p.shape)
+
+
+class FusedAdamW(torch.optim.Optimizer):
+    """AdamW on fused flat buffers; grads accumulate in fp32 main_grad."""
+
+    def __init__(self, named_params: Iterable[Tuple[str, torch.nn.Parameter]],
+                 lr: float = 1e-4, beta1: float = 0.9, beta2: float = 0.95,
+                 epsilon: float = 1e-8, weight_decay: float = 0.01,
+                 multi_precision: bool = True, grad_clip: Optional[float] = None,
+                 tensor_fusion: bool = True, **unused):
+        named = [(n, p) for n, p in named_params if p.requires_grad]
+        params = [p for _, p in named]
+        defaults = dict(lr=lr, beta1=beta1, beta2=beta2, epsilon=epsilon,
+                        weight_decay=weight_decay)
+        super().__init__(params, defaults)
+        self.lr = lr
+        self.beta1, self.beta2, self.eps = beta1, beta2, epsilon
+        self.weight_decay = weight_decay
+        self.grad_clip = grad_clip
+        self._step = 0
+        self.found_inf = False
+
+        device = params[0].device if params else torch.device("cpu")
+        # split by (dtype, decay?) and pack into <=256MB buckets
+        self.buckets: List[_Bucket] = []
+        groups: Dict[Tuple[torch.dtype, bool], List] = {}
+        for n, p in named:
+            groups.setdefault((p.dtype, not _no_decay(n)), []).append(p)
+        for (dtype, decay), ps in groups.items():
+            cur: List[torch.nn.Parameter] = []
+            cur_bytes = 0
+            for p in ps:
+                nbytes = p.numel() * 4
+                if cur and cur_bytes + nbytes > BUCKET_BYTES:
+                    self.buckets.append(_Bucket(cur, dtype, p.device,
+                                                weight_decay if decay else 0.0))
+                    cur, cur_bytes = [], 0
+                cur.append(p)
+                cur_bytes += nbytes
+            if cur:
+                self.buckets.append(_Bucket(cur, dtype, cur[0].device,
+                                            weight_decay if decay else 0.0))
+        # accumulate bf16 autograd grads into fp32 main_grad, free the bf16 grad
+        self._hooks = []
+        for n, p in named:
+            self._hooks.append(
+                p.register_post_accumulate_grad_hook(_MainGradHook()))
+        n_params = sum(p.numel() for p in params)
+        logger.info(f"FusedAdamW: {len(params)} params ({n_params/1e6:.1f}M) "
+                    f"in {len(self.buckets)} fused buckets")
+
+    # --- gradient plumbing -------------------------------------------------
+    def zero_grad(self, set_to_none: bool = True):
+        for b in self.buckets:
+            b.main_grad.zero_()
+        super().zero_grad(set_to_none=set_to_none)
+
+    def grad_buffers(self) -> List[torch.Tensor]:
+        return [b.main_grad for b in self.buckets]
+
+    def reduce_gradients(self, group, avg_factor: Optional[float] = None):
+        """Allreduce fused grad buffers over a process group (DP)."""
+        if group is None or getattr(group, "world_size", 1) == 1:
+            return
+        pg = group.group if hasattr(group, "group") else group
+        ws = group.world_size if hasattr(group, "world_size") else dist.get_world_size(pg)
+        for b in self.buckets:
+            dist.all_reduce(b.main_grad, group=pg)
+            b.main_grad.div_(avg_factor or ws)
+
+    # --- norm / clip -------------------------------------------------------
+    def grad_global_norm(self, extra_sq: float = 0.0,
+                         mp_group=None, pp_group=None) -> torch.Tensor:
+        """Global grad norm; TP-aware: mp-sharded param grads summed across mp.
+
+        To avoid double counting, replicated (non-mp) params only counted on
+        mp_rank 0; then allreduce over mp and pp groups.
+        """
+        device = self.buckets[0].main_grad.device if self.buckets else "cpu"
+        sq = torch.zeros((), dtype=torch.float32, device=device)
+        mp_ws = mp_group.world_size if mp_group is not None else 1
+        for b in self.buckets:
+            if mp_ws == 1:
+                sq += b.main_grad.pow(2).sum()
+            else:
+                for p, off in zip(b.params, b.offsets):
+                    g = b.main_grad[off:off + p.numel()]
+                    if getattr(p, "is_mp", False) or mp_group.rank == 0:
+                        sq += g.pow(2).sum()
+        sq += extra_sq
+        if mp_ws > 1:
+            dist.all_reduce(sq, group=mp_group.group)
+        if pp_group is not None and pp_group.world_size > 1:
+            dist.all_reduce(sq, group=pp_group.group)
+        return sq.sqrt()
+
+    def clip_grads(self, max_norm: float, mp_group=None, pp_group=None):
+        norm = self.grad_global_norm(mp_group=mp_group, pp_group=pp_group)
+        scale = max_norm / (norm + 1e-6)
+        if scale < 1.0:
+            for b in self.buckets:
+                b.main_grad.mul_(scale)
+        return norm
+
+    # --- step --------------------------------------------------------------
+    @torch.no_grad()
+    def step(self, closure=None, lr: Optional[float] = None):
+        if lr is not None:
+            self.lr = lr
+        self._step += 1
+        for b in self.buckets:
+            fused_adamw_flat(b.master, b.main_grad, b.exp_avg, b.exp_avg_sq,
+                             b.model_flat, self.lr, self.beta1, self.beta2,
+                             self.eps, b.weight_decay, self._step)
+
+    # --- checkpoint --------------------------------------------------------
+    def state_dict(self):
+        return {
+            "step": self._step,
+            "lr": self.lr,
+            "buckets": [{
+                "master": b.master,
+                "exp_avg": b.exp_avg,
+                "exp_avg_sq": b.exp_avg_sq,
+            } for b in self.buckets],
+        }
+
+    def load_state_dict(self, sd):
+        self._step = sd["step"]
+        self.lr = sd.get("lr", self.lr)
+        assert len(sd["buckets"]) == len(self.buckets), \
+            "optimizer bucket layout mismatch on load"
+        for b, s in zip(self.buckets, sd["buckets"]):
+            b.master.copy_(s["master"])
+            b.exp_avg.copy_(s["exp_avg"])
+            b.exp_avg_sq.copy_(s["exp_avg_sq"])
+            b.model_flat.copy_(b.master.to(b.dtype))
+
+
+class _MainGradHook:
+    def __call__(self, p: torch.nn.Parameter):
+        if p.grad is not None:
+            p.main_grad.add_(p.grad.float())
+            p.grad = None
+
+
+class AdamW(torch.optim.AdamW):
+    """Plain torch AdamW (no fusion) — kept for small/CPU runs."""
+
+    def __init__(self, named_params, lr=1e-4, beta1=0.9, beta2=0.95,
+                 epsilon=1e-8, weight_decay=0.01, **unused):
+        named = [(n, p) for n, p in named_params if p.requires_grad]
+        decay = [p for n, p in named if not _no_decay(n)]
+        nodecay = [p for n, p in named if _no_decay(n)]
+        super().__init__([
+            {"params": decay, "weight_decay": weight_decay},
+            {"params": nodecay, "weight_decay": 0.0},
+        ], lr=lr, betas=(beta1, beta2), eps=epsilon)

@@ -1,0 +1,20 @@
+import os
+import sys
+
+import pytest
+import torch
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
+
+
+def pytest_configure(config):
+    config.addinivalue_line("markers", "gpu: needs an MI355X GPU (run via gpurun)")
+
+
+def pytest_collection_modifyitems(config, items):
+    if torch.cuda.is_available():
+        return
+    skip = pytest.mark.skip(reason="no GPU on this host")
+    for item in items:
+        if "gpu" in item.keywords:
+            item.add_marker(skip)
