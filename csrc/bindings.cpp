@@ -1,0 +1,79 @@
+// Torch extension bindings for the gfx950 fused-op kernels.
+#include <torch/extension.h>
+
+#include <vector>
+
+// layernorm.hip
+std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w,
+                                         torch::Tensor b, double eps);
+std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
+                                         torch::Tensor w, torch::Tensor mean,
+                                         torch::Tensor rstd);
+std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w,
+                                       double eps);
+std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
+                                       torch::Tensor w, torch::Tensor rstd);
+// elementwise.hip
+torch::Tensor bias_gelu_fwd(torch::Tensor x, torch::Tensor bias);
+std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
+                                         torch::Tensor bias);
+void adamw_flat(torch::Tensor master, torch::Tensor grad, torch::Tensor m,
+                torch::Tensor v, torch::Tensor model, double lr, double beta1,
+                double beta2, double eps, double wd, long step);
+torch::Tensor rope_fwd(torch::Tensor x, torch::Tensor cos, torch::Tensor sin);
+// softmax.hip
+torch::Tensor softmax_causal_fwd(torch::Tensor s, double scale);
+torch::Tensor softmax_causal_bwd(torch::Tensor dy, torch::Tensor y,
+                                 double scale);
+std::vector<torch::Tensor> cross_entropy_fwd(torch::Tensor logits,
+                                             torch::Tensor labels,
+                                             long ignore_index);
+torch::Tensor cross_entropy_bwd(torch::Tensor dloss, torch::Tensor logits,
+                                torch::Tensor labels, torch::Tensor lse,
+                                long ignore_index);
+torch::Tensor row_max(torch::Tensor x);
+torch::Tensor row_sumexp(torch::Tensor x, torch::Tensor gmax);
+torch::Tensor gather_label_logit(torch::Tensor x, torch::Tensor labels,
+                                 long start, long ignore_index);
+torch::Tensor vp_ce_bwd(torch::Tensor dloss, torch::Tensor logits,
+                        torch::Tensor labels, torch::Tensor lse, long start,
+                        long ignore_index);
+// attention.hip
+std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor v, bool causal,
+                                    double scale);
+std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
+                                    torch::Tensor k, torch::Tensor v,
+                                    torch::Tensor o, torch::Tensor lse,
+                                    bool causal, double scale);
+// topp.hip
+std::vector<torch::Tensor> topp_select(torch::Tensor sorted_p,
+                                       torch::Tensor sorted_idx,
+                                       torch::Tensor top_p, torch::Tensor u);
+// mfma_probe.hip
+torch::Tensor mfma_gemm16_probe(torch::Tensor a, torch::Tensor b);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("layernorm_fwd", &layernorm_fwd, "fused LayerNorm fwd (gfx950)");
+  m.def("layernorm_bwd", &layernorm_bwd, "fused LayerNorm bwd");
+  m.def("rmsnorm_fwd", &rmsnorm_fwd, "fused RMSNorm fwd");
+  m.def("rmsnorm_bwd", &rmsnorm_bwd, "fused RMSNorm bwd");
+  m.def("bias_gelu_fwd", &bias_gelu_fwd, "fused bias+gelu fwd");
+  m.def("bias_gelu_bwd", &bias_gelu_bwd, "fused bias+gelu bwd");
+  m.def("adamw_flat", &adamw_flat, "fused AdamW on flat buffers");
+  m.def("rope_fwd", &rope_fwd, "rotary embedding fwd");
+  m.def("softmax_causal_fwd", &softmax_causal_fwd,
+        "fused scale+causal-mask+softmax fwd");
+  m.def("softmax_causal_bwd", &softmax_causal_bwd, "causal softmax bwd");
+  m.def("cross_entropy_fwd", &cross_entropy_fwd, "softmax CE fwd");
+  m.def("cross_entropy_bwd", &cross_entropy_bwd, "softmax CE bwd");
+  m.def("row_max", &row_max, "rowwise max (vocab-parallel CE)");
+  m.def("row_sumexp", &row_sumexp, "rowwise sum(exp(x-m))");
+  m.def("gather_label_logit", &gather_label_logit, "gather label logits");
+  m.def("vp_ce_bwd", &vp_ce_bwd, "vocab-parallel CE bwd");
+  m.def("attn_fwd", &attn_fwd, "flash attention fwd (MFMA, causal)");
+  m.def("attn_bwd", &attn_bwd, "flash attention bwd (MFMA)");
+  m.def("topp_select", &topp_select, "top-p nucleus cutoff + draw");
+  m.def("mfma_gemm16_probe", &mfma_gemm16_probe,
+        "debug: 16x16x32 MFMA fragment-layout probe");
+}

@@ -1,0 +1,109 @@
+// Common helpers for gfx950 (MI355X / CDNA4) kernels.
+// Wavefront = 64 lanes; LDS 160 KiB/CU; vectorize bf16 as short4/short8.
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <hip/hip_fp16.h>
+
+#define WAVE 64
+#define DEV_INLINE __device__ __forceinline__
+
+typedef __attribute__((ext_vector_type(2))) short short2v;
+typedef __attribute__((ext_vector_type(4))) short short4v;
+typedef __attribute__((ext_vector_type(8))) short short8v;
+typedef __attribute__((ext_vector_type(4))) float float4v;
+typedef __attribute__((ext_vector_type(2))) float float2v;
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
+
+// MFMA fragment types (gfx950 16x16x32 bf16: 8 bf16 in = 4 VGPRs, 4 f32 acc)
+typedef __attribute__((ext_vector_type(8))) __bf16 mfma_bf16x8;
+typedef __attribute__((ext_vector_type(4))) float mfma_f32x4;
+
+DEV_INLINE float bf2f(__hip_bfloat16 x) { return __bfloat162float(x); }
+DEV_INLINE __hip_bfloat16 f2bf(float x) { return __float2bfloat16(x); }
+
+DEV_INLINE float bf_raw2f(unsigned short u) {
+  union { unsigned int i; float f; } v;
+  v.i = ((unsigned int)u) << 16;
+  return v.f;
+}
+DEV_INLINE unsigned short f2bf_raw(float f) {
+  union { unsigned int i; float f; } v;
+  v.f = f;
+  unsigned int lsb = (v.i >> 16) & 1u;
+  v.i += 0x7fffu + lsb;  // round-to-nearest-even
+  return (unsigned short)(v.i >> 16);
+}
+
+// ---- wave reductions (64-lane) ----
+DEV_INLINE float wave_reduce_sum(float x) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) x += __shfl_xor(x, off, WAVE);
+  return x;
+}
+DEV_INLINE float wave_reduce_max(float x) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) x = fmaxf(x, __shfl_xor(x, off, WAVE));
+  return x;
+}
+// reduce within contiguous 16-lane groups
+DEV_INLINE float group16_reduce_sum(float x) {
+#pragma unroll
+  for (int off = 8; off > 0; off >>= 1) x += __shfl_xor(x, off, WAVE);
+  return x;
+}
+DEV_INLINE float group16_reduce_max(float x) {
+#pragma unroll
+  for (int off = 8; off > 0; off >>= 1) x = fmaxf(x, __shfl_xor(x, off, WAVE));
+  return x;
+}
+
+// ---- block reductions via LDS (expects <= 16 waves) ----
+template <int BLOCK>
+DEV_INLINE float block_reduce_sum(float x, float* lds_scratch) {
+  constexpr int NW = BLOCK / WAVE;
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  x = wave_reduce_sum(x);
+  if (lane == 0) lds_scratch[wid] = x;
+  __syncthreads();
+  float r = (threadIdx.x < NW) ? lds_scratch[threadIdx.x] : 0.f;
+#pragma unroll
+  for (int off = NW / 2; off > 0; off >>= 1) r += __shfl_xor(r, off, WAVE);
+  r = __shfl(r, 0, WAVE);
+  if (threadIdx.x == 0) lds_scratch[0] = r;
+  __syncthreads();
+  r = lds_scratch[0];
+  __syncthreads();
+  return r;
+}
+
+template <int BLOCK>
+DEV_INLINE float block_reduce_max(float x, float* lds_scratch) {
+  constexpr int NW = BLOCK / WAVE;
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  x = wave_reduce_max(x);
+  if (lane == 0) lds_scratch[wid] = x;
+  __syncthreads();
+  float r = (threadIdx.x < NW) ? lds_scratch[threadIdx.x] : -INFINITY;
+#pragma unroll
+  for (int off = NW / 2; off > 0; off >>= 1) r = fmaxf(r, __shfl_xor(r, off, WAVE));
+  r = __shfl(r, 0, WAVE);
+  if (threadIdx.x == 0) lds_scratch[0] = r;
+  __syncthreads();
+  r = lds_scratch[0];
+  __syncthreads();
+  return r;
+}
+
+#define HIP_CHECK(x)                                                       \
+  do {                                                                     \
+    hipError_t _e = (x);                                                   \
+    if (_e != hipSuccess) {                                                \
+      throw std::runtime_error(std::string("HIP error: ") +                \
+                               hipGetErrorString(_e));                     \
+    }                                                                      \
+  } while (0)

@@ -1,0 +1,275 @@
+#include "hip/hip_runtime.h"
+// Elementwise fused kernels: bias+gelu (fwd/bwd), flat AdamW, RoPE.
+// All memory-bound: vectorized loads (G13), grid-stride (G11).
+#include "common.h"
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+namespace {
+
+constexpr int BLOCK = 256;
+constexpr float GELU_K = 0.7978845608028654f;  // sqrt(2/pi)
+
+DEV_INLINE float gelu_tanh(float x) {
+  float t = tanhf(GELU_K * (x + 0.044715f * x * x * x));
+  return 0.5f * x * (1.f + t);
+}
+DEV_INLINE float gelu_tanh_grad(float x) {
+  float t = tanhf(GELU_K * (x + 0.044715f * x * x * x));
+  return 0.5f * (1.f + t) +
+         0.5f * x * (1.f - t * t) * GELU_K * (1.f + 3.f * 0.044715f * x * x);
+}
+
+// ---- bias + gelu: x [N, H] + bias[H] ----
+template <typename T, bool HAS_BIAS>
+__global__ void bias_gelu_fwd_kernel(const T* __restrict__ x,
+                                     const T* __restrict__ bias,
+                                     T* __restrict__ y, long total, int H) {
+  const int VEC = 4;
+  for (long i = ((long)blockIdx.x * BLOCK + threadIdx.x) * VEC; i < total;
+       i += (long)gridDim.x * BLOCK * VEC) {
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      float v;
+      if constexpr (sizeof(T) == 2)
+        v = bf_raw2f(((const unsigned short*)x)[i + j]);
+      else
+        v = ((const float*)x)[i + j];
+      if constexpr (HAS_BIAS) {
+        int col = (int)((i + j) % H);
+        if constexpr (sizeof(T) == 2)
+          v += bf_raw2f(((const unsigned short*)bias)[col]);
+        else
+          v += ((const float*)bias)[col];
+      }
+      float o = gelu_tanh(v);
+      if constexpr (sizeof(T) == 2)
+        ((unsigned short*)y)[i + j] = f2bf_raw(o);
+      else
+        ((float*)y)[i + j] = o;
+    }
+  }
+}
+
+template <typename T, bool HAS_BIAS>
+__global__ void bias_gelu_bwd_kernel(const T* __restrict__ dy,
+                                     const T* __restrict__ x,
+                                     const T* __restrict__ bias,
+                                     T* __restrict__ dx, long total, int H) {
+  const int VEC = 4;
+  for (long i = ((long)blockIdx.x * BLOCK + threadIdx.x) * VEC; i < total;
+       i += (long)gridDim.x * BLOCK * VEC) {
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      float v, d;
+      if constexpr (sizeof(T) == 2) {
+        v = bf_raw2f(((const unsigned short*)x)[i + j]);
+        d = bf_raw2f(((const unsigned short*)dy)[i + j]);
+      } else {
+        v = ((const float*)x)[i + j];
+        d = ((const float*)dy)[i + j];
+      }
+      if constexpr (HAS_BIAS) {
+        int col = (int)((i + j) % H);
+        if constexpr (sizeof(T) == 2)
+          v += bf_raw2f(((const unsigned short*)bias)[col]);
+        else
+          v += ((const float*)bias)[col];
+      }
+      float o = d * gelu_tanh_grad(v);
+      if constexpr (sizeof(T) == 2)
+        ((unsigned short*)dx)[i + j] = f2bf_raw(o);
+      else
+        ((float*)dx)[i + j] = o;
+    }
+  }
+}
+
+// ---- AdamW over flat fp32 buffers, writes back low-precision model copy ----
+template <typename TM>  // model dtype
+__global__ void adamw_flat_kernel(float* __restrict__ master,
+                                  const float* __restrict__ grad,
+                                  float* __restrict__ m, float* __restrict__ v,
+                                  TM* __restrict__ model, long n, float lr,
+                                  float beta1, float beta2, float eps,
+                                  float wd, float bc1, float bc2) {
+  const int VEC = 4;
+  for (long i = ((long)blockIdx.x * BLOCK + threadIdx.x) * VEC; i < n;
+       i += (long)gridDim.x * BLOCK * VEC) {
+    float4v pm = *reinterpret_cast<float4v*>(master + i);
+    float4v pg = *reinterpret_cast<const float4v*>(grad + i);
+    float4v pmm = *reinterpret_cast<float4v*>(m + i);
+    float4v pvv = *reinterpret_cast<float4v*>(v + i);
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      float g = ((float*)&pg)[j];
+      float mm = beta1 * ((float*)&pmm)[j] + (1.f - beta1) * g;
+      float vv = beta2 * ((float*)&pvv)[j] + (1.f - beta2) * g * g;
+      float p = ((float*)&pm)[j];
+      p *= (1.f - lr * wd);
+      float denom = sqrtf(vv / bc2) + eps;
+      p -= lr * (mm / bc1) / denom;
+      ((float*)&pmm)[j] = mm;
+      ((float*)&pvv)[j] = vv;
+      ((float*)&pm)[j] = p;
+    }
+    *reinterpret_cast<float4v*>(master + i) = pm;
+    *reinterpret_cast<float4v*>(m + i) = pmm;
+    *reinterpret_cast<float4v*>(v + i) = pvv;
+    if (model) {
+      if constexpr (sizeof(TM) == 2) {
+        short4v out;
+#pragma unroll
+        for (int j = 0; j < VEC; ++j)
+          ((unsigned short*)&out)[j] = f2bf_raw(((float*)&pm)[j]);
+        *reinterpret_cast<short4v*>((unsigned short*)model + i) = out;
+      } else {
+        *reinterpret_cast<float4v*>((float*)model + i) = pm;
+      }
+    }
+  }
+}
+
+// ---- RoPE: x [B, H, S, D], cos/sin [S, D/2] fp32, interleaved pairs ----
+template <typename T>
+__global__ void rope_kernel(const T* __restrict__ x, const float* __restrict__ cs,
+                            const float* __restrict__ sn, T* __restrict__ y,
+                            long rows, int S, int D) {
+  // one row = one (b, h, s); lanes cover D/2 pairs
+  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+    int s = (int)(row % S);
+    const T* xr = x + row * D;
+    T* yr = y + row * D;
+    const float* cr = cs + (long)s * (D / 2);
+    const float* sr = sn + (long)s * (D / 2);
+    for (int p = threadIdx.x; p < D / 2; p += BLOCK) {
+      float x1, x2;
+      if constexpr (sizeof(T) == 2) {
+        x1 = bf_raw2f(((const unsigned short*)xr)[2 * p]);
+        x2 = bf_raw2f(((const unsigned short*)xr)[2 * p + 1]);
+      } else {
+        x1 = ((const float*)xr)[2 * p];
+        x2 = ((const float*)xr)[2 * p + 1];
+      }
+      float c = cr[p], s_ = sr[p];
+      float o1 = x1 * c - x2 * s_;
+      float o2 = x2 * c + x1 * s_;
+      if constexpr (sizeof(T) == 2) {
+        ((unsigned short*)yr)[2 * p] = f2bf_raw(o1);
+        ((unsigned short*)yr)[2 * p + 1] = f2bf_raw(o2);
+      } else {
+        ((float*)yr)[2 * p] = o1;
+        ((float*)yr)[2 * p + 1] = o2;
+      }
+    }
+  }
+}
+
+long grid_for(long total, int per_block) {
+  long g = (total + per_block - 1) / per_block;
+  return std::min<long>(g, 2048);
+}
+
+}  // namespace
+
+torch::Tensor bias_gelu_fwd(torch::Tensor x, torch::Tensor bias) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  long total = x.numel();
+  int H = x.size(-1);
+  TORCH_CHECK(total % 4 == 0, "numel must be divisible by 4");
+  bool has_bias = bias.defined() && bias.numel() > 0;
+  if (has_bias) TORCH_CHECK(H % 4 == 0);
+  auto y = torch::empty_like(x);
+  auto stream = at::hip::getCurrentHIPStream();
+  dim3 grid(grid_for(total, BLOCK * 4));
+#define LAUNCH_BG(T, HB)                                                     \
+  hipLaunchKernelGGL((bias_gelu_fwd_kernel<T, HB>), grid, dim3(BLOCK), 0,    \
+                     stream, (const T*)x.data_ptr(),                         \
+                     has_bias ? (const T*)bias.data_ptr() : nullptr,         \
+                     (T*)y.data_ptr(), total, H)
+  if (x.scalar_type() == torch::kBFloat16) {
+    if (has_bias) LAUNCH_BG(__hip_bfloat16, true);
+    else LAUNCH_BG(__hip_bfloat16, false);
+  } else {
+    if (has_bias) LAUNCH_BG(float, true);
+    else LAUNCH_BG(float, false);
+  }
+#undef LAUNCH_BG
+  return y;
+}
+
+std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
+                                         torch::Tensor bias) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && dy.is_contiguous());
+  long total = x.numel();
+  int H = x.size(-1);
+  bool has_bias = bias.defined() && bias.numel() > 0;
+  auto dx = torch::empty_like(x);
+  auto stream = at::hip::getCurrentHIPStream();
+  dim3 grid(grid_for(total, BLOCK * 4));
+#define LAUNCH_BG(T, HB)                                                     \
+  hipLaunchKernelGGL((bias_gelu_bwd_kernel<T, HB>), grid, dim3(BLOCK), 0,    \
+                     stream, (const T*)dy.data_ptr(), (const T*)x.data_ptr(), \
+                     has_bias ? (const T*)bias.data_ptr() : nullptr,         \
+                     (T*)dx.data_ptr(), total, H)
+  if (x.scalar_type() == torch::kBFloat16) {
+    if (has_bias) LAUNCH_BG(__hip_bfloat16, true);
+    else LAUNCH_BG(__hip_bfloat16, false);
+  } else {
+    if (has_bias) LAUNCH_BG(float, true);
+    else LAUNCH_BG(float, false);
+  }
+#undef LAUNCH_BG
+  torch::Tensor db;
+  if (has_bias) db = dx.view({-1, H}).to(torch::kFloat).sum(0);
+  return {dx, db};
+}
+
+void adamw_flat(torch::Tensor master, torch::Tensor grad, torch::Tensor m,
+                torch::Tensor v, torch::Tensor model, double lr, double beta1,
+                double beta2, double eps, double wd, long step) {
+  TORCH_CHECK(master.is_cuda() && master.scalar_type() == torch::kFloat);
+  long n = master.numel();
+  TORCH_CHECK(n % 4 == 0);
+  bool has_model = model.defined() && model.numel() > 0;
+  float bc1 = 1.f - powf((float)beta1, (float)step);
+  float bc2 = 1.f - powf((float)beta2, (float)step);
+  auto stream = at::hip::getCurrentHIPStream();
+  dim3 grid(grid_for(n, BLOCK * 4));
+  if (has_model && model.scalar_type() == torch::kBFloat16) {
+    hipLaunchKernelGGL((adamw_flat_kernel<__hip_bfloat16>), grid, dim3(BLOCK),
+                       0, stream, master.data_ptr<float>(),
+                       grad.data_ptr<float>(), m.data_ptr<float>(),
+                       v.data_ptr<float>(), (__hip_bfloat16*)model.data_ptr(),
+                       n, (float)lr, (float)beta1, (float)beta2, (float)eps,
+                       (float)wd, bc1, bc2);
+  } else {
+    hipLaunchKernelGGL((adamw_flat_kernel<float>), grid, dim3(BLOCK), 0,
+                       stream, master.data_ptr<float>(), grad.data_ptr<float>(),
+                       m.data_ptr<float>(), v.data_ptr<float>(),
+                       has_model ? model.data_ptr<float>() : nullptr, n,
+                       (float)lr, (float)beta1, (float)beta2, (float)eps,
+                       (float)wd, bc1, bc2);
+  }
+}
+
+torch::Tensor rope_fwd(torch::Tensor x, torch::Tensor cos, torch::Tensor sin) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 4);
+  int S = x.size(2), D = x.size(3);
+  long rows = x.numel() / D;
+  auto y = torch::empty_like(x);
+  auto stream = at::hip::getCurrentHIPStream();
+  dim3 grid((unsigned)std::min<long>(rows, 8192));
+  if (x.scalar_type() == torch::kBFloat16) {
+    hipLaunchKernelGGL((rope_kernel<__hip_bfloat16>), grid, dim3(BLOCK), 0,
+                       stream, (const __hip_bfloat16*)x.data_ptr(),
+                       cos.data_ptr<float>(), sin.data_ptr<float>(),
+                       (__hip_bfloat16*)y.data_ptr(), rows, S, D);
+  } else {
+    hipLaunchKernelGGL((rope_kernel<float>), grid, dim3(BLOCK), 0, stream,
+                       x.data_ptr<float>(), cos.data_ptr<float>(),
+                       sin.data_ptr<float>(), y.data_ptr<float>(), rows, S, D);
+  }
+  return y;
+}

@@ -313,5 +313,17 @@ def topp_sampling(probs: torch.Tensor, top_p: torch.Tensor, seed: int = -1
     a gfx950 kernel.
     """
     if use_hip(probs):
-        return hip_ext().topp_sampling(probs.contiguous(), top_p, seed)
+        # rocPRIM radix sort (torch.sort) + HIP scan/cutoff/draw kernel
+        pf = probs.float().contiguous()
+        sorted_p, sorted_idx = torch.sort(pf, dim=-1, descending=True)
+        if seed >= 0:
+            g = torch.Generator(device=probs.device)
+            g.manual_seed(seed)
+            u = torch.rand(probs.shape[0], device=probs.device, generator=g)
+        else:
+            u = torch.rand(probs.shape[0], device=probs.device)
+        ids, pp = hip_ext().topp_select(sorted_p, sorted_idx,
+                                        top_p.float().contiguous(),
+                                        u.contiguous())
+        return ids, pp.to(probs.dtype)
     return ref.topp_sampling(probs, top_p, seed)

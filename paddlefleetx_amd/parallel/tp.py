@@ -234,24 +234,42 @@ class _VocabParallelCE(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, logits, labels, vocab_start, vocab_end, group, ignore_index):
-        lf = logits.float()
-        lmax = lf.max(dim=-1).values
-        if group is not None:
-            dist.all_reduce(lmax, op=dist.ReduceOp.MAX, group=group)
-        sumexp = torch.exp(lf - lmax[:, None]).sum(dim=-1)
-        if group is not None:
-            dist.all_reduce(sumexp, group=group)
-        lse = torch.log(sumexp) + lmax
-
-        valid = labels != ignore_index
-        in_part = (labels >= vocab_start) & (labels < vocab_end) & valid
-        local_lab = (labels - vocab_start).masked_fill(~in_part, 0)
-        picked = lf.gather(1, local_lab[:, None]).squeeze(1)
-        picked = torch.where(in_part, picked, torch.zeros_like(picked))
-        if group is not None:
-            dist.all_reduce(picked, group=group)
-        loss = torch.where(valid, lse - picked, torch.zeros_like(lse))
-        ctx.save_for_backward(logits, labels, lse)
+        if logits.is_cuda:
+            from paddlefleetx_amd.ops import hip_ext
+            ext = hip_ext()
+            lc = logits.contiguous()
+            lmax = ext.row_max(lc)
+            if group is not None:
+                dist.all_reduce(lmax, op=dist.ReduceOp.MAX, group=group)
+            sumexp = ext.row_sumexp(lc, lmax)
+            if group is not None:
+                dist.all_reduce(sumexp, group=group)
+            lse = torch.log(sumexp) + lmax
+            picked = ext.gather_label_logit(lc, labels, vocab_start,
+                                            ignore_index)
+            if group is not None:
+                dist.all_reduce(picked, group=group)
+            valid = labels != ignore_index
+            loss = torch.where(valid, lse - picked, torch.zeros_like(lse))
+            ctx.save_for_backward(lc, labels, lse)
+        else:
+            lf = logits.float()
+            lmax = lf.max(dim=-1).values
+            if group is not None:
+                dist.all_reduce(lmax, op=dist.ReduceOp.MAX, group=group)
+            sumexp = torch.exp(lf - lmax[:, None]).sum(dim=-1)
+            if group is not None:
+                dist.all_reduce(sumexp, group=group)
+            lse = torch.log(sumexp) + lmax
+            valid = labels != ignore_index
+            in_part = (labels >= vocab_start) & (labels < vocab_end) & valid
+            local_lab = (labels - vocab_start).masked_fill(~in_part, 0)
+            picked = lf.gather(1, local_lab[:, None]).squeeze(1)
+            picked = torch.where(in_part, picked, torch.zeros_like(picked))
+            if group is not None:
+                dist.all_reduce(picked, group=group)
+            loss = torch.where(valid, lse - picked, torch.zeros_like(lse))
+            ctx.save_for_backward(logits, labels, lse)
         ctx.vocab_start, ctx.vocab_end = vocab_start, vocab_end
         ctx.ignore_index = ignore_index
         return loss
@@ -259,6 +277,11 @@ class _VocabParallelCE(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dloss):
         logits, labels, lse = ctx.saved_tensors
+        if logits.is_cuda:
+            from paddlefleetx_amd.ops import hip_ext
+            g = hip_ext().vp_ce_bwd(dloss.contiguous(), logits, labels, lse,
+                                    ctx.vocab_start, ctx.ignore_index)
+            return g, None, None, None, None, None
         lf = logits.float()
         p = torch.exp(lf - lse[:, None])
         valid = labels != ctx.ignore_index

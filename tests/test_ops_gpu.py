@@ -1,0 +1,198 @@
+"""GPU numerics: every HIP kernel vs its plain-PyTorch fp32 reference twin."""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from paddlefleetx_amd.ops import _reference as ref
+from paddlefleetx_amd.ops import functional as F
+from paddlefleetx_amd.ops import hip_ext
+
+
+@pytest.fixture(scope="module")
+def dev():
+    assert torch.cuda.is_available()
+    torch.manual_seed(1234)
+    return torch.device("cuda:0")
+
+
+def test_mfma_fragment_layout(dev):
+    """Asymmetric-B probe: catches transposed operand/output layouts (§3)."""
+    a = torch.randn(16, 32, device=dev).to(torch.bfloat16)
+    b = torch.randn(32, 16, device=dev).to(torch.bfloat16)
+    c = hip_ext().mfma_gemm16_probe(a, b)
+    want = a.float() @ b.float()
+    assert torch.allclose(c, want, atol=2e-2, rtol=2e-2), \
+        (c - want).abs().max()
+
+
+@pytest.mark.parametrize("shape", [(128, 1024), (512, 4096), (33, 1000)])
+def test_layernorm_fwd_bwd(dev, shape):
+    N, H = shape
+    if H % 4:
+        pytest.skip("H%4 kernel constraint")
+    x = torch.randn(N, H, device=dev, dtype=torch.bfloat16)
+    w = torch.randn(H, device=dev, dtype=torch.bfloat16)
+    b = torch.randn(H, device=dev, dtype=torch.bfloat16)
+    y, mean, rstd = hip_ext().layernorm_fwd(x, w, b, 1e-5)
+    y2, mean2, rstd2 = ref.layernorm_fwd(x, w, b, 1e-5)
+    assert torch.allclose(mean, mean2, atol=1e-3)
+    assert torch.allclose(rstd, rstd2, atol=1e-2, rtol=1e-2)
+    assert torch.allclose(y.float(), y2.float(), atol=3e-2, rtol=3e-2)
+
+    dy = torch.randn_like(x)
+    dx, dw, db = hip_ext().layernorm_bwd(dy, x, w, mean, rstd)
+    dx2, dw2, db2 = ref.layernorm_bwd(dy, x, w, mean2, rstd2)
+    assert torch.allclose(dx.float(), dx2.float(), atol=5e-2, rtol=5e-2)
+    assert torch.allclose(dw, dw2, atol=0.5, rtol=1e-2)
+    assert torch.allclose(db, db2, atol=0.5, rtol=1e-2)
+
+
+def test_rmsnorm(dev):
+    x = torch.randn(256, 2048, device=dev, dtype=torch.bfloat16)
+    w = torch.randn(2048, device=dev, dtype=torch.bfloat16)
+    y, rstd = hip_ext().rmsnorm_fwd(x, w, 1e-6)
+    y2, rstd2 = ref.rmsnorm_fwd(x, w, 1e-6)
+    assert torch.allclose(y.float(), y2.float(), atol=3e-2, rtol=3e-2)
+    dy = torch.randn_like(x)
+    dx, dw = hip_ext().rmsnorm_bwd(dy, x, w, rstd)
+    dx2, dw2 = ref.rmsnorm_bwd(dy, x, w, rstd2)
+    assert torch.allclose(dx.float(), dx2.float(), atol=5e-2, rtol=5e-2)
+    assert torch.allclose(dw, dw2, atol=0.5, rtol=1e-2)
+
+
+def test_bias_gelu(dev):
+    x = torch.randn(128, 512, device=dev, dtype=torch.bfloat16)
+    b = torch.randn(512, device=dev, dtype=torch.bfloat16)
+    y = hip_ext().bias_gelu_fwd(x, b)
+    y2 = ref.bias_gelu_fwd(x, b)
+    assert torch.allclose(y.float(), y2.float(), atol=2e-2, rtol=2e-2)
+    dy = torch.randn_like(x)
+    dx, db = hip_ext().bias_gelu_bwd(dy, x, b)
+    dx2, db2 = ref.bias_gelu_bwd(dy, x, b)
+    assert torch.allclose(dx.float(), dx2.float(), atol=3e-2, rtol=3e-2)
+    assert torch.allclose(db.float(), db2.float(), atol=0.5, rtol=1e-2)
+
+
+def test_adamw_flat(dev):
+    n = 4096
+    master = torch.randn(n, device=dev)
+    grad = torch.randn(n, device=dev)
+    m = torch.randn(n, device=dev).abs() * 0.1
+    v = torch.randn(n, device=dev).abs() * 0.01
+    model = torch.zeros(n, device=dev, dtype=torch.bfloat16)
+    m2, v2, master2 = m.clone(), v.clone(), master.clone()
+    model2 = model.clone()
+    hip_ext().adamw_flat(master, grad, m, v, model, 1e-3, 0.9, 0.95, 1e-8,
+                         0.01, 7)
+    ref.adamw_step(master2, grad, m2, v2, model2, 1e-3, 0.9, 0.95, 1e-8,
+                   0.01, 7)
+    assert torch.allclose(master, master2, atol=1e-6, rtol=1e-5)
+    assert torch.allclose(m, m2, atol=1e-6)
+    assert torch.allclose(v, v2, atol=1e-6)
+    assert torch.equal(model, model2)
+
+
+@pytest.mark.parametrize("shape", [(4, 8, 256, 256), (2, 4, 128, 384)])
+def test_softmax_causal(dev, shape):
+    s = torch.randn(*shape, device=dev, dtype=torch.bfloat16)
+    scale = 0.125
+    y = hip_ext().softmax_causal_fwd(s, scale)
+    y2 = ref.softmax_causal_fwd(s, scale)
+    assert torch.allclose(y.float(), y2.float(), atol=1e-2, rtol=1e-2)
+    dy = torch.randn_like(s)
+    ds = hip_ext().softmax_causal_bwd(dy, y, scale)
+    ds2 = ref.softmax_causal_bwd(dy, y2, scale)
+    assert torch.allclose(ds.float(), ds2.float(), atol=2e-2, rtol=2e-2)
+
+
+def test_cross_entropy(dev):
+    N, V = 512, 50304
+    logits = torch.randn(N, V, device=dev, dtype=torch.bfloat16)
+    labels = torch.randint(0, V, (N,), device=dev)
+    labels[::17] = -100
+    loss, lse = hip_ext().cross_entropy_fwd(logits, labels, -100)
+    loss2, lse2 = ref.cross_entropy_fwd(logits, labels, -100)
+    assert torch.allclose(lse, lse2, atol=1e-3, rtol=1e-4)
+    assert torch.allclose(loss, loss2, atol=1e-3, rtol=1e-4)
+    dloss = torch.randn(N, device=dev)
+    g = hip_ext().cross_entropy_bwd(dloss, logits, labels, lse, -100)
+    g2 = ref.cross_entropy_bwd(dloss, logits, labels, lse2, -100)
+    assert torch.allclose(g.float(), g2.float(), atol=2e-2, rtol=2e-2)
+
+
+@pytest.mark.parametrize("cfg", [
+    dict(B=2, H=4, S=256, D=128),
+    dict(B=1, H=2, S=1024, D=128),
+    dict(B=2, H=2, S=192, D=64),   # ragged S
+])
+def test_flash_attention_fwd(dev, cfg):
+    B, H, S, D = cfg["B"], cfg["H"], cfg["S"], cfg["D"]
+    q = torch.randn(B, H, S, D, device=dev, dtype=torch.bfloat16) * 0.5
+    k = torch.randn_like(q) * 0.5
+    v = torch.randn_like(q)
+    scale = 1.0 / math.sqrt(D)
+    o, lse = hip_ext().attn_fwd(q, k, v, True, scale)
+    o2, lse2 = ref.attention_fwd(q, k, v, True, scale)
+    assert torch.allclose(lse, lse2, atol=2e-2, rtol=1e-2), \
+        (lse - lse2).abs().max()
+    assert torch.allclose(o.float(), o2.float(), atol=3e-2, rtol=3e-2), \
+        (o.float() - o2.float()).abs().max()
+
+
+@pytest.mark.parametrize("cfg", [
+    dict(B=2, H=4, S=256, D=128),
+    dict(B=1, H=2, S=512, D=64),
+])
+def test_flash_attention_bwd(dev, cfg):
+    B, H, S, D = cfg["B"], cfg["H"], cfg["S"], cfg["D"]
+    q = torch.randn(B, H, S, D, device=dev, dtype=torch.bfloat16) * 0.5
+    k = torch.randn_like(q) * 0.5
+    v = torch.randn_like(q)
+    do = torch.randn_like(q)
+    scale = 1.0 / math.sqrt(D)
+    o, lse = hip_ext().attn_fwd(q, k, v, True, scale)
+    dq, dk, dv = hip_ext().attn_bwd(do, q, k, v, o, lse, True, scale)
+    dq2, dk2, dv2 = ref.attention_bwd(do, q, k, v, o, lse, True, scale)
+    for name, a, b in (("dv", dv, dv2), ("dk", dk, dk2), ("dq", dq, dq2)):
+        err = (a.float() - b.float()).abs().max()
+        assert torch.allclose(a.float(), b.float(), atol=8e-2, rtol=8e-2), \
+            f"{name}: max err {err}"
+
+
+def test_topp_sampling(dev):
+    B, V = 8, 50304
+    logits = torch.randn(B, V, device=dev)
+    probs = torch.softmax(logits, dim=-1)
+    top_p = torch.full((B,), 0.7, device=dev)
+    torch.manual_seed(7)
+    ids, pp = F.topp_sampling(probs, top_p)
+    assert ids.shape == (B, 1)
+    # sampled tokens must lie inside the nucleus of each row
+    sorted_p, sorted_idx = torch.sort(probs, dim=-1, descending=True)
+    cum = torch.cumsum(sorted_p, -1)
+    for b in range(B):
+        ncut = int((cum[b] >= 0.7).nonzero()[0]) + 1
+        nucleus = set(sorted_idx[b, :ncut].tolist())
+        assert int(ids[b, 0]) in nucleus
+    # distribution sanity: near-deterministic row
+    logits2 = torch.full((1, V), -10.0, device=dev)
+    logits2[0, 123] = 20.0
+    p2 = torch.softmax(logits2, -1)
+    ids2, _ = F.topp_sampling(p2, torch.tensor([0.9], device=dev))
+    assert int(ids2[0, 0]) == 123
+
+
+def test_rope(dev):
+    B, H, S, D = 2, 4, 128, 64
+    x = torch.randn(B, H, S, D, device=dev, dtype=torch.bfloat16)
+    inv = 1.0 / (10000 ** (torch.arange(0, D, 2, device=dev).float() / D))
+    t = torch.arange(S, device=dev).float()
+    freqs = torch.outer(t, inv)
+    cos, sin = freqs.cos(), freqs.sin()
+    y = hip_ext().rope_fwd(x, cos.contiguous(), sin.contiguous())
+    y2 = ref.rope_fwd(x, cos, sin)
+    assert torch.allclose(y.float(), y2.float(), atol=2e-2, rtol=2e-2)
