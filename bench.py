@@ -1,0 +1,151 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: GPT-3 6.7B pretrain step (BASELINE.json north star).
+
+Single node, one rank per GPU over RCCL:
+  python bench.py --gpus 1 --steps 10 --warmup 3
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N ...
+
+Parallelism by N (scaled-down versions of the 8-GPU DP2xTP2xPP2 target):
+  1 -> single | 2 -> TP2 | 4 -> TP2xPP2 | 8 -> DP2xTP2xPP2
+
+Rank 0 prints ONE JSON line with the whole-job tokens/s.
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import torch
+import torch.distributed as dist
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=None)
+    p.add_argument("--steps", type=int, default=10)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--model", type=str, default="GPT-6.7B",
+                   choices=["GPT-345M", "GPT-1.3B", "GPT-6.7B"])
+    p.add_argument("--micro-batch", type=int, default=None)
+    p.add_argument("--acc-steps", type=int, default=None)
+    p.add_argument("--seq-len", type=int, default=1024)
+    return p.parse_args()
+
+
+MODELS = {
+    "GPT-345M": dict(hidden_size=1024, num_layers=24, num_attention_heads=16),
+    "GPT-1.3B": dict(hidden_size=2048, num_layers=24, num_attention_heads=16),
+    "GPT-6.7B": dict(hidden_size=4096, num_layers=32, num_attention_heads=32),
+}
+
+
+def topology_for(n):
+    return {1: (1, 1, 1), 2: (1, 2, 1), 4: (1, 2, 2), 8: (2, 2, 2)}.get(
+        n, (n, 1, 1))  # fallback: pure DP
+
+
+def main():
+    args = parse_args()
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    n = args.gpus or world
+    assert world in (1, n), f"launched with WORLD_SIZE={world} but --gpus={n}"
+    dp, tp, pp = topology_for(n)
+
+    from paddlefleetx_amd.utils.config import get_config
+    from paddlefleetx_amd.parallel.env import init_dist_env, get_hcg
+    from paddlefleetx_amd.models import build_module
+    from paddlefleetx_amd.core import EagerEngine
+
+    micro = args.micro_batch or 2
+    acc = args.acc_steps or 8
+    local_bs = micro * acc
+    seq = args.seq_len
+    cfg_path = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                            "paddlefleetx_amd/configs/nlp/gpt/pretrain_gpt_base.yaml")
+    shape = MODELS[args.model]
+    overrides = [
+        f"Model.hidden_size={shape['hidden_size']}",
+        f"Model.num_layers={shape['num_layers']}",
+        f"Model.num_attention_heads={shape['num_attention_heads']}",
+        f"Model.max_position_embeddings={seq}",
+        "Model.hidden_dropout_prob=0.0",
+        "Model.attention_probs_dropout_prob=0.0",
+        f"Global.micro_batch_size={micro}",
+        f"Global.local_batch_size={local_bs}",
+        "Global.eval_freq=", "Global.save_steps=",
+        f"Distributed.dp_degree={dp}", f"Distributed.mp_degree={tp}",
+        f"Distributed.pp_degree={pp}",
+    ]
+    cfg = get_config(cfg_path, overrides=overrides)
+    hcg = init_dist_env(cfg)
+    rank = hcg.global_rank
+    device = torch.device("cuda") if torch.cuda.is_available() else torch.device("cpu")
+
+    module = build_module(cfg)
+    engine = EagerEngine(cfg, module)
+
+    vocab = cfg["Model"]["padded_vocab_size"]
+    torch.manual_seed(1234 + hcg.get_data_world_rank())
+    batch = (torch.randint(0, vocab, (local_bs, seq), device=device),
+             torch.arange(seq, device=device).unsqueeze(0).repeat(local_bs, 1),
+             torch.randint(0, vocab, (local_bs, seq), device=device),
+             torch.ones(local_bs, seq, device=device))
+
+    def barrier_sync():
+        if dist.is_initialized():
+            dist.barrier()
+        if device.type == "cuda":
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        engine._fit_impl(batch)
+    barrier_sync()
+    t0 = time.time()
+    loss = None
+    for _ in range(args.steps):
+        loss = engine._fit_impl(batch)
+    barrier_sync()
+    elapsed = time.time() - t0
+
+    # max over ranks
+    if dist.is_initialized():
+        t = torch.tensor(elapsed, device=device if device.type == "cuda" else None)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t)
+
+    global_batch = cfg["Global"]["global_batch_size"]
+    tokens_per_step = global_batch * seq
+    toks_per_s = tokens_per_step * args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank == 0:
+        out = {
+            "metric": "tokens/sec/GPU GPT-3-6.7B pretrain, DP2×TP2×PP2 at 1/2/4/8 MI355X",
+            "value": round(toks_per_s, 1),
+            "unit": "tokens/s",
+            "n_gpus": n,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": args.model, "global_batch": global_batch,
+                "seq_len": seq, "micro_batch": micro,
+                "parallelism": f"dp{dp}tp{tp}pp{pp}",
+                "loss": float(loss) if loss is not None else None,
+            },
+        }
+        print(json.dumps(out))
+
+
+if __name__ == "__main__":
+    main()
