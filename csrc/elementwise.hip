@@ -28,25 +28,38 @@ __global__ void bias_gelu_fwd_kernel(const T* __restrict__ x,
   const int VEC = 4;
   for (long i = ((long)blockIdx.x * BLOCK + threadIdx.x) * VEC; i < total;
        i += (long)gridDim.x * BLOCK * VEC) {
+    float v[VEC];
+    if constexpr (sizeof(T) == 2) {
+      short4v pk = *reinterpret_cast<const short4v*>((const unsigned short*)x + i);
 #pragma unroll
-    for (int j = 0; j < VEC; ++j) {
-      float v;
-      if constexpr (sizeof(T) == 2)
-        v = bf_raw2f(((const unsigned short*)x)[i + j]);
-      else
-        v = ((const float*)x)[i + j];
-      if constexpr (HAS_BIAS) {
-        int col = (int)((i + j) % H);
-        if constexpr (sizeof(T) == 2)
-          v += bf_raw2f(((const unsigned short*)bias)[col]);
-        else
-          v += ((const float*)bias)[col];
+      for (int j = 0; j < VEC; ++j) v[j] = bf_raw2f(((unsigned short*)&pk)[j]);
+    } else {
+      float4v pk = *reinterpret_cast<const float4v*>((const float*)x + i);
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) v[j] = ((float*)&pk)[j];
+    }
+    if constexpr (HAS_BIAS) {
+      int col = (int)(i % H);  // VEC divides H, so col..col+3 stay in row
+      if constexpr (sizeof(T) == 2) {
+        short4v bk = *reinterpret_cast<const short4v*>((const unsigned short*)bias + col);
+#pragma unroll
+        for (int j = 0; j < VEC; ++j) v[j] += bf_raw2f(((unsigned short*)&bk)[j]);
+      } else {
+        float4v bk = *reinterpret_cast<const float4v*>((const float*)bias + col);
+#pragma unroll
+        for (int j = 0; j < VEC; ++j) v[j] += ((float*)&bk)[j];
       }
-      float o = gelu_tanh(v);
-      if constexpr (sizeof(T) == 2)
-        ((unsigned short*)y)[i + j] = f2bf_raw(o);
-      else
-        ((float*)y)[i + j] = o;
+    }
+    if constexpr (sizeof(T) == 2) {
+      short4v out;
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) ((unsigned short*)&out)[j] = f2bf_raw(gelu_tanh(v[j]));
+      *reinterpret_cast<short4v*>((unsigned short*)y + i) = out;
+    } else {
+      float4v out;
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) ((float*)&out)[j] = gelu_tanh(v[j]);
+      *reinterpret_cast<float4v*>((float*)y + i) = out;
     }
   }
 }
@@ -59,36 +72,52 @@ __global__ void bias_gelu_bwd_kernel(const T* __restrict__ dy,
   const int VEC = 4;
   for (long i = ((long)blockIdx.x * BLOCK + threadIdx.x) * VEC; i < total;
        i += (long)gridDim.x * BLOCK * VEC) {
+    float v[VEC], d[VEC];
+    if constexpr (sizeof(T) == 2) {
+      short4v px = *reinterpret_cast<const short4v*>((const unsigned short*)x + i);
+      short4v pd = *reinterpret_cast<const short4v*>((const unsigned short*)dy + i);
 #pragma unroll
-    for (int j = 0; j < VEC; ++j) {
-      float v, d;
+      for (int j = 0; j < VEC; ++j) {
+        v[j] = bf_raw2f(((unsigned short*)&px)[j]);
+        d[j] = bf_raw2f(((unsigned short*)&pd)[j]);
+      }
+    } else {
+      float4v px = *reinterpret_cast<const float4v*>((const float*)x + i);
+      float4v pd = *reinterpret_cast<const float4v*>((const float*)dy + i);
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) { v[j] = ((float*)&px)[j]; d[j] = ((float*)&pd)[j]; }
+    }
+    if constexpr (HAS_BIAS) {
+      int col = (int)(i % H);
       if constexpr (sizeof(T) == 2) {
-        v = bf_raw2f(((const unsigned short*)x)[i + j]);
-        d = bf_raw2f(((const unsigned short*)dy)[i + j]);
+        short4v bk = *reinterpret_cast<const short4v*>((const unsigned short*)bias + col);
+#pragma unroll
+        for (int j = 0; j < VEC; ++j) v[j] += bf_raw2f(((unsigned short*)&bk)[j]);
       } else {
-        v = ((const float*)x)[i + j];
-        d = ((const float*)dy)[i + j];
+        float4v bk = *reinterpret_cast<const float4v*>((const float*)bias + col);
+#pragma unroll
+        for (int j = 0; j < VEC; ++j) v[j] += ((float*)&bk)[j];
       }
-      if constexpr (HAS_BIAS) {
-        int col = (int)((i + j) % H);
-        if constexpr (sizeof(T) == 2)
-          v += bf_raw2f(((const unsigned short*)bias)[col]);
-        else
-          v += ((const float*)bias)[col];
-      }
-      float o = d * gelu_tanh_grad(v);
-      if constexpr (sizeof(T) == 2)
-        ((unsigned short*)dx)[i + j] = f2bf_raw(o);
-      else
-        ((float*)dx)[i + j] = o;
+    }
+    if constexpr (sizeof(T) == 2) {
+      short4v out;
+#pragma unroll
+      for (int j = 0; j < VEC; ++j)
+        ((unsigned short*)&out)[j] = f2bf_raw(d[j] * gelu_tanh_grad(v[j]));
+      *reinterpret_cast<short4v*>((unsigned short*)dx + i) = out;
+    } else {
+      float4v out;
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) ((float*)&out)[j] = d[j] * gelu_tanh_grad(v[j]);
+      *reinterpret_cast<float4v*>((float*)dx + i) = out;
     }
   }
 }
 
-// ---- AdamW over flat fp32 buffers, writes back low-precision model copy ----
-template <typename TM>  // model dtype
+// ---- AdamW over flat fp32 state, low-precision model copy + grads ----
+template <typename TM, typename TG>  // model dtype, grad dtype
 __global__ void adamw_flat_kernel(float* __restrict__ master,
-                                  const float* __restrict__ grad,
+                                  const TG* __restrict__ grad,
                                   float* __restrict__ m, float* __restrict__ v,
                                   TM* __restrict__ model, long n, float lr,
                                   float beta1, float beta2, float eps,
@@ -97,7 +126,15 @@ __global__ void adamw_flat_kernel(float* __restrict__ master,
   for (long i = ((long)blockIdx.x * BLOCK + threadIdx.x) * VEC; i < n;
        i += (long)gridDim.x * BLOCK * VEC) {
     float4v pm = *reinterpret_cast<float4v*>(master + i);
-    float4v pg = *reinterpret_cast<const float4v*>(grad + i);
+    float4v pg;
+    if constexpr (sizeof(TG) == 2) {
+      short4v graw = *reinterpret_cast<const short4v*>((const unsigned short*)grad + i);
+#pragma unroll
+      for (int j = 0; j < VEC; ++j)
+        ((float*)&pg)[j] = bf_raw2f(((unsigned short*)&graw)[j]);
+    } else {
+      pg = *reinterpret_cast<const float4v*>((const float*)grad + i);
+    }
     float4v pmm = *reinterpret_cast<float4v*>(m + i);
     float4v pvv = *reinterpret_cast<float4v*>(v + i);
 #pragma unroll
@@ -236,21 +273,21 @@ void adamw_flat(torch::Tensor master, torch::Tensor grad, torch::Tensor m,
   float bc2 = 1.f - powf((float)beta2, (float)step);
   auto stream = at::hip::getCurrentHIPStream();
   dim3 grid(grid_for(n, BLOCK * 4));
-  if (has_model && model.scalar_type() == torch::kBFloat16) {
-    hipLaunchKernelGGL((adamw_flat_kernel<__hip_bfloat16>), grid, dim3(BLOCK),
-                       0, stream, master.data_ptr<float>(),
-                       grad.data_ptr<float>(), m.data_ptr<float>(),
-                       v.data_ptr<float>(), (__hip_bfloat16*)model.data_ptr(),
-                       n, (float)lr, (float)beta1, (float)beta2, (float)eps,
-                       (float)wd, bc1, bc2);
-  } else {
-    hipLaunchKernelGGL((adamw_flat_kernel<float>), grid, dim3(BLOCK), 0,
-                       stream, master.data_ptr<float>(), grad.data_ptr<float>(),
-                       m.data_ptr<float>(), v.data_ptr<float>(),
-                       has_model ? model.data_ptr<float>() : nullptr, n,
-                       (float)lr, (float)beta1, (float)beta2, (float)eps,
-                       (float)wd, bc1, bc2);
-  }
+  bool model_bf16 = has_model && model.scalar_type() == torch::kBFloat16;
+  bool grad_bf16 = grad.scalar_type() == torch::kBFloat16;
+#define LAUNCH_ADAMW(TM, TG, MP, GP)                                          \
+  hipLaunchKernelGGL((adamw_flat_kernel<TM, TG>), grid, dim3(BLOCK), 0,       \
+                     stream, master.data_ptr<float>(), (const TG*)(GP),       \
+                     m.data_ptr<float>(), v.data_ptr<float>(), (TM*)(MP), n,  \
+                     (float)lr, (float)beta1, (float)beta2, (float)eps,       \
+                     (float)wd, bc1, bc2)
+  void* mp = has_model ? model.data_ptr() : nullptr;
+  const void* gp = grad.data_ptr();
+  if (model_bf16 && grad_bf16) LAUNCH_ADAMW(__hip_bfloat16, __hip_bfloat16, mp, gp);
+  else if (model_bf16) LAUNCH_ADAMW(__hip_bfloat16, float, mp, gp);
+  else if (grad_bf16) LAUNCH_ADAMW(float, __hip_bfloat16, mp, gp);
+  else LAUNCH_ADAMW(float, float, mp, gp);
+#undef LAUNCH_ADAMW
 }
 
 torch::Tensor rope_fwd(torch::Tensor x, torch::Tensor cos, torch::Tensor sin) {

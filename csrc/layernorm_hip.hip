@@ -250,8 +250,10 @@ static std::vector<torch::Tensor> ln_bwd_impl(torch::Tensor dy, torch::Tensor x,
   auto stream = at::hip::getCurrentHIPStream();
   int grid = std::min(N, 2048);
   const float* mean_p = rms ? nullptr : mean_or_empty.data_ptr<float>();
-  int nchunks = std::min(64, (N + 255) / 256);
-  int rows_per_chunk = (N + nchunks - 1) / nchunks;
+  // enough (col-tile x row-chunk) blocks to fill 256 CUs; each thread owns
+  // one column over rows_per_chunk rows (coalesced across the 256 lanes)
+  int rows_per_chunk = 16;
+  int nchunks = (N + rows_per_chunk - 1) / rows_per_chunk;
   dim3 grid2((H + BLOCK - 1) / BLOCK, nchunks);
   if (x.scalar_type() == torch::kBFloat16) {
     using T = __hip_bfloat16;

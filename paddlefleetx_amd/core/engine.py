@@ -221,9 +221,7 @@ class EagerEngine(BasicEngine):
             elif dp.world_size > 1:
                 self.optimizer.reduce_gradients(dp, avg_factor=1.0)
             inv = 1.0 / (n_replicas * self.loss_scale)
-            if inv != 1.0:
-                for b in self.optimizer.buckets:
-                    b.main_grad.mul_(inv)
+            self.optimizer.scale_grads(inv)
             if self.loss_scale != 1.0:
                 self._found_inf = self._check_found_inf()
                 if self._found_inf:
@@ -264,11 +262,7 @@ class EagerEngine(BasicEngine):
                 self.optimizer.zero_grad(set_to_none=True)
 
     def _check_found_inf(self) -> float:
-        found = 0.0
-        for b in self.optimizer.buckets:
-            if not torch.isfinite(b.main_grad).all():
-                found = 1.0
-                break
+        found = 0.0 if self.optimizer.check_finite() else 1.0
         if dist.is_initialized():
             t = torch.tensor(found, device=self.device)
             dist.all_reduce(t, op=dist.ReduceOp.MAX)
@@ -281,7 +275,9 @@ class EagerEngine(BasicEngine):
             return
         for p in self.module.model.parameters():
             if getattr(p, "sequence_parallel", False):
-                buf = p.main_grad if hasattr(p, "main_grad") else p.grad
+                buf = getattr(p, "main_grad", None)
+                if buf is None:
+                    buf = p.grad
                 if buf is not None:
                     dist.all_reduce(buf, group=g.group)
 

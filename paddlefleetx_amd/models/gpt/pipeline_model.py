@@ -1,0 +1,90 @@
+"""GPT as a pipeline of LayerDescs (reference hybrid_model.py:999-1206:
+EmbeddingPipe :1012, LayerNormPipe :1026, GPTForPretrainingPipe :1055 with
+SharedLayerDesc tied embeddings :1116/1169, seg_method :1190-1192)."""
+
+from __future__ import annotations
+
+from typing import Any, Optional
+
+import torch
+import torch.nn as nn
+
+from paddlefleetx_amd.models.gpt.model import (GPTEmbeddings,
+                                               TransformerDecoderLayer)
+from paddlefleetx_amd.ops import FusedLayerNorm
+from paddlefleetx_amd.parallel.pp import (LayerDesc, PipelineModule,
+                                          SharedLayerDesc)
+from paddlefleetx_amd.parallel.tp import parallel_matmul
+
+
+class EmbeddingPipe(GPTEmbeddings):
+    """forward(tokens, position_ids) -> hidden."""
+
+    def forward(self, input_ids, position_ids=None):
+        return super().forward(input_ids, position_ids)
+
+
+class LayerNormPipe(FusedLayerNorm):
+    pass
+
+
+class TiedLogitsPipe(nn.Module):
+    """Last-stage copy of the word-embedding table used as the LM head;
+    weight synced with stage 0 by PipelineModule's shared-key machinery."""
+
+    def __init__(self, vocab_size: int, hidden_size: int,
+                 dtype: Optional[torch.dtype] = None, init_std: float = 0.02,
+                 **unused):
+        super().__init__()
+        from paddlefleetx_amd.parallel.tp import VocabParallelEmbedding
+        self.word_embeddings = VocabParallelEmbedding(
+            vocab_size, hidden_size, dtype=dtype, init_std=init_std)
+
+    def forward(self, x):
+        return parallel_matmul(x, self.word_embeddings.weight,
+                               parallel_output=True)
+
+
+class GPTForPretrainingPipe(PipelineModule):
+    def __init__(self, vocab_size: int = 50304, hidden_size: int = 1024,
+                 num_layers: int = 24, num_attention_heads: int = 16,
+                 ffn_hidden_size: Optional[int] = None,
+                 max_position_embeddings: int = 1024,
+                 hidden_dropout_prob: float = 0.0,
+                 attention_probs_dropout_prob: float = 0.0,
+                 fused_attn: bool = True, use_recompute: bool = False,
+                 recompute_granularity: str = "full",
+                 sequence_parallel: bool = False,
+                 initializer_range: float = 0.02,
+                 dtype: Optional[torch.dtype] = None, **unused: Any):
+        ffn_hidden_size = ffn_hidden_size or 4 * hidden_size
+        if sequence_parallel:
+            raise NotImplementedError(
+                "sequence_parallel with pipeline: round-2 item")
+        embed_kwargs = dict(vocab_size=vocab_size, hidden_size=hidden_size,
+                            max_position_embeddings=max_position_embeddings,
+                            dropout=hidden_dropout_prob, dtype=dtype,
+                            init_std=initializer_range)
+        descs = [
+            SharedLayerDesc("embed", EmbeddingPipe,
+                            shared_weight_attr="word_embeddings.weight",
+                            **embed_kwargs),
+        ]
+        for _ in range(num_layers):
+            descs.append(LayerDesc(
+                TransformerDecoderLayer, hidden_size, num_attention_heads,
+                ffn_hidden_size, hidden_dropout=hidden_dropout_prob,
+                attn_dropout=attention_probs_dropout_prob,
+                fused_attn=fused_attn, dtype=dtype,
+                init_std=initializer_range, num_layers_for_scale=num_layers,
+                use_recompute=use_recompute,
+                recompute_granularity=recompute_granularity))
+        descs.append(LayerDesc(LayerNormPipe, hidden_size, dtype=dtype))
+        descs.append(SharedLayerDesc("embed", TiedLogitsPipe,
+                                     shared_weight_attr="word_embeddings.weight",
+                                     vocab_size=vocab_size,
+                                     hidden_size=hidden_size, dtype=dtype,
+                                     init_std=initializer_range))
+        super().__init__(descs, seg_method="layer:TransformerDecoderLayer",
+                         act_dtype=dtype or torch.float32)
+        self.hidden_size = hidden_size
