@@ -10,8 +10,11 @@
 //     (parallel over q tiles) so neither needs atomics; both recompute
 //     P from (q, k, lse) — the standard flash backward decomposition.
 //
-// Layouts: q,k,v,o,do [B, H, S, D] bf16 contiguous; lse/delta [B, H, S] f32.
-// D in {64, 128}; any S (ragged tiles zero-padded in LDS).
+// All tensors are STRIDED: (batch_stride, head_stride, row_stride) in
+// elements with the head_dim axis contiguous. This lets the kernels read
+// the fused-QKV linear output [B, S, h, 3, D] and write attention output
+// [B, S, h*D] directly — no split/transpose/cat copies on either side.
+// lse/delta are [B, H, S] fp32 contiguous. D in {64, 128}; any S.
 #include "common.h"
 
 #include <torch/extension.h>
@@ -27,27 +30,40 @@ typedef __attribute__((ext_vector_type(4))) float f4;
 constexpr int TILE = 64;       // q-tile and kv-tile rows
 constexpr int NWAVES = 4;      // waves per block
 constexpr int BLOCKT = NWAVES * WAVE;
-// LDS row strides (elements): +8 bf16 (16 B) pad keeps ds_read_b128
-// 16-B-aligned and breaks the power-of-2 bank stride.
-constexpr int PAD = 8;
+constexpr int PAD = 8;         // +16 B LDS row pad: alignment + bank spread
+
+struct Strided {
+  const __hip_bfloat16* p;
+  long bs, hs, rs;  // batch/head/row strides (elements)
+  DEV_INLINE const __hip_bfloat16* at(int b, int h) const {
+    return p + (long)b * bs + (long)h * hs;
+  }
+};
+struct StridedMut {
+  __hip_bfloat16* p;
+  long bs, hs, rs;
+  DEV_INLINE __hip_bfloat16* at(int b, int h) const {
+    return p + (long)b * bs + (long)h * hs;
+  }
+};
 
 // ---------------------------------------------------------------------------
-// LDS staging helpers. Row-major [rows][D+PAD] and transposed [D][rows+PAD].
-// Zero-fills rows beyond `nvalid`.
+// LDS staging. Row-major [TILE][D+PAD] and transposed [D][TILE+PAD].
+// Rows beyond `nvalid` zero-filled. `g` points at row 0 of the tile.
 // ---------------------------------------------------------------------------
 template <int D>
-DEV_INLINE void stage_rowmajor(const __hip_bfloat16* __restrict__ g, int nvalid,
-                               unsigned short* lds /*[TILE][D+PAD]*/) {
+DEV_INLINE void stage_rowmajor(const __hip_bfloat16* __restrict__ g, long rs,
+                               int nvalid, unsigned short* lds) {
   constexpr int RS = D + PAD;
-  // each thread copies (TILE*D)/(BLOCKT*8) short8 vectors
-  constexpr int NV = TILE * D / 8;  // 8-elem vectors total
+  constexpr int NV = TILE * D / 8;
   for (int v = threadIdx.x; v < NV; v += BLOCKT) {
     int row = v / (D / 8);
     int col = (v % (D / 8)) * 8;
     short4v lo = {0, 0, 0, 0}, hi = {0, 0, 0, 0};
     if (row < nvalid) {
-      lo = *reinterpret_cast<const short4v*>(g + (long)row * D + col);
-      hi = *reinterpret_cast<const short4v*>(g + (long)row * D + col + 4);
+      const unsigned short* src = (const unsigned short*)g + (long)row * rs + col;
+      lo = *reinterpret_cast<const short4v*>(src);
+      hi = *reinterpret_cast<const short4v*>(src + 4);
     }
     *reinterpret_cast<short4v*>(lds + row * RS + col) = lo;
     *reinterpret_cast<short4v*>(lds + row * RS + col + 4) = hi;
@@ -55,49 +71,75 @@ DEV_INLINE void stage_rowmajor(const __hip_bfloat16* __restrict__ g, int nvalid,
 }
 
 template <int D>
-DEV_INLINE void stage_transposed(const __hip_bfloat16* __restrict__ g,
-                                 int nvalid,
-                                 unsigned short* lds /*[D][TILE+PAD]*/) {
+DEV_INLINE void stage_transposed(const __hip_bfloat16* __restrict__ g, long rs,
+                                 int nvalid, unsigned short* lds) {
   constexpr int RS = TILE + PAD;
   constexpr int NV = TILE * D / 8;
   for (int v = threadIdx.x; v < NV; v += BLOCKT) {
-    int row = v / (D / 8);        // source row (kv index)
-    int col = (v % (D / 8)) * 8;  // source col (d)
+    int row = v / (D / 8);
+    int col = (v % (D / 8)) * 8;
     unsigned short tmp[8] = {0, 0, 0, 0, 0, 0, 0, 0};
     if (row < nvalid) {
-      *reinterpret_cast<short4v*>(tmp) =
-          *reinterpret_cast<const short4v*>(g + (long)row * D + col);
+      const unsigned short* src = (const unsigned short*)g + (long)row * rs + col;
+      *reinterpret_cast<short4v*>(tmp) = *reinterpret_cast<const short4v*>(src);
       *reinterpret_cast<short4v*>(tmp + 4) =
-          *reinterpret_cast<const short4v*>(g + (long)row * D + col + 4);
+          *reinterpret_cast<const short4v*>(src + 4);
     }
 #pragma unroll
     for (int j = 0; j < 8; ++j) lds[(col + j) * RS + row] = tmp[j];
   }
 }
 
-// Load this wave's 16 rows of a [S, D] matrix into MFMA A-fragments.
-// A[row][k]: row = lane&15, k = (lane>>4)*8 + e  (k-chunk kc adds kc*32).
+// Stage both images in one global read.
 template <int D>
-DEV_INLINE void load_a_frags(const __hip_bfloat16* __restrict__ g, int row0,
-                             int nvalid, int lane, bf8* frags /*[D/32]*/) {
+DEV_INLINE void stage_both(const __hip_bfloat16* __restrict__ g, long rs,
+                           int nvalid, unsigned short* lds_rm,
+                           unsigned short* lds_tr) {
+  constexpr int RSM = D + PAD;
+  constexpr int RST = TILE + PAD;
+  constexpr int NV = TILE * D / 8;
+  for (int v = threadIdx.x; v < NV; v += BLOCKT) {
+    int row = v / (D / 8);
+    int col = (v % (D / 8)) * 8;
+    short4v lo = {0, 0, 0, 0}, hi = {0, 0, 0, 0};
+    if (row < nvalid) {
+      const unsigned short* src = (const unsigned short*)g + (long)row * rs + col;
+      lo = *reinterpret_cast<const short4v*>(src);
+      hi = *reinterpret_cast<const short4v*>(src + 4);
+    }
+    *reinterpret_cast<short4v*>(lds_rm + row * RSM + col) = lo;
+    *reinterpret_cast<short4v*>(lds_rm + row * RSM + col + 4) = hi;
+    unsigned short tmp[8];
+    *reinterpret_cast<short4v*>(tmp) = lo;
+    *reinterpret_cast<short4v*>(tmp + 4) = hi;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) lds_tr[(col + j) * RST + row] = tmp[j];
+  }
+}
+
+// Load this wave's 16 rows into MFMA A-fragments.
+// A[row][k]: row = lane&15, k = (lane>>4)*8 + e (+ kc*32).
+template <int D>
+DEV_INLINE void load_a_frags(const __hip_bfloat16* __restrict__ g, long rs,
+                             int row0, int nvalid, int lane, bf8* frags) {
   int row = row0 + (lane & 15);
   int d0 = (lane >> 4) * 8;
 #pragma unroll
   for (int kc = 0; kc < D / 32; ++kc) {
     bf8 f = {};
     if (row < nvalid) {
+      const unsigned short* src =
+          (const unsigned short*)g + (long)row * rs + kc * 32 + d0;
       union { short4v v[2]; bf8 b; } u;
-      u.v[0] = *reinterpret_cast<const short4v*>(g + (long)row * D + kc * 32 + d0);
-      u.v[1] = *reinterpret_cast<const short4v*>(g + (long)row * D + kc * 32 + d0 + 4);
+      u.v[0] = *reinterpret_cast<const short4v*>(src);
+      u.v[1] = *reinterpret_cast<const short4v*>(src + 4);
       f = u.b;
     }
     frags[kc] = f;
   }
 }
 
-// Read a B-fragment from a row-major LDS tile [R][C+PAD]:
-// B[k][col] with col = col0 + (lane&15) selecting LDS row, k consecutive 8
-// selecting LDS cols -> one 16-B read per chunk.
+// B-fragment from a row-major LDS tile: B[k][col], col selects LDS row.
 DEV_INLINE bf8 read_b_frag(const unsigned short* lds, int row_stride, int col0,
                            int k0, int lane) {
   const unsigned short* p =
@@ -110,7 +152,6 @@ DEV_INLINE bf8 read_b_frag(const unsigned short* lds, int row_stride, int col0,
 
 DEV_INLINE bf8 read_a_frag_lds(const unsigned short* lds, int row_stride,
                                int k0, int lane) {
-  // A[row][k]: row = lane&15 -> LDS row, k consecutive 8 -> LDS cols
   const unsigned short* p =
       lds + (lane & 15) * row_stride + k0 + (lane >> 4) * 8;
   union { short4v v[2]; bf8 b; } u;
@@ -124,32 +165,30 @@ DEV_INLINE bf8 read_a_frag_lds(const unsigned short* lds, int row_stride,
 // ===========================================================================
 template <int D>
 __global__ __launch_bounds__(BLOCKT) void attn_fwd_kernel(
-    const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
-    const __hip_bfloat16* __restrict__ v, __hip_bfloat16* __restrict__ o,
-    float* __restrict__ lse_out, int S, float scale, int q_tiles) {
-  constexpr int KRS = D + PAD;       // K row stride
-  constexpr int VRS = TILE + PAD;    // V^T row stride
-  constexpr int PRS = TILE + PAD;    // P row stride
-  constexpr int NDT = D / 16;        // output d-tiles per wave
+    Strided q, Strided k, Strided v, StridedMut o, float* __restrict__ lse_out,
+    int H, int S, float scale, int q_tiles) {
+  constexpr int KRS = D + PAD;
+  constexpr int VRS = TILE + PAD;
+  constexpr int PRS = TILE + PAD;
+  constexpr int NDT = D / 16;
 
   __shared__ unsigned short k_lds[TILE * KRS];
   __shared__ unsigned short vt_lds[D * VRS];
   __shared__ unsigned short p_lds[NWAVES * 16 * PRS];
 
   const int qt = blockIdx.x;
+  const int b = blockIdx.y / H, hh = blockIdx.y % H;
   const long bh = blockIdx.y;
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
 
-  const __hip_bfloat16* qp = q + bh * (long)S * D;
-  const __hip_bfloat16* kp = k + bh * (long)S * D;
-  const __hip_bfloat16* vp = v + bh * (long)S * D;
+  const __hip_bfloat16* qp = q.at(b, hh);
+  const __hip_bfloat16* kp = k.at(b, hh);
+  const __hip_bfloat16* vp = v.at(b, hh);
 
-  // this wave's 16 q rows
   const int qrow0 = qt * TILE + wid * 16;
   bf8 qfrag[D / 32];
-  load_a_frags<D>(qp, qrow0, S, lane, qfrag);
-  // scale folded into q? keep on scores (exactness of mask path)
+  load_a_frags<D>(qp, q.rs, qrow0, S, lane, qfrag);
 
   f4 oacc[NDT];
 #pragma unroll
@@ -158,10 +197,9 @@ __global__ __launch_bounds__(BLOCKT) void attn_fwd_kernel(
 #pragma unroll
   for (int r = 0; r < 4; ++r) { m_r[r] = -INFINITY; l_r[r] = 0.f; }
 
-  // C-layout coordinates of this lane
-  const int ccol = lane & 15;          // + ct*16 -> kv col
-  const int crow4 = (lane >> 4) * 4;   // + r -> local q row
-  const int my_qrow = qrow0 + crow4;   // + r
+  const int ccol = lane & 15;
+  const int crow4 = (lane >> 4) * 4;
+  const int my_qrow = qrow0 + crow4;
 
   const int kv_tiles = min(q_tiles, qt + 1);
   unsigned short* myp = p_lds + wid * 16 * PRS;
@@ -170,11 +208,11 @@ __global__ __launch_bounds__(BLOCKT) void attn_fwd_kernel(
     const int kv0 = kt * TILE;
     const int nvalid = min(TILE, S - kv0);
     __syncthreads();
-    stage_rowmajor<D>(kp + (long)kv0 * D, nvalid, k_lds);
-    stage_transposed<D>(vp + (long)kv0 * D, nvalid, vt_lds);
+    stage_rowmajor<D>(kp + (long)kv0 * k.rs, k.rs, nvalid, k_lds);
+    stage_transposed<D>(vp + (long)kv0 * v.rs, v.rs, nvalid, vt_lds);
     __syncthreads();
 
-    // ---- QK^T: 4 col-tiles x (D/32) k-chunks ----
+    // ---- QK^T ----
     f4 s[4];
 #pragma unroll
     for (int ct = 0; ct < 4; ++ct) {
@@ -207,7 +245,7 @@ __global__ __launch_bounds__(BLOCKT) void attn_fwd_kernel(
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       float nm = fmaxf(m_r[r], pmax[r]);
-      if (nm == -INFINITY) nm = 0.f;  // fully-masked row guard
+      if (nm == -INFINITY) nm = 0.f;
       alpha[r] = (m_r[r] == -INFINITY) ? 0.f : expf(m_r[r] - nm);
       m_r[r] = (m_r[r] == -INFINITY && pmax[r] == -INFINITY) ? -INFINITY : nm;
     }
@@ -227,14 +265,13 @@ __global__ __launch_bounds__(BLOCKT) void attn_fwd_kernel(
       psum[r] = group16_reduce_sum(psum[r]);
       l_r[r] = l_r[r] * alpha[r] + psum[r];
     }
-    // rescale O
 #pragma unroll
     for (int i = 0; i < NDT; ++i) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) oacc[i][r] *= alpha[r];
     }
 
-    // ---- P -> LDS (bf16) for A-fragments ----
+    // ---- P -> LDS for A-fragments ----
 #pragma unroll
     for (int ct = 0; ct < 4; ++ct) {
 #pragma unroll
@@ -242,11 +279,10 @@ __global__ __launch_bounds__(BLOCKT) void attn_fwd_kernel(
         myp[(crow4 + r) * PRS + ct * 16 + ccol] = f2bf_raw(s[ct][r]);
       }
     }
-    // wave-local LDS RAW: ordered by lgkmcnt the compiler inserts
 
-    // ---- PV: O[16][D] += P[16][64] V[64][D] ----
+    // ---- PV ----
 #pragma unroll
-    for (int kc = 0; kc < 2; ++kc) {  // 64 kv = 2 x K32
+    for (int kc = 0; kc < 2; ++kc) {
       bf8 pa = read_a_frag_lds(myp, PRS, kc * 32, lane);
 #pragma unroll
       for (int dt = 0; dt < NDT; ++dt) {
@@ -256,17 +292,17 @@ __global__ __launch_bounds__(BLOCKT) void attn_fwd_kernel(
     }
   }
 
-  // ---- epilogue ----
-  __hip_bfloat16* op = o + bh * (long)S * D;
+  // ---- epilogue (strided o) ----
+  __hip_bfloat16* op = o.at(b, hh);
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     int qrow = my_qrow + r;
     if (qrow >= S) continue;
     float inv = (l_r[r] > 0.f) ? 1.f / l_r[r] : 0.f;
+    unsigned short* orow = (unsigned short*)op + (long)qrow * o.rs;
 #pragma unroll
     for (int dt = 0; dt < NDT; ++dt) {
-      ((unsigned short*)op)[(long)qrow * D + dt * 16 + ccol] =
-          f2bf_raw(oacc[dt][r] * inv);
+      orow[dt * 16 + ccol] = f2bf_raw(oacc[dt][r] * inv);
     }
     if (ccol == 0 && lse_out) {
       float lv = (l_r[r] > 0.f) ? m_r[r] + logf(l_r[r]) : -INFINITY;
@@ -276,62 +312,68 @@ __global__ __launch_bounds__(BLOCKT) void attn_fwd_kernel(
 }
 
 // ===========================================================================
-// Backward: delta = rowsum(dO * O)
+// Backward: delta = rowsum(dO * O), strided inputs
 // ===========================================================================
-__global__ void attn_bwd_delta_kernel(const __hip_bfloat16* __restrict__ dout,
-                                      const __hip_bfloat16* __restrict__ o,
-                                      float* __restrict__ delta, long rows,
+__global__ void attn_bwd_delta_kernel(Strided dout, Strided o,
+                                      float* __restrict__ delta, int H, int S,
                                       int D) {
-  // one wave per row
   long row = (long)blockIdx.x * (BLOCKT / WAVE) + threadIdx.x / WAVE;
-  if (row >= rows) return;
+  long rows = (long)gridDim.x * (BLOCKT / WAVE);
+  long total = 0;  // computed by caller grid; bounds via S*H*B passed in grid
+  (void)rows; (void)total;
+  // decode (b, h, s)
+  long bhs = row;
+  int s = (int)(bhs % S);
+  long bh = bhs / S;
+  int hh = (int)(bh % H);
+  int b = (int)(bh / H);
+  if (s >= S) return;
   int lane = threadIdx.x % WAVE;
-  const unsigned short* dp = (const unsigned short*)dout + row * D;
-  const unsigned short* op = (const unsigned short*)o + row * D;
-  float s = 0.f;
+  const unsigned short* dp =
+      (const unsigned short*)dout.at(b, hh) + (long)s * dout.rs;
+  const unsigned short* op = (const unsigned short*)o.at(b, hh) + (long)s * o.rs;
+  float acc = 0.f;
   for (int i = lane; i < D; i += WAVE)
-    s += bf_raw2f(dp[i]) * bf_raw2f(op[i]);
-  s = wave_reduce_sum(s);
-  if (lane == 0) delta[row] = s;
+    acc += bf_raw2f(dp[i]) * bf_raw2f(op[i]);
+  acc = wave_reduce_sum(acc);
+  if (lane == 0) delta[bh * (long)S + s] = acc;
 }
 
 // ===========================================================================
-// Backward dK/dV: block = one kv tile; loop over q tiles >= diagonal.
-// Works in transposed score space: S^T[kv][q] = K Q^T.
+// Backward dK/dV (transposed score space S^T = K Q^T)
 // ===========================================================================
 template <int D>
 __global__ __launch_bounds__(BLOCKT) void attn_bwd_dkv_kernel(
-    const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
-    const __hip_bfloat16* __restrict__ v, const __hip_bfloat16* __restrict__ dout,
+    Strided q, Strided k, Strided v, Strided dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
-    __hip_bfloat16* __restrict__ dk, __hip_bfloat16* __restrict__ dv, int S,
-    float scale, int q_tiles) {
-  constexpr int RS = D + PAD;        // row-major stride
-  constexpr int TS = TILE + PAD;     // transposed stride
+    StridedMut dk, StridedMut dv, int H, int S, float scale, int q_tiles) {
+  constexpr int RS = D + PAD;
+  constexpr int TS = TILE + PAD;
   constexpr int NDT = D / 16;
 
-  __shared__ unsigned short q_lds[TILE * RS];    // Q row-major (B of S^T)
-  __shared__ unsigned short qt_lds[D * TS];      // Q^T (B of dK)
-  __shared__ unsigned short do_lds[TILE * RS];   // dO row-major (B of dP^T)
-  __shared__ unsigned short dot_lds[D * TS];     // dO^T (B of dV)
+  __shared__ unsigned short q_lds[TILE * RS];
+  __shared__ unsigned short qt_lds[D * TS];
+  __shared__ unsigned short do_lds[TILE * RS];
+  __shared__ unsigned short dot_lds[D * TS];
   __shared__ unsigned short p_lds[NWAVES * 16 * TS];
-  __shared__ float lsed_lds[2 * TILE];           // lse tile + delta tile
+  __shared__ float lsed_lds[2 * TILE];
 
   const int kt = blockIdx.x;
+  const int b = blockIdx.y / H, hh = blockIdx.y % H;
   const long bh = blockIdx.y;
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
 
-  const __hip_bfloat16* qp = q + bh * (long)S * D;
-  const __hip_bfloat16* kp = k + bh * (long)S * D;
-  const __hip_bfloat16* vp = v + bh * (long)S * D;
-  const __hip_bfloat16* dop = dout + bh * (long)S * D;
+  const __hip_bfloat16* qp = q.at(b, hh);
+  const __hip_bfloat16* kp = k.at(b, hh);
+  const __hip_bfloat16* vp = v.at(b, hh);
+  const __hip_bfloat16* dop = dout.at(b, hh);
 
   const int kv0 = kt * TILE;
   const int kvrow0 = kv0 + wid * 16;
   bf8 kfrag[D / 32], vfrag[D / 32];
-  load_a_frags<D>(kp, kvrow0, S, lane, kfrag);
-  load_a_frags<D>(vp, kvrow0, S, lane, vfrag);
+  load_a_frags<D>(kp, k.rs, kvrow0, S, lane, kfrag);
+  load_a_frags<D>(vp, v.rs, kvrow0, S, lane, vfrag);
 
   f4 dvacc[NDT], dkacc[NDT];
 #pragma unroll
@@ -340,8 +382,8 @@ __global__ __launch_bounds__(BLOCKT) void attn_bwd_dkv_kernel(
     dkacc[i] = f4{0.f, 0.f, 0.f, 0.f};
   }
 
-  const int ccol = lane & 15;         // + ct*16 -> q col (in S^T space)
-  const int crow4 = (lane >> 4) * 4;  // + r -> local kv row
+  const int ccol = lane & 15;
+  const int crow4 = (lane >> 4) * 4;
   const int my_kvrow = kvrow0 + crow4;
   unsigned short* myp = p_lds + wid * 16 * TS;
 
@@ -349,10 +391,8 @@ __global__ __launch_bounds__(BLOCKT) void attn_bwd_dkv_kernel(
     const int q0 = qt * TILE;
     const int nvalid = min(TILE, S - q0);
     __syncthreads();
-    stage_rowmajor<D>(qp + (long)q0 * D, nvalid, q_lds);
-    stage_transposed<D>(qp + (long)q0 * D, nvalid, qt_lds);
-    stage_rowmajor<D>(dop + (long)q0 * D, nvalid, do_lds);
-    stage_transposed<D>(dop + (long)q0 * D, nvalid, dot_lds);
+    stage_both<D>(qp + (long)q0 * q.rs, q.rs, nvalid, q_lds, qt_lds);
+    stage_both<D>(dop + (long)q0 * dout.rs, dout.rs, nvalid, do_lds, dot_lds);
     for (int i = threadIdx.x; i < TILE; i += BLOCKT) {
       int qi = q0 + i;
       lsed_lds[i] = (qi < S) ? lse[bh * (long)S + qi] : 0.f;
@@ -360,8 +400,8 @@ __global__ __launch_bounds__(BLOCKT) void attn_bwd_dkv_kernel(
     }
     __syncthreads();
 
-    // ---- S^T = K Q^T (then P^T) ----
-    f4 st[4];
+    // ---- S^T = K Q^T; P^T = exp(scale*S^T - lse[q]) ----
+    f4 pt[4];
 #pragma unroll
     for (int ct = 0; ct < 4; ++ct) {
       f4 acc = {0.f, 0.f, 0.f, 0.f};
@@ -370,12 +410,6 @@ __global__ __launch_bounds__(BLOCKT) void attn_bwd_dkv_kernel(
         bf8 qb = read_b_frag(q_lds, RS, ct * 16, kc * 32, lane);
         acc = MFMA_BF16(kfrag[kc], qb, acc);
       }
-      st[ct] = acc;
-    }
-    // P^T = exp(scale*S^T - lse[q]); causal: q >= kv
-    f4 pt[4];
-#pragma unroll
-    for (int ct = 0; ct < 4; ++ct) {
       int qcol = q0 + ct * 16 + ccol;
       float l = lsed_lds[ct * 16 + ccol];
 #pragma unroll
@@ -383,12 +417,12 @@ __global__ __launch_bounds__(BLOCKT) void attn_bwd_dkv_kernel(
         int kvr = my_kvrow + r;
         float p = 0.f;
         if (qcol >= kvr && qcol < S && kvr < S)
-          p = expf(st[ct][r] * scale - l);
+          p = expf(acc[r] * scale - l);
         pt[ct][r] = p;
       }
     }
 
-    // ---- stage P^T -> LDS, dV += P^T dO ----
+    // ---- stage P^T; dV += P^T dO ----
 #pragma unroll
     for (int ct = 0; ct < 4; ++ct)
 #pragma unroll
@@ -417,8 +451,8 @@ __global__ __launch_bounds__(BLOCKT) void attn_bwd_dkv_kernel(
       dpt[ct] = acc;
     }
 
-    // ---- dS^T = P^T (dP^T - delta[q]) * scale -> LDS, dK += dS^T Q ----
-    __syncthreads();  // everyone done reading p_lds as P^T
+    // ---- dS^T -> LDS; dK += dS^T Q ----
+    __syncthreads();
 #pragma unroll
     for (int ct = 0; ct < 4; ++ct) {
       float dlt = lsed_lds[TILE + ct * 16 + ccol];
@@ -439,55 +473,54 @@ __global__ __launch_bounds__(BLOCKT) void attn_bwd_dkv_kernel(
     }
   }
 
-  // ---- epilogue: write dK, dV (bf16) ----
-  __hip_bfloat16* dkp = dk + bh * (long)S * D;
-  __hip_bfloat16* dvp = dv + bh * (long)S * D;
+  __hip_bfloat16* dkp = dk.at(b, hh);
+  __hip_bfloat16* dvp = dv.at(b, hh);
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     int kvr = my_kvrow + r;
     if (kvr >= S) continue;
+    unsigned short* krow = (unsigned short*)dkp + (long)kvr * dk.rs;
+    unsigned short* vrow = (unsigned short*)dvp + (long)kvr * dv.rs;
 #pragma unroll
     for (int dt = 0; dt < NDT; ++dt) {
-      ((unsigned short*)dkp)[(long)kvr * D + dt * 16 + ccol] =
-          f2bf_raw(dkacc[dt][r]);
-      ((unsigned short*)dvp)[(long)kvr * D + dt * 16 + ccol] =
-          f2bf_raw(dvacc[dt][r]);
+      krow[dt * 16 + ccol] = f2bf_raw(dkacc[dt][r]);
+      vrow[dt * 16 + ccol] = f2bf_raw(dvacc[dt][r]);
     }
   }
 }
 
 // ===========================================================================
-// Backward dQ: block = one q tile; loop kv tiles <= diagonal.
+// Backward dQ
 // ===========================================================================
 template <int D>
 __global__ __launch_bounds__(BLOCKT) void attn_bwd_dq_kernel(
-    const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
-    const __hip_bfloat16* __restrict__ v, const __hip_bfloat16* __restrict__ dout,
+    Strided q, Strided k, Strided v, Strided dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
-    __hip_bfloat16* __restrict__ dq, int S, float scale, int q_tiles) {
+    StridedMut dq, int H, int S, float scale, int q_tiles) {
   constexpr int RS = D + PAD;
   constexpr int TS = TILE + PAD;
   constexpr int NDT = D / 16;
 
-  __shared__ unsigned short k_lds[TILE * RS];   // K row-major (B of S)
-  __shared__ unsigned short kt_lds[D * TS];     // K^T (B of dQ)
-  __shared__ unsigned short v_lds[TILE * RS];   // V row-major (B of dP)
+  __shared__ unsigned short k_lds[TILE * RS];
+  __shared__ unsigned short kt_lds[D * TS];
+  __shared__ unsigned short v_lds[TILE * RS];
   __shared__ unsigned short p_lds[NWAVES * 16 * TS];
 
   const int qt = blockIdx.x;
+  const int b = blockIdx.y / H, hh = blockIdx.y % H;
   const long bh = blockIdx.y;
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
 
-  const __hip_bfloat16* qp = q + bh * (long)S * D;
-  const __hip_bfloat16* kp = k + bh * (long)S * D;
-  const __hip_bfloat16* vp = v + bh * (long)S * D;
-  const __hip_bfloat16* dop = dout + bh * (long)S * D;
+  const __hip_bfloat16* qp = q.at(b, hh);
+  const __hip_bfloat16* kp = k.at(b, hh);
+  const __hip_bfloat16* vp = v.at(b, hh);
+  const __hip_bfloat16* dop = dout.at(b, hh);
 
   const int qrow0 = qt * TILE + wid * 16;
   bf8 qfrag[D / 32], dofrag[D / 32];
-  load_a_frags<D>(qp, qrow0, S, lane, qfrag);
-  load_a_frags<D>(dop, qrow0, S, lane, dofrag);
+  load_a_frags<D>(qp, q.rs, qrow0, S, lane, qfrag);
+  load_a_frags<D>(dop, dout.rs, qrow0, S, lane, dofrag);
 
   f4 dqacc[NDT];
 #pragma unroll
@@ -498,7 +531,6 @@ __global__ __launch_bounds__(BLOCKT) void attn_bwd_dq_kernel(
   const int my_qrow = qrow0 + crow4;
   unsigned short* myp = p_lds + wid * 16 * TS;
 
-  // per-row lse/delta (C layout rows)
   float lse_r[4], dlt_r[4];
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
@@ -512,12 +544,11 @@ __global__ __launch_bounds__(BLOCKT) void attn_bwd_dq_kernel(
     const int kv0 = ktl * TILE;
     const int nvalid = min(TILE, S - kv0);
     __syncthreads();
-    stage_rowmajor<D>(kp + (long)kv0 * D, nvalid, k_lds);
-    stage_transposed<D>(kp + (long)kv0 * D, nvalid, kt_lds);
-    stage_rowmajor<D>(vp + (long)kv0 * D, nvalid, v_lds);
+    stage_both<D>(kp + (long)kv0 * k.rs, k.rs, nvalid, k_lds, kt_lds);
+    stage_rowmajor<D>(vp + (long)kv0 * v.rs, v.rs, nvalid, v_lds);
     __syncthreads();
 
-    // ---- S = Q K^T, P = exp(scale*S - lse) ----
+    // ---- S = Q K^T; P ----
     f4 p[4];
 #pragma unroll
     for (int ct = 0; ct < 4; ++ct) {
@@ -550,7 +581,7 @@ __global__ __launch_bounds__(BLOCKT) void attn_bwd_dq_kernel(
       dp[ct] = acc;
     }
 
-    // ---- dS = P (dP - delta) scale -> LDS; dQ += dS K ----
+    // ---- dS -> LDS; dQ += dS K ----
 #pragma unroll
     for (int ct = 0; ct < 4; ++ct)
 #pragma unroll
@@ -569,16 +600,79 @@ __global__ __launch_bounds__(BLOCKT) void attn_bwd_dq_kernel(
     }
   }
 
-  __hip_bfloat16* dqp = dq + bh * (long)S * D;
+  __hip_bfloat16* dqp = dq.at(b, hh);
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     int qr = my_qrow + r;
     if (qr >= S) continue;
+    unsigned short* qrow_p = (unsigned short*)dqp + (long)qr * dq.rs;
 #pragma unroll
     for (int dt = 0; dt < NDT; ++dt) {
-      ((unsigned short*)dqp)[(long)qr * D + dt * 16 + ccol] =
-          f2bf_raw(dqacc[dt][r]);
+      qrow_p[dt * 16 + ccol] = f2bf_raw(dqacc[dt][r]);
     }
+  }
+}
+
+Strided strided_of(const torch::Tensor& t, int b_dim, int h_dim, int s_dim) {
+  Strided s;
+  s.p = (const __hip_bfloat16*)t.data_ptr();
+  s.bs = t.stride(b_dim);
+  s.hs = t.stride(h_dim);
+  s.rs = t.stride(s_dim);
+  return s;
+}
+StridedMut strided_mut_of(torch::Tensor& t, int b_dim, int h_dim, int s_dim) {
+  StridedMut s;
+  s.p = (__hip_bfloat16*)t.data_ptr();
+  s.bs = t.stride(b_dim);
+  s.hs = t.stride(h_dim);
+  s.rs = t.stride(s_dim);
+  return s;
+}
+
+void check_attn_tensor(const torch::Tensor& t, int d_dim, int D) {
+  TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(t.size(d_dim) == D && t.stride(d_dim) == 1,
+              "head_dim axis must be contiguous");
+}
+
+void launch_fwd(Strided q, Strided k, Strided v, StridedMut o, float* lse,
+                int B, int H, int S, int D, float scale) {
+  int q_tiles = (S + TILE - 1) / TILE;
+  dim3 grid(q_tiles, B * H);
+  auto stream = at::hip::getCurrentHIPStream();
+  if (D == 128)
+    hipLaunchKernelGGL((attn_fwd_kernel<128>), grid, dim3(BLOCKT), 0, stream,
+                       q, k, v, o, lse, H, S, scale, q_tiles);
+  else
+    hipLaunchKernelGGL((attn_fwd_kernel<64>), grid, dim3(BLOCKT), 0, stream,
+                       q, k, v, o, lse, H, S, scale, q_tiles);
+}
+
+void launch_bwd(Strided q, Strided k, Strided v, Strided dout, Strided o,
+                StridedMut dq, StridedMut dk, StridedMut dv, float* lse,
+                float* delta, int B, int H, int S, int D, float scale) {
+  auto stream = at::hip::getCurrentHIPStream();
+  long rows = (long)B * H * S;
+  hipLaunchKernelGGL(attn_bwd_delta_kernel,
+                     dim3((rows + (BLOCKT / WAVE) - 1) / (BLOCKT / WAVE)),
+                     dim3(BLOCKT), 0, stream, dout, o, delta, H, S, D);
+  int q_tiles = (S + TILE - 1) / TILE;
+  dim3 grid(q_tiles, B * H);
+  if (D == 128) {
+    hipLaunchKernelGGL((attn_bwd_dkv_kernel<128>), grid, dim3(BLOCKT), 0,
+                       stream, q, k, v, dout, lse, delta, dk, dv, H, S, scale,
+                       q_tiles);
+    hipLaunchKernelGGL((attn_bwd_dq_kernel<128>), grid, dim3(BLOCKT), 0,
+                       stream, q, k, v, dout, lse, delta, dq, H, S, scale,
+                       q_tiles);
+  } else {
+    hipLaunchKernelGGL((attn_bwd_dkv_kernel<64>), grid, dim3(BLOCKT), 0,
+                       stream, q, k, v, dout, lse, delta, dk, dv, H, S, scale,
+                       q_tiles);
+    hipLaunchKernelGGL((attn_bwd_dq_kernel<64>), grid, dim3(BLOCKT), 0,
+                       stream, q, k, v, dout, lse, delta, dq, H, S, scale,
+                       q_tiles);
   }
 }
 
@@ -588,35 +682,21 @@ __global__ __launch_bounds__(BLOCKT) void attn_bwd_dq_kernel(
 // Host bindings
 // ===========================================================================
 
+// q,k,v: [B, H, S, D] contiguous (layout-compat path)
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, bool causal,
                                     double scale) {
-  TORCH_CHECK(q.is_cuda() && q.dim() == 4 && q.is_contiguous());
-  TORCH_CHECK(q.scalar_type() == torch::kBFloat16,
-              "attn_fwd: bf16 only (got ", q.scalar_type(), ")");
+  TORCH_CHECK(q.dim() == 4 && q.is_contiguous());
   TORCH_CHECK(causal, "attn_fwd: only causal attention implemented");
   int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
-  TORCH_CHECK(k.size(2) == S, "attn_fwd: q and k seq length must match");
-  TORCH_CHECK(D == 64 || D == 128, "attn_fwd: head_dim must be 64 or 128");
+  TORCH_CHECK(k.size(2) == S, "q and k seq length must match");
+  TORCH_CHECK(D == 64 || D == 128, "head_dim must be 64 or 128");
+  check_attn_tensor(q, 3, D);
   auto o = torch::empty_like(q);
   auto lse = torch::empty({B, H, S}, q.options().dtype(torch::kFloat));
-  int q_tiles = (S + TILE - 1) / TILE;
-  dim3 grid(q_tiles, B * H);
-  auto stream = at::hip::getCurrentHIPStream();
-  if (D == 128)
-    hipLaunchKernelGGL((attn_fwd_kernel<128>), grid, dim3(BLOCKT), 0, stream,
-                       (const __hip_bfloat16*)q.data_ptr(),
-                       (const __hip_bfloat16*)k.data_ptr(),
-                       (const __hip_bfloat16*)v.data_ptr(),
-                       (__hip_bfloat16*)o.data_ptr(), lse.data_ptr<float>(), S,
-                       (float)scale, q_tiles);
-  else
-    hipLaunchKernelGGL((attn_fwd_kernel<64>), grid, dim3(BLOCKT), 0, stream,
-                       (const __hip_bfloat16*)q.data_ptr(),
-                       (const __hip_bfloat16*)k.data_ptr(),
-                       (const __hip_bfloat16*)v.data_ptr(),
-                       (__hip_bfloat16*)o.data_ptr(), lse.data_ptr<float>(), S,
-                       (float)scale, q_tiles);
+  launch_fwd(strided_of(q, 0, 1, 2), strided_of(k, 0, 1, 2),
+             strided_of(v, 0, 1, 2), strided_mut_of(o, 0, 1, 2),
+             lse.data_ptr<float>(), B, H, S, D, (float)scale);
   return {o, lse};
 }
 
@@ -624,44 +704,64 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     torch::Tensor k, torch::Tensor v,
                                     torch::Tensor o, torch::Tensor lse,
                                     bool causal, double scale) {
-  TORCH_CHECK(causal, "attn_bwd: only causal attention implemented");
-  TORCH_CHECK(q.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(causal);
   int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
   auto dq = torch::empty_like(q);
   auto dk = torch::empty_like(k);
   auto dv = torch::empty_like(v);
   auto delta = torch::empty({B, H, S}, q.options().dtype(torch::kFloat));
-  long rows = (long)B * H * S;
-  auto stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(attn_bwd_delta_kernel,
-                     dim3((rows + NWAVES - 1) / NWAVES), dim3(BLOCKT), 0,
-                     stream, (const __hip_bfloat16*)dout.data_ptr(),
-                     (const __hip_bfloat16*)o.data_ptr(),
-                     delta.data_ptr<float>(), rows, D);
-  int q_tiles = (S + TILE - 1) / TILE;
-  dim3 grid(q_tiles, B * H);
-#define LAUNCH_BWD(DV)                                                        \
-  do {                                                                        \
-    hipLaunchKernelGGL((attn_bwd_dkv_kernel<DV>), grid, dim3(BLOCKT), 0,      \
-                       stream, (const __hip_bfloat16*)q.data_ptr(),           \
-                       (const __hip_bfloat16*)k.data_ptr(),                   \
-                       (const __hip_bfloat16*)v.data_ptr(),                   \
-                       (const __hip_bfloat16*)dout.data_ptr(),                \
-                       lse.data_ptr<float>(), delta.data_ptr<float>(),        \
-                       (__hip_bfloat16*)dk.data_ptr(),                        \
-                       (__hip_bfloat16*)dv.data_ptr(), S, (float)scale,       \
-                       q_tiles);                                              \
-    hipLaunchKernelGGL((attn_bwd_dq_kernel<DV>), grid, dim3(BLOCKT), 0,       \
-                       stream, (const __hip_bfloat16*)q.data_ptr(),           \
-                       (const __hip_bfloat16*)k.data_ptr(),                   \
-                       (const __hip_bfloat16*)v.data_ptr(),                   \
-                       (const __hip_bfloat16*)dout.data_ptr(),                \
-                       lse.data_ptr<float>(), delta.data_ptr<float>(),        \
-                       (__hip_bfloat16*)dq.data_ptr(), S, (float)scale,       \
-                       q_tiles);                                              \
-  } while (0)
-  if (D == 128) LAUNCH_BWD(128);
-  else LAUNCH_BWD(64);
-#undef LAUNCH_BWD
+  auto dc = dout.contiguous();
+  launch_bwd(strided_of(q, 0, 1, 2), strided_of(k, 0, 1, 2),
+             strided_of(v, 0, 1, 2), strided_of(dc, 0, 1, 2),
+             strided_of(o, 0, 1, 2), strided_mut_of(dq, 0, 1, 2),
+             strided_mut_of(dk, 0, 1, 2), strided_mut_of(dv, 0, 1, 2),
+             lse.data_ptr<float>(), delta.data_ptr<float>(), B, H, S, D,
+             (float)scale);
   return {dq, dk, dv};
+}
+
+// Packed path: qkv [B, S, Hh, 3, D] (the fused-QKV linear output viewed);
+// returns o [B, S, Hh*D] + lse [B, Hh, S]. Zero layout copies.
+std::vector<torch::Tensor> attn_fwd_packed(torch::Tensor qkv, long num_heads,
+                                           double scale) {
+  TORCH_CHECK(qkv.dim() == 5 && qkv.is_contiguous());
+  int B = qkv.size(0), S = qkv.size(1), H = qkv.size(2), D = qkv.size(4);
+  TORCH_CHECK(qkv.size(3) == 3 && H == num_heads);
+  TORCH_CHECK(D == 64 || D == 128);
+  check_attn_tensor(qkv, 4, D);
+  auto o = torch::empty({B, S, (long)H * D}, qkv.options());
+  auto lse = torch::empty({B, H, S}, qkv.options().dtype(torch::kFloat));
+  Strided q{(const __hip_bfloat16*)qkv.data_ptr(), qkv.stride(0),
+            qkv.stride(2), qkv.stride(1)};
+  Strided k = q; k.p += D;
+  Strided v = q; v.p += 2 * D;
+  StridedMut om{(__hip_bfloat16*)o.data_ptr(), (long)S * H * D, (long)D,
+                (long)H * D};
+  launch_fwd(q, k, v, om, lse.data_ptr<float>(), B, H, S, D, (float)scale);
+  return {o, lse};
+}
+
+// dout [B, S, Hh*D]; returns dqkv [B, S, Hh, 3, D]
+torch::Tensor attn_bwd_packed(torch::Tensor dout, torch::Tensor qkv,
+                              torch::Tensor o, torch::Tensor lse,
+                              long num_heads, double scale) {
+  int B = qkv.size(0), S = qkv.size(1), H = qkv.size(2), D = qkv.size(4);
+  auto dqkv = torch::empty_like(qkv);
+  auto delta = torch::empty({B, H, S}, qkv.options().dtype(torch::kFloat));
+  auto dc = dout.contiguous();
+  Strided q{(const __hip_bfloat16*)qkv.data_ptr(), qkv.stride(0),
+            qkv.stride(2), qkv.stride(1)};
+  Strided k = q; k.p += D;
+  Strided v = q; v.p += 2 * D;
+  Strided dos{(const __hip_bfloat16*)dc.data_ptr(), (long)S * H * D, (long)D,
+              (long)H * D};
+  Strided os{(const __hip_bfloat16*)o.data_ptr(), (long)S * H * D, (long)D,
+             (long)H * D};
+  StridedMut dqm{(__hip_bfloat16*)dqkv.data_ptr(), dqkv.stride(0),
+                 dqkv.stride(2), dqkv.stride(1)};
+  StridedMut dkm = dqm; dkm.p += D;
+  StridedMut dvm = dqm; dvm.p += 2 * D;
+  launch_bwd(q, k, v, dos, os, dqm, dkm, dvm, lse.data_ptr<float>(),
+             delta.data_ptr<float>(), B, H, S, D, (float)scale);
+  return dqkv;
 }

@@ -46,6 +46,11 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     torch::Tensor k, torch::Tensor v,
                                     torch::Tensor o, torch::Tensor lse,
                                     bool causal, double scale);
+std::vector<torch::Tensor> attn_fwd_packed(torch::Tensor qkv, long num_heads,
+                                           double scale);
+torch::Tensor attn_bwd_packed(torch::Tensor dout, torch::Tensor qkv,
+                              torch::Tensor o, torch::Tensor lse,
+                              long num_heads, double scale);
 // topp.hip
 std::vector<torch::Tensor> topp_select(torch::Tensor sorted_p,
                                        torch::Tensor sorted_idx,
@@ -73,6 +78,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("vp_ce_bwd", &vp_ce_bwd, "vocab-parallel CE bwd");
   m.def("attn_fwd", &attn_fwd, "flash attention fwd (MFMA, causal)");
   m.def("attn_bwd", &attn_bwd, "flash attention bwd (MFMA)");
+  m.def("attn_fwd_packed", &attn_fwd_packed,
+        "flash attention fwd on packed QKV (no layout copies)");
+  m.def("attn_bwd_packed", &attn_bwd_packed,
+        "flash attention bwd writing packed dQKV");
   m.def("topp_select", &topp_select, "top-p nucleus cutoff + draw");
   m.def("mfma_gemm16_probe", &mfma_gemm16_probe,
         "debug: 16x16x32 MFMA fragment-layout probe");

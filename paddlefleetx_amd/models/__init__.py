@@ -27,4 +27,10 @@ def build_module(config):
     if name not in table:
         raise ValueError(f"unknown module {name}")
     logger.info(f"building module {name}")
+    import torch
+    if torch.cuda.is_available():
+        # construct parameters directly on the GPU: CPU-side init of a 6.7B
+        # model costs ~100 s; on-device init is seconds
+        with torch.device("cuda"):
+            return table[name](config)
     return table[name](config)

@@ -23,6 +23,7 @@ import torch.nn.functional as F
 from torch.utils.checkpoint import checkpoint
 
 from paddlefleetx_amd.ops import (FusedLayerNorm, bias_gelu, flash_attention,
+                                  flash_attention_packed,
                                   fused_softmax_causal)
 from paddlefleetx_amd.parallel.env import get_hcg
 from paddlefleetx_amd.parallel.rng import model_parallel_rng
@@ -85,6 +86,15 @@ class MultiHeadAttention(nn.Module):
                 use_cache: bool = False
                 ) -> Tuple[torch.Tensor, Optional[KVCache]]:
         qkv = self.qkv(x)
+        if (self.fused_attn and self.attn_dropout_p == 0.0 and cache is None
+                and not use_cache and not self.sequence_parallel):
+            # packed fast path: kernel reads the fused-QKV linear output
+            # directly and writes [B, S, h*D] — zero layout copies
+            B, S, _ = qkv.shape
+            packed = qkv.view(B, S, self.num_heads_local, 3, self.head_dim)
+            o = flash_attention_packed(packed, self.num_heads_local,
+                                       scale=self.scale)
+            return self.out_proj(o), None
         if self.sequence_parallel:
             # x: [s/mp, B, H]; qkv allgathered the seq dim -> [S, B, 3H/mp]
             S, B = qkv.shape[0], qkv.shape[1]

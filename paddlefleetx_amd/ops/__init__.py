@@ -56,6 +56,7 @@ from paddlefleetx_amd.ops.functional import (  # noqa: E402,F401
     bias_gelu,
     cross_entropy,
     flash_attention,
+    flash_attention_packed,
     fused_adamw_flat,
     fused_softmax_causal,
     layernorm,

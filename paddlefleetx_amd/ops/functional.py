@@ -22,7 +22,7 @@ from paddlefleetx_amd.ops import hip_ext, use_hip
 
 __all__ = [
     "layernorm", "rmsnorm", "FusedLayerNorm", "FusedRMSNorm", "bias_gelu",
-    "flash_attention", "fused_softmax_causal", "cross_entropy",
+    "flash_attention", "flash_attention_packed", "fused_softmax_causal", "cross_entropy",
     "fused_adamw_flat", "rope", "topp_sampling",
 ]
 
@@ -184,6 +184,52 @@ class _FlashAttnFn(torch.autograd.Function):
 def flash_attention(q, k, v, causal: bool = True, scale: Optional[float] = None):
     """q,k,v: [B, H, S, D] -> o: [B, H, S, D]. O(S) memory, online softmax."""
     return _FlashAttnFn.apply(q, k, v, causal, scale)
+
+
+class _FlashAttnPackedFn(torch.autograd.Function):
+    """Packed-QKV flash attention: input [B, S, h, 3, D] (fused-QKV linear
+    output viewed), output [B, S, h*D] — no split/transpose/cat copies."""
+
+    @staticmethod
+    def forward(ctx, qkv, num_heads, scale):
+        scale = scale if scale is not None else 1.0 / math.sqrt(qkv.shape[-1])
+        if use_hip(qkv):
+            o, lse = hip_ext().attn_fwd_packed(qkv, num_heads, scale)
+        else:
+            B, S, h, _, D = qkv.shape
+            q = qkv[:, :, :, 0].permute(0, 2, 1, 3)
+            k = qkv[:, :, :, 1].permute(0, 2, 1, 3)
+            v = qkv[:, :, :, 2].permute(0, 2, 1, 3)
+            o4, lse = ref.attention_fwd(q, k, v, True, scale)
+            o = o4.permute(0, 2, 1, 3).reshape(B, S, h * D)
+        ctx.save_for_backward(qkv, o, lse)
+        ctx.num_heads, ctx.scale = num_heads, scale
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        qkv, o, lse = ctx.saved_tensors
+        if use_hip(qkv):
+            dqkv = hip_ext().attn_bwd_packed(do.contiguous(), qkv, o, lse,
+                                             ctx.num_heads, ctx.scale)
+        else:
+            B, S, h, _, D = qkv.shape
+            q = qkv[:, :, :, 0].permute(0, 2, 1, 3)
+            k = qkv[:, :, :, 1].permute(0, 2, 1, 3)
+            v = qkv[:, :, :, 2].permute(0, 2, 1, 3)
+            o4 = o.view(B, S, h, D).permute(0, 2, 1, 3)
+            do4 = do.view(B, S, h, D).permute(0, 2, 1, 3)
+            dq, dk, dv = ref.attention_bwd(do4, q, k, v, o4, lse, True,
+                                           ctx.scale)
+            dqkv = torch.stack(
+                (dq.permute(0, 2, 1, 3), dk.permute(0, 2, 1, 3),
+                 dv.permute(0, 2, 1, 3)), dim=3)
+        return dqkv, None, None
+
+
+def flash_attention_packed(qkv, num_heads: int, scale: Optional[float] = None):
+    """qkv [B, S, h, 3, D] -> o [B, S, h*D]; causal."""
+    return _FlashAttnPackedFn.apply(qkv.contiguous(), num_heads, scale)
 
 
 # ---------------------------------------------------------------------------
