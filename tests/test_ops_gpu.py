@@ -242,7 +242,9 @@ def test_gpt_module_end_to_end_gpu(dev):
                    "Model.hidden_dropout_prob=0.0",
                    "Model.attention_probs_dropout_prob=0.0",
                    "Global.micro_batch_size=2", "Global.local_batch_size=4",
-                   "Global.eval_freq=", "Global.save_steps="])
+                   "Global.eval_freq=", "Global.save_steps=",
+                   "Optimizer.lr.name=ConstantLR",
+                   "Optimizer.lr.learning_rate=1e-3"])
     init_dist_env(cfg)
     module = build_module(cfg)
     engine = EagerEngine(cfg, module)
