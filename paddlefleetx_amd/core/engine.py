@@ -96,17 +96,14 @@ class EagerEngine(BasicEngine):
             self.grad_clip_norm = float(gc)
 
         if mode == "train":
-            if self.sharding_degree > 1 and self.sharding_stage >= 2:
-                from paddlefleetx_amd.parallel.zero import ShardedOptimizer
-                self.optimizer = ShardedOptimizer(
-                    self.module.model, opt_cfg, self.hcg,
-                    stage=self.sharding_stage,
-                    lr_value=self.lr_scheduler.get_lr())
-            else:
-                self.optimizer = build_optimizer(
-                    opt_cfg, self.module.model,
-                    lr_value=self.lr_scheduler.get_lr())
-            # broadcast initial params across dp so replicas agree
+            sharding_group = self.hcg.get_sharding_parallel_group() \
+                if self.sharding_degree > 1 else None
+            self.optimizer = build_optimizer(
+                opt_cfg, self.module.model,
+                lr_value=self.lr_scheduler.get_lr(),
+                sharding_group=sharding_group,
+                sharding_stage=self.sharding_stage)
+            # broadcast initial params across dp (and sharding) so replicas agree
             self._sync_params()
         else:
             self.optimizer = None
@@ -120,17 +117,16 @@ class EagerEngine(BasicEngine):
     def _sync_params(self):
         """Broadcast params from dp-rank-0 (reference strategy.py:43 sync_params_buffers)."""
         dp = self.hcg.get_data_parallel_group()
-        groups = [dp]
         sd = self.hcg.get_sharding_parallel_group()
-        if self.sharding_stage == 1:
-            groups.append(sd)
-        for g in groups:
+        for g in (dp, sd):
             if g.world_size <= 1 or not dist.is_initialized():
                 continue
             if isinstance(self.optimizer, FusedAdamW):
                 for b in self.optimizer.buckets:
                     dist.broadcast(b.model_flat, src=g.ranks[0], group=g.group)
-                    b.master.copy_(b.model_flat.float())
+                    b.master.copy_(b.model_shard.float()
+                                   if self.optimizer.sharding_group is not None
+                                   else b.model_flat.float())
             else:
                 for p in self.module.model.parameters():
                     dist.broadcast(p.data, src=g.ranks[0], group=g.group)
@@ -210,16 +206,10 @@ class EagerEngine(BasicEngine):
             self._allreduce_sp_main_grads()
         self._found_inf = 0.0
         if isinstance(self.optimizer, FusedAdamW):
-            # DP (+ZeRO-1 sharding) reduce on fused buffers
+            # ZeRO reduce-scatter over sharding + DP allreduce on fused buffers
             dp = self.hcg.get_data_parallel_group()
-            n_replicas = dp.world_size
-            sd = self.hcg.get_sharding_parallel_group()
-            if self.sharding_degree > 1 and self.sharding_stage == 1:
-                n_replicas *= sd.world_size
-                self.optimizer.reduce_gradients(
-                    self.hcg.get_data_world_group(), avg_factor=1.0)
-            elif dp.world_size > 1:
-                self.optimizer.reduce_gradients(dp, avg_factor=1.0)
+            n_replicas = dp.world_size * self.sharding_degree
+            self.optimizer.reduce_gradients(dp, avg_factor=1.0)
             inv = 1.0 / (n_replicas * self.loss_scale)
             self.optimizer.scale_grads(inv)
             if self.loss_scale != 1.0:

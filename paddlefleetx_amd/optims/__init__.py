@@ -14,7 +14,7 @@ __all__ = ["build_optimizer", "build_lr_scheduler", "FusedAdamW", "AdamW",
 _OPTIMIZERS = {"FusedAdamW": FusedAdamW, "AdamW": AdamW}
 
 
-def build_optimizer(cfg, model, lr_value: float = None):
+def build_optimizer(cfg, model, lr_value: float = None, **extra):
     cfg = dict(cfg or {})
     name = cfg.pop("name", "FusedAdamW")
     cfg.pop("lr", None)
@@ -25,8 +25,12 @@ def build_optimizer(cfg, model, lr_value: float = None):
         raise ValueError(f"unknown optimizer {name}")
     kwargs = {k: v for k, v in cfg.items()
               if k in ("weight_decay", "beta1", "beta2", "epsilon",
-                       "multi_precision", "tensor_fusion")}
-    return _OPTIMIZERS[name](model.named_parameters(),
-                             lr=lr_value or 1e-4, grad_clip=grad_clip,
-                             **kwargs) if name == "FusedAdamW" else \
-        _OPTIMIZERS[name](model.named_parameters(), lr=lr_value or 1e-4, **kwargs)
+                       "multi_precision", "tensor_fusion", "grad_dtype")}
+    if name == "FusedAdamW":
+        return _OPTIMIZERS[name](
+            model.named_parameters(), lr=lr_value or 1e-4,
+            grad_clip=grad_clip,
+            sharding_group=extra.get("sharding_group"),
+            sharding_stage=extra.get("sharding_stage", 1), **kwargs)
+    return _OPTIMIZERS[name](model.named_parameters(), lr=lr_value or 1e-4,
+                             **kwargs)

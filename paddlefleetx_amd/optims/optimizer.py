@@ -40,21 +40,31 @@ def _no_decay(name: str) -> bool:
 
 
 class _Bucket:
-    """One fused storage: flat model copy + grads + fp32 master/m/v."""
+    """One fused storage: flat model copy + grads + fp32 master/m/v.
+
+    With `shard_world > 1` (ZeRO-1/2, reference group_sharded_parallel
+    eager_engine.py:281-307) the fp32 master/exp_avg/exp_avg_sq cover only
+    this rank's 1/shard_world slice; grads are reduce-scattered onto the
+    slice and updated params all-gathered back into model_flat.
+    """
 
     def __init__(self, params: List[torch.nn.Parameter], dtype: torch.dtype,
                  device: torch.device, weight_decay: float,
-                 grad_dtype: torch.dtype):
+                 grad_dtype: torch.dtype, shard_rank: int = 0,
+                 shard_world: int = 1):
         self.params = params
         self.dtype = dtype
         self.grad_dtype = grad_dtype
         self.weight_decay = weight_decay
+        self.shard_rank, self.shard_world = shard_rank, shard_world
         offs = []
         total = 0
         for p in params:
             offs.append(total)
             n = p.numel()
             total += (n + ALIGN_ELEMS - 1) // ALIGN_ELEMS * ALIGN_ELEMS
+        align = ALIGN_ELEMS * shard_world
+        total = (total + align - 1) // align * align
         self.numel = total
         self.offsets = offs
         self.model_flat = torch.zeros(total, dtype=dtype, device=device)
@@ -62,11 +72,24 @@ class _Bucket:
             n = p.numel()
             self.model_flat[off:off + n].copy_(p.data.reshape(-1).to(dtype))
             p.data = self.model_flat[off:off + n].view(p.shape)
-        self.master = self.model_flat.float()
-        self.exp_avg = torch.zeros(total, dtype=torch.float32, device=device)
-        self.exp_avg_sq = torch.zeros(total, dtype=torch.float32, device=device)
+        self.shard_len = total // shard_world
+        self.shard_lo = shard_rank * self.shard_len
+        self.shard_hi = self.shard_lo + self.shard_len
+        self.master = self.model_flat[self.shard_lo:self.shard_hi].float()
+        self.exp_avg = torch.zeros(self.shard_len, dtype=torch.float32,
+                                   device=device)
+        self.exp_avg_sq = torch.zeros(self.shard_len, dtype=torch.float32,
+                                      device=device)
         self.grad_flat = torch.zeros(total, dtype=grad_dtype, device=device)
         self.attach_grads()
+
+    @property
+    def grad_shard(self):
+        return self.grad_flat[self.shard_lo:self.shard_hi]
+
+    @property
+    def model_shard(self):
+        return self.model_flat[self.shard_lo:self.shard_hi]
 
     def attach_grads(self):
         """Point p.grad (same dtype) or p.main_grad (fp32 mode) at views."""
@@ -86,7 +109,7 @@ class FusedAdamW(torch.optim.Optimizer):
                  epsilon: float = 1e-8, weight_decay: float = 0.01,
                  multi_precision: bool = True, grad_clip: Optional[float] = None,
                  tensor_fusion: bool = True, grad_dtype: str = "param",
-                 **unused):
+                 sharding_group=None, sharding_stage: int = 1, **unused):
         named = [(n, p) for n, p in named_params if p.requires_grad]
         params = [p for _, p in named]
         defaults = dict(lr=lr, beta1=beta1, beta2=beta2, epsilon=epsilon,
@@ -99,6 +122,12 @@ class FusedAdamW(torch.optim.Optimizer):
         self._step = 0
         self.found_inf = False
         self._fp32_main_grad = grad_dtype == "float32"
+        self.sharding_group = sharding_group \
+            if sharding_group is not None and sharding_group.world_size > 1 \
+            else None
+        self.sharding_stage = sharding_stage
+        shard_rank = self.sharding_group.rank if self.sharding_group else 0
+        shard_world = self.sharding_group.world_size if self.sharding_group else 1
 
         # split by (dtype, decay?) and pack into <=256MB buckets
         self.buckets: List[_Bucket] = []
@@ -114,14 +143,15 @@ class FusedAdamW(torch.optim.Optimizer):
                 if cur and cur_bytes + nbytes > BUCKET_BYTES:
                     self.buckets.append(_Bucket(cur, dtype, p.device,
                                                 weight_decay if decay else 0.0,
-                                                gdtype))
+                                                gdtype, shard_rank,
+                                                shard_world))
                     cur, cur_bytes = [], 0
                 cur.append(p)
                 cur_bytes += nbytes
             if cur:
                 self.buckets.append(_Bucket(cur, dtype, cur[0].device,
                                             weight_decay if decay else 0.0,
-                                            gdtype))
+                                            gdtype, shard_rank, shard_world))
         self._hooks = []
         if self._fp32_main_grad:
             # fp32 accumulation via post-accumulate hooks (fp16 runs)
@@ -144,26 +174,36 @@ class FusedAdamW(torch.optim.Optimizer):
         return [b.grad_flat for b in self.buckets]
 
     def reduce_gradients(self, group, avg_factor: Optional[float] = None):
-        """Allreduce fused grad buffers over a process group (DP)."""
-        if group is None or getattr(group, "world_size", 1) == 1:
-            return
-        pg = group.group if hasattr(group, "group") else group
-        ws = group.world_size if hasattr(group, "world_size") else dist.get_world_size(pg)
-        for b in self.buckets:
-            dist.all_reduce(b.grad_flat, group=pg)
-            if avg_factor is None or avg_factor != 1.0:
-                b.grad_flat.div_(avg_factor or ws)
+        """Reduce fused grad buffers: reduce-scatter over the sharding group
+        (ZeRO) then allreduce own shard over DP."""
+        sg = self.sharding_group
+        if sg is not None:
+            for b in self.buckets:
+                shard = torch.empty_like(b.grad_shard)
+                dist.reduce_scatter_tensor(shard, b.grad_flat, group=sg.group)
+                b.grad_shard.copy_(shard)
+        if group is not None and getattr(group, "world_size", 1) > 1:
+            pg = group.group if hasattr(group, "group") else group
+            for b in self.buckets:
+                g = b.grad_shard if sg is not None else b.grad_flat
+                dist.all_reduce(g, group=pg)
+        if avg_factor is not None and avg_factor != 1.0:
+            for b in self.buckets:
+                g = b.grad_shard if sg is not None else b.grad_flat
+                g.div_(avg_factor)
 
     def scale_grads(self, factor: float):
         if factor == 1.0:
             return
         for b in self.buckets:
-            b.grad_flat.mul_(factor)
+            g = b.grad_shard if self.sharding_group else b.grad_flat
+            g.mul_(factor)
 
     def check_finite(self) -> bool:
         for b in self.buckets:
+            g = b.grad_shard if self.sharding_group else b.grad_flat
             if not torch.isfinite(
-                    torch.linalg.vector_norm(b.grad_flat, dtype=torch.float32)):
+                    torch.linalg.vector_norm(g, dtype=torch.float32)):
                 return False
         return True
 
@@ -173,16 +213,24 @@ class FusedAdamW(torch.optim.Optimizer):
         device = self.buckets[0].grad_flat.device if self.buckets else "cpu"
         sq = torch.zeros((), dtype=torch.float32, device=device)
         mp_ws = mp_group.world_size if mp_group is not None else 1
+        sg = self.sharding_group
         for b in self.buckets:
             if mp_ws == 1:
-                sq += torch.linalg.vector_norm(
-                    b.grad_flat, dtype=torch.float32) ** 2
+                g = b.grad_shard if sg is not None else b.grad_flat
+                sq += torch.linalg.vector_norm(g, dtype=torch.float32) ** 2
             else:
                 for p, off in zip(b.params, b.offsets):
-                    g = b.grad_flat[off:off + p.numel()]
+                    lo, hi = off, off + p.numel()
+                    if sg is not None:
+                        lo, hi = max(lo, b.shard_lo), min(hi, b.shard_hi)
+                        if lo >= hi:
+                            continue
+                    g = b.grad_flat[lo:hi]
                     if getattr(p, "is_mp", False) or mp_group.rank == 0:
                         sq += torch.linalg.vector_norm(
                             g, dtype=torch.float32) ** 2
+        if sg is not None:
+            dist.all_reduce(sq, group=sg.group)
         if mp_ws > 1:
             dist.all_reduce(sq, group=mp_group.group)
         if pp_group is not None and pp_group.world_size > 1:
@@ -202,10 +250,18 @@ class FusedAdamW(torch.optim.Optimizer):
         if lr is not None:
             self.lr = lr
         self._step += 1
+        sg = self.sharding_group
         for b in self.buckets:
-            fused_adamw_flat(b.master, b.grad_flat, b.exp_avg, b.exp_avg_sq,
-                             b.model_flat, self.lr, self.beta1, self.beta2,
+            grad = b.grad_shard if sg is not None else b.grad_flat
+            model = b.model_shard if sg is not None else b.model_flat
+            fused_adamw_flat(b.master, grad, b.exp_avg, b.exp_avg_sq,
+                             model, self.lr, self.beta1, self.beta2,
                              self.eps, b.weight_decay, self._step)
+        if sg is not None:
+            for b in self.buckets:
+                out = torch.empty_like(b.model_flat)
+                dist.all_gather_into_tensor(out, b.model_shard, group=sg.group)
+                b.model_flat.copy_(out)
 
     # --- checkpoint --------------------------------------------------------
     def state_dict(self):

@@ -271,3 +271,78 @@ def _dp_worker(rank, world, port):
 @pytest.mark.timeout(300)
 def test_dp_engine_sync():
     _run(_dp_worker, 2)
+
+
+# ---------------------------------------------------------------------------
+# ZeRO sharding2: sharded optimizer == unsharded on identical data
+# ---------------------------------------------------------------------------
+
+def _shard_worker(rank, world, port, tmpdir):
+    hcg = _init(rank, world, port, sharding=2)
+    from paddlefleetx_amd.utils.config import get_config
+    from paddlefleetx_amd.models import build_module
+    from paddlefleetx_amd.core import EagerEngine
+    overrides = ["Model.hidden_size=32", "Model.num_layers=2",
+                 "Model.num_attention_heads=4", "Model.vocab_size=128",
+                 "Model.max_position_embeddings=32",
+                 "Model.hidden_dropout_prob=0.0",
+                 "Model.attention_probs_dropout_prob=0.0",
+                 "Global.micro_batch_size=2", "Global.local_batch_size=2",
+                 "Engine.mix_precision.enable=False",
+                 "Distributed.sharding.sharding_degree=2",
+                 "Distributed.sharding.sharding_stage=2",
+                 "Distributed.world_size=2"]
+    cfg = get_config(
+        os.path.join(REPO, "paddlefleetx_amd/configs/nlp/gpt/"
+                     "pretrain_gpt_345M_single_card.yaml"), overrides=overrides)
+    module = build_module(cfg)
+    engine = EagerEngine(cfg, module)
+    torch.manual_seed(42)  # SAME batch on both shards -> grads avg to the same
+    batch = (torch.randint(0, 128, (2, 32)),
+             torch.arange(32).unsqueeze(0).repeat(2, 1),
+             torch.randint(0, 128, (2, 32)), torch.ones(2, 32))
+    engine._fit_impl(batch)
+    # replicas in sync after allgather
+    for b in engine.optimizer.buckets:
+        refb = b.model_flat.clone()
+        dist.broadcast(refb, src=0)
+        assert torch.equal(refb, b.model_flat)
+    if rank == 0:
+        torch.save({n: p.detach().clone()
+                    for n, p in module.model.named_parameters()},
+                   os.path.join(tmpdir, "sharded.pt"))
+    dist.destroy_process_group()
+
+
+def _unsharded_worker(rank, world, port, tmpdir):
+    _init(rank, world, port)
+    from paddlefleetx_amd.utils.config import get_config
+    from paddlefleetx_amd.models import build_module
+    from paddlefleetx_amd.core import EagerEngine
+    cfg = get_config(
+        os.path.join(REPO, "paddlefleetx_amd/configs/nlp/gpt/"
+                     "pretrain_gpt_345M_single_card.yaml"),
+        overrides=["Model.hidden_size=32", "Model.num_layers=2",
+                   "Model.num_attention_heads=4", "Model.vocab_size=128",
+                   "Model.max_position_embeddings=32",
+                   "Model.hidden_dropout_prob=0.0",
+                   "Model.attention_probs_dropout_prob=0.0",
+                   "Global.micro_batch_size=2", "Global.local_batch_size=2",
+                   "Engine.mix_precision.enable=False"])
+    module = build_module(cfg)
+    engine = EagerEngine(cfg, module)
+    torch.manual_seed(42)
+    batch = (torch.randint(0, 128, (2, 32)),
+             torch.arange(32).unsqueeze(0).repeat(2, 1),
+             torch.randint(0, 128, (2, 32)), torch.ones(2, 32))
+    engine._fit_impl(batch)
+    sharded = torch.load(os.path.join(tmpdir, "sharded.pt"), weights_only=False)
+    for n, p in module.model.named_parameters():
+        assert torch.allclose(p.detach(), sharded[n], atol=1e-6), n
+
+
+@pytest.mark.timeout(600)
+def test_zero_sharding_matches_unsharded():
+    with tempfile.TemporaryDirectory() as tmpdir:
+        _run(_shard_worker, 2, (tmpdir,))
+        _run(_unsharded_worker, 1, (tmpdir,))
