@@ -272,9 +272,23 @@ class GPTModel(nn.Module):
                  recompute_granularity: str = "full",
                  sequence_parallel: bool = False,
                  initializer_range: float = 0.02,
+                 moe_configs: Optional[Dict[str, Any]] = None,
                  dtype: Optional[torch.dtype] = None, **unused: Any):
         super().__init__()
         ffn_hidden_size = ffn_hidden_size or 4 * hidden_size
+
+        def _make_expert_module():
+            """MoE FFN when expert_mode (hybrid_model.py:535-587)."""
+            if not (moe_configs and moe_configs.get("expert_mode", False)):
+                return None
+            from paddlefleetx_amd.models.moe import MoELayer
+            return MoELayer(
+                hidden_size, ffn_hidden_size,
+                num_experts=moe_configs.get("num_experts", 1),
+                gate=moe_configs.get("gate", "gshard"),
+                top_k=moe_configs.get("top_k", 2),
+                capacity_factor=moe_configs.get("capacity_factor"),
+                dtype=dtype)
         self.use_recompute = use_recompute
         self.recompute_granularity = recompute_granularity
         self.sequence_parallel = sequence_parallel
@@ -293,7 +307,8 @@ class GPTModel(nn.Module):
                                     init_std=initializer_range,
                                     num_layers_for_scale=num_layers,
                                     use_recompute=use_recompute,
-                                    recompute_granularity=recompute_granularity)
+                                    recompute_granularity=recompute_granularity,
+                                    expert_module=_make_expert_module())
             for _ in range(num_layers)])
         self.final_ln = FusedLayerNorm(hidden_size, dtype=dtype)
         if sequence_parallel:

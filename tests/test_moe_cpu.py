@@ -1,0 +1,182 @@
+"""MoE: gates, capacity, dispatch/combine (single rank + EP2 over gloo)."""
+
+import os
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from tests.test_distributed_cpu import _init, _run
+
+REPO = os.path.join(os.path.dirname(__file__), "..")
+
+
+@pytest.fixture(autouse=True)
+def _single_rank_env():
+    from paddlefleetx_amd.parallel.env import set_hcg
+    from paddlefleetx_amd.parallel.topology import HybridTopology
+    set_hcg(HybridTopology())
+    yield
+
+
+# ---------------------------------------------------------------------------
+# gates
+# ---------------------------------------------------------------------------
+
+def test_naive_gate_topk():
+    from paddlefleetx_amd.models.moe import NaiveGate
+    g = NaiveGate(16, 8, top_k=2)
+    idx, score = g(torch.randn(10, 16))
+    assert idx.shape == (10, 2) and score.shape == (10, 2)
+    assert torch.allclose(score.sum(-1), torch.ones(10), atol=1e-5)
+    assert g.get_loss() is None
+
+
+def test_gshard_gate_aux_loss_balanced_vs_skewed():
+    from paddlefleetx_amd.models.moe import GShardGate
+    torch.manual_seed(0)
+    g = GShardGate(16, 4, random_routing=False).eval()
+    g(torch.randn(256, 16))
+    balanced = float(g.get_loss())
+    # force heavy skew: all logits favor expert 0
+    with torch.no_grad():
+        g.gate.weight.zero_()
+        g.gate.weight[0] += 10.0
+    g(torch.randn(256, 16))
+    skewed = float(g.get_loss())
+    assert skewed > balanced
+
+
+def test_switch_gate_top1():
+    from paddlefleetx_amd.models.moe import SwitchGate
+    g = SwitchGate(16, 4).eval()
+    idx, score = g(torch.randn(32, 16))
+    assert idx.shape == (32, 1)
+    assert torch.all(score == 1.0)
+    assert g.get_loss() is not None
+
+
+def test_build_gate_unknown():
+    from paddlefleetx_amd.models.moe import build_gate
+    with pytest.raises(ValueError):
+        build_gate("nope", 8, 4, 2)
+
+
+# ---------------------------------------------------------------------------
+# single-rank MoE layer == dense expert mixture reference
+# ---------------------------------------------------------------------------
+
+def test_moe_layer_single_rank_matches_reference():
+    from paddlefleetx_amd.models.moe import MoELayer, NaiveGate
+    torch.manual_seed(1)
+    layer = MoELayer(8, 16, num_experts=4, gate=NaiveGate(8, 4, top_k=2))
+    x = torch.randn(3, 5, 8, requires_grad=True)
+    y = layer(x)
+    assert y.shape == x.shape
+
+    # reference: explicit per-token loop
+    xf = x.detach().reshape(-1, 8)
+    layer.gate.eval()
+    idx, score = layer.gate(xf)
+    ref = torch.zeros_like(xf)
+    for t in range(xf.shape[0]):
+        for k in range(2):
+            e = int(idx[t, k])
+            ref[t] += score[t, k] * layer.experts[e](xf[t:t + 1])[0]
+    assert torch.allclose(y.reshape(-1, 8), ref, atol=1e-5), \
+        (y.reshape(-1, 8) - ref).abs().max()
+
+    # gradients flow to experts and gate
+    y.sum().backward()
+    assert x.grad is not None
+    assert any(p.grad is not None and p.grad.abs().sum() > 0
+               for p in layer.experts.parameters())
+    assert layer.gate.gate.weight.grad is not None
+
+
+def test_moe_capacity_drops_overflow():
+    from paddlefleetx_amd.models.moe import MoELayer, NaiveGate
+    torch.manual_seed(2)
+    # capacity so small that most slots drop
+    layer = MoELayer(8, 16, num_experts=2, gate=NaiveGate(8, 2, top_k=1),
+                     top_k=1, capacity_factor=0.25)
+    x = torch.randn(16, 8)
+    y = layer(x)
+    assert y.shape == x.shape
+    # dropped tokens produce zero output rows; capacity 0.25*16/2 = 2 per expert
+    nonzero_rows = (y.abs().sum(-1) > 0).sum()
+    assert nonzero_rows <= 4
+
+
+def test_moe_grad_clip_separates_experts():
+    from paddlefleetx_amd.optims.grad_clip import \
+        clip_grad_for_moe_by_global_norm
+    p_shared = torch.nn.Parameter(torch.ones(4))
+    p_expert = torch.nn.Parameter(torch.ones(4))
+    p_expert.is_expert = True
+    p_shared.grad = torch.full((4,), 3.0)
+    p_expert.grad = torch.full((4,), 4.0)
+    gn = clip_grad_for_moe_by_global_norm([p_shared, p_expert], clip_norm=1.0)
+    assert abs(gn - 10.0) < 1e-4  # sqrt(9*4 + 16*4) = 10
+    assert torch.allclose(p_shared.grad, torch.full((4,), 0.3), atol=1e-4)
+
+
+# ---------------------------------------------------------------------------
+# EP2 over gloo: distributed dispatch == single-rank result
+# ---------------------------------------------------------------------------
+
+def _ep_worker(rank, world, port):
+    hcg = _init(rank, world, port, dp=2)  # ep group = dp when no mp
+    from paddlefleetx_amd.models.moe import MoELayer, NaiveGate
+    from paddlefleetx_amd.parallel.env import set_hcg
+    from paddlefleetx_amd.parallel.topology import HybridTopology
+
+    torch.manual_seed(42)  # same on both ranks
+    gate = NaiveGate(8, 4, top_k=2)
+    ep_layer = MoELayer(8, 16, num_experts=4, gate=gate)
+    assert ep_layer.num_local_experts == 2
+
+    # Build the single-rank twin holding ALL experts with identical weights:
+    # gather expert weights from both ranks.
+    g = ep_layer.ep_group_info.group
+    all_params = []
+    for e in ep_layer.experts:
+        vec = torch.cat([p.detach().reshape(-1) for p in e.parameters()])
+        gathered = [torch.empty_like(vec) for _ in range(world)]
+        dist.all_gather(gathered, vec, group=g)
+        all_params.append(gathered)
+
+    from paddlefleetx_amd.models.moe import ExpertLayer
+    full_experts = []
+    with torch.no_grad():
+        for e_idx in range(4):
+            src_rank, local_e = divmod(e_idx, 2)
+            vec = all_params[local_e][src_rank]
+            e = ExpertLayer(8, 16)
+            off = 0
+            for p in e.parameters():
+                p.copy_(vec[off:off + p.numel()].view(p.shape))
+                off += p.numel()
+            full_experts.append(e)
+
+    torch.manual_seed(100 + rank)  # different tokens per rank
+    x = torch.randn(6, 8)
+    y = ep_layer(x)
+
+    # reference: per-token loop over the full (gathered) expert set
+    gate.eval()
+    idx, score = gate(x)
+    y_ref = torch.zeros_like(x)
+    with torch.no_grad():
+        for t in range(x.shape[0]):
+            for k in range(2):
+                y_ref[t] += score[t, k] * \
+                    full_experts[int(idx[t, k])](x[t:t + 1])[0]
+    assert torch.allclose(y, y_ref, atol=1e-5), (y - y_ref).abs().max()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_moe_ep2_matches_single_rank():
+    _run(_ep_worker, 2)
