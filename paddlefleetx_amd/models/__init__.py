@@ -6,7 +6,9 @@ from paddlefleetx_amd.utils.log import logger
 
 
 def build_module(config):
-    name = config["Model"]["name"]
+    # vision configs name the module under Model.module (vit base.yaml),
+    # language configs under Model.name
+    name = config["Model"].get("module") or config["Model"]["name"]
     from paddlefleetx_amd.models.language_module import (GPTGenerationModule,
                                                          GPTModule)
     table = {

@@ -75,6 +75,43 @@ class ConstantLR(_Scheduler):
         return self.lr
 
 
+class ViTLRScheduler(_Scheduler):
+    """Linear warmup + cosine/linear decay (reference lr_scheduler.py
+    ViTLRScheduler; decay_type 'cosine' | 'linear')."""
+
+    def __init__(self, learning_rate: float, warmup_steps: int = 10000,
+                 total_steps: int = 300000, decay_type: str = "cosine",
+                 **unused):
+        super().__init__()
+        self.lr = learning_rate
+        self.warmup_steps = warmup_steps
+        self.total_steps = total_steps
+        self.decay_type = decay_type
+
+    def get_lr(self) -> float:
+        if self.warmup_steps > 0 and self.num < self.warmup_steps:
+            return self.lr * self.num / max(1, self.warmup_steps)
+        ratio = (self.num - self.warmup_steps) / max(
+            1, self.total_steps - self.warmup_steps)
+        ratio = min(1.0, ratio)
+        if self.decay_type == "cosine":
+            return self.lr * 0.5 * (1.0 + math.cos(math.pi * ratio))
+        return self.lr * (1.0 - ratio)
+
+
+class MultiStepDecay(_Scheduler):
+    def __init__(self, learning_rate: float, milestones=(30, 60, 90),
+                 gamma: float = 0.1, **unused):
+        super().__init__()
+        self.lr = learning_rate
+        self.milestones = sorted(milestones)
+        self.gamma = gamma
+
+    def get_lr(self) -> float:
+        n = sum(1 for m in self.milestones if self.num >= m)
+        return self.lr * (self.gamma ** n)
+
+
 def build_lr_scheduler(cfg) -> _Scheduler:
     cfg = dict(cfg or {})
     name = cfg.pop("name", "ConstantLR")
@@ -82,6 +119,8 @@ def build_lr_scheduler(cfg) -> _Scheduler:
         "CosineAnnealingWithWarmupDecay": CosineAnnealingWithWarmupDecay,
         "LinearDecayWithWarmup": LinearDecayWithWarmup,
         "ConstantLR": ConstantLR,
+        "ViTLRScheduler": ViTLRScheduler,
+        "MultiStepDecay": MultiStepDecay,
     }
     if name not in table:
         raise ValueError(f"unknown lr scheduler {name}")
