@@ -1,19 +1,26 @@
-// Flash attention (causal) forward + backward for gfx950 / CDNA4.
+// Flash attention forward + backward for gfx950 / CDNA4.
 //
 // Replaces the reference's paddle `flash_attention` consumption
-// (ppfleetx hybrid_model.py:284-301) with hand-written MFMA kernels:
-//   - mfma_f32_16x16x32_bf16 tiles, LDS-staged K/V with +16B row padding
-//     (bank-conflict fix, guide §6 G4), online softmax (m, l per row).
-//   - fwd: one 4-wave block per 64 q rows; each wave owns 16 rows.
-//   - bwd: split into dKV kernel (parallel over kv tiles) and dQ kernel
-//     (parallel over q tiles) so neither needs atomics; both recompute
-//     P from (q, k, lse) — the standard flash backward decomposition.
+// (ppfleetx hybrid_model.py:284-301) with hand-written MFMA kernels.
+//
+// Forward (v2): 8-wave workgroup, 128-row Q tile (16 rows/wave), 64-row
+// KV tiles double-buffered in LDS with split staging (issue global->reg
+// loads BEFORE the tile's compute, ds_write after — guide T14): HBM
+// latency for tile t+1 hides under tile t's MFMAs, one __syncthreads per
+// tile. Online softmax in registers per 16-row fragment; hardware
+// v_exp (__expf). mfma_f32_16x16x32_bf16 tiles; K row-major +16B-padded
+// (conflict-free b128 reads), V transposed at stage time.
+//
+// Backward: split into dKV kernel (parallel over kv tiles) and dQ kernel
+// (parallel over q tiles) so neither needs atomics; both recompute
+// P from (q, k, lse) — the standard flash backward decomposition.
 //
 // All tensors are STRIDED: (batch_stride, head_stride, row_stride) in
 // elements with the head_dim axis contiguous. This lets the kernels read
 // the fused-QKV linear output [B, S, h, 3, D] and write attention output
 // [B, S, h*D] directly — no split/transpose/cat copies on either side.
 // lse/delta are [B, H, S] fp32 contiguous. D in {64, 128}; any S.
+// Both causal and bidirectional (ViT) attention.
 #include "common.h"
 
 #include <torch/extension.h>
@@ -23,13 +30,19 @@ namespace {
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf8;
 typedef __attribute__((ext_vector_type(4))) float f4;
+typedef __attribute__((ext_vector_type(8))) short short8v;
 
 #define MFMA_BF16(a, b, c) __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0)
 
-constexpr int TILE = 64;       // q-tile and kv-tile rows
-constexpr int NWAVES = 4;      // waves per block
+constexpr int TILE = 64;       // kv-tile rows (and bwd q-tile rows)
+constexpr int NWAVES = 4;      // waves per block (backward kernels)
 constexpr int BLOCKT = NWAVES * WAVE;
 constexpr int PAD = 8;         // +16 B LDS row pad: alignment + bank spread
+
+// forward geometry
+constexpr int FW_WAVES = 8;            // 512 threads
+constexpr int FW_BLOCKT = FW_WAVES * WAVE;
+constexpr int QTILE = FW_WAVES * 16;   // 128 q rows per block
 
 struct Strided {
   const __hip_bfloat16* p;
@@ -47,7 +60,8 @@ struct StridedMut {
 };
 
 // ---------------------------------------------------------------------------
-// LDS staging. Row-major [TILE][D+PAD] and transposed [D][TILE+PAD].
+// LDS staging (synchronous forms, used by the backward kernels).
+// Row-major [TILE][D+PAD] and transposed [D][TILE+PAD].
 // Rows beyond `nvalid` zero-filled. `g` points at row 0 of the tile.
 // ---------------------------------------------------------------------------
 template <int D>
@@ -66,26 +80,6 @@ DEV_INLINE void stage_rowmajor(const __hip_bfloat16* __restrict__ g, long rs,
     }
     *reinterpret_cast<short4v*>(lds + row * RS + col) = lo;
     *reinterpret_cast<short4v*>(lds + row * RS + col + 4) = hi;
-  }
-}
-
-template <int D>
-DEV_INLINE void stage_transposed(const __hip_bfloat16* __restrict__ g, long rs,
-                                 int nvalid, unsigned short* lds) {
-  constexpr int RS = TILE + PAD;
-  constexpr int NV = TILE * D / 8;
-  for (int v = threadIdx.x; v < NV; v += BLOCKT) {
-    int row = v / (D / 8);
-    int col = (v % (D / 8)) * 8;
-    unsigned short tmp[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-    if (row < nvalid) {
-      const unsigned short* src = (const unsigned short*)g + (long)row * rs + col;
-      *reinterpret_cast<short4v*>(tmp) = *reinterpret_cast<const short4v*>(src);
-      *reinterpret_cast<short4v*>(tmp + 4) =
-          *reinterpret_cast<const short4v*>(src + 4);
-    }
-#pragma unroll
-    for (int j = 0; j < 8; ++j) lds[(col + j) * RS + row] = tmp[j];
   }
 }
 
@@ -160,20 +154,24 @@ DEV_INLINE bf8 read_a_frag_lds(const unsigned short* lds, int row_stride,
 }
 
 // ===========================================================================
-// Forward
+// Forward v2: 8 waves, 128-row Q tile, double-buffered KV, split staging.
 // ===========================================================================
-template <int D>
-__global__ __launch_bounds__(BLOCKT) void attn_fwd_kernel(
+template <int D, bool CAUSAL>
+__global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
     Strided q, Strided k, Strided v, StridedMut o, float* __restrict__ lse_out,
-    int H, int S, float scale, int q_tiles) {
-  constexpr int KRS = D + PAD;
-  constexpr int VRS = TILE + PAD;
-  constexpr int PRS = TILE + PAD;
+    int H, int S, float scale, int kv_total) {
+  constexpr int KRS = D + PAD;          // K row-major row stride
+  constexpr int VRS = TILE + PAD;       // V^T row stride
+  constexpr int PRS = TILE + PAD;       // per-wave P row stride
   constexpr int NDT = D / 16;
+  constexpr int KSZ = TILE * KRS;       // one K buffer (shorts)
+  constexpr int VSZ = D * VRS;          // one V^T buffer
 
-  __shared__ unsigned short k_lds[TILE * KRS];
-  __shared__ unsigned short vt_lds[D * VRS];
-  __shared__ unsigned short p_lds[NWAVES * 16 * PRS];
+  // single __shared__ object (guide §5 trap 4a)
+  __shared__ unsigned short smem[2 * KSZ + 2 * VSZ + FW_WAVES * 16 * PRS];
+  unsigned short* k_lds = smem;                  // [2][KSZ]
+  unsigned short* vt_lds = smem + 2 * KSZ;       // [2][VSZ]
+  unsigned short* p_lds = smem + 2 * KSZ + 2 * VSZ;
 
   const int qt = blockIdx.x;
   const int b = blockIdx.y / H, hh = blockIdx.y % H;
@@ -185,7 +183,7 @@ __global__ __launch_bounds__(BLOCKT) void attn_fwd_kernel(
   const __hip_bfloat16* kp = k.at(b, hh);
   const __hip_bfloat16* vp = v.at(b, hh);
 
-  const int qrow0 = qt * TILE + wid * 16;
+  const int qrow0 = qt * QTILE + wid * 16;
   bf8 qfrag[D / 32];
   load_a_frags<D>(qp, q.rs, qrow0, S, lane, qfrag);
 
@@ -199,96 +197,172 @@ __global__ __launch_bounds__(BLOCKT) void attn_fwd_kernel(
   const int ccol = lane & 15;
   const int crow4 = (lane >> 4) * 4;
   const int my_qrow = qrow0 + crow4;
+  const int wave_last_row = min(qrow0 + 15, S - 1);
 
-  const int kv_tiles = min(q_tiles, qt + 1);
+  const int kv_tiles = CAUSAL
+      ? min(kv_total, (qt * QTILE + QTILE - 1) / TILE + 1)
+      : kv_total;
   unsigned short* myp = p_lds + wid * 16 * PRS;
+
+  // ---- staging thread map: D/16 threads per row, 16 shorts each ----
+  constexpr int TPR = D / 16;  // threads per kv row
+  const int st_row = threadIdx.x / TPR;          // > TILE-1 threads idle (D=64)
+  const int st_col = (threadIdx.x % TPR) * 16;
+  const bool st_on = st_row < TILE;
+  short8v kreg[2], vreg[2];
+
+  // prologue: load tile 0 and write buffer 0
+  {
+    const int nvalid = min(TILE, S);
+#pragma unroll
+    for (int h2 = 0; h2 < 2; ++h2) { kreg[h2] = short8v{}; vreg[h2] = short8v{}; }
+    if (st_on && st_row < nvalid) {
+      const unsigned short* ks = (const unsigned short*)kp +
+                                 (long)st_row * k.rs + st_col;
+      const unsigned short* vs = (const unsigned short*)vp +
+                                 (long)st_row * v.rs + st_col;
+      kreg[0] = *reinterpret_cast<const short8v*>(ks);
+      kreg[1] = *reinterpret_cast<const short8v*>(ks + 8);
+      vreg[0] = *reinterpret_cast<const short8v*>(vs);
+      vreg[1] = *reinterpret_cast<const short8v*>(vs + 8);
+    }
+    if (st_on) {
+      *reinterpret_cast<short8v*>(k_lds + st_row * KRS + st_col) = kreg[0];
+      *reinterpret_cast<short8v*>(k_lds + st_row * KRS + st_col + 8) = kreg[1];
+      const unsigned short* vr = (const unsigned short*)&vreg[0];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        vt_lds[(st_col + j) * VRS + st_row] = vr[j];
+      vr = (const unsigned short*)&vreg[1];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        vt_lds[(st_col + 8 + j) * VRS + st_row] = vr[j];
+    }
+  }
+  __syncthreads();
 
   for (int kt = 0; kt < kv_tiles; ++kt) {
     const int kv0 = kt * TILE;
-    const int nvalid = min(TILE, S - kv0);
+    const int cur = kt & 1;
+    const unsigned short* kb_lds = k_lds + cur * KSZ;
+    const unsigned short* vb_lds = vt_lds + cur * VSZ;
+
+    // ---- issue next tile's global loads (land during this tile's MFMAs) --
+    const bool have_next = (kt + 1) < kv_tiles;
+    if (have_next) {
+      const int nrow = (kt + 1) * TILE + st_row;
+#pragma unroll
+      for (int h2 = 0; h2 < 2; ++h2) { kreg[h2] = short8v{}; vreg[h2] = short8v{}; }
+      if (st_on && nrow < S) {
+        const unsigned short* ks = (const unsigned short*)kp +
+                                   (long)nrow * k.rs + st_col;
+        const unsigned short* vs = (const unsigned short*)vp +
+                                   (long)nrow * v.rs + st_col;
+        kreg[0] = *reinterpret_cast<const short8v*>(ks);
+        kreg[1] = *reinterpret_cast<const short8v*>(ks + 8);
+        vreg[0] = *reinterpret_cast<const short8v*>(vs);
+        vreg[1] = *reinterpret_cast<const short8v*>(vs + 8);
+      }
+    }
+
+    // waves whose rows are entirely above this kv tile skip compute
+    // (they still stage and hit the barrier)
+    const bool active = !CAUSAL || (kv0 <= wave_last_row);
+    if (active) {
+      // ---- QK^T ----
+      f4 s[4];
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct) {
+        f4 acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int kc = 0; kc < D / 32; ++kc) {
+          bf8 kb = read_b_frag(kb_lds, KRS, ct * 16, kc * 32, lane);
+          acc = MFMA_BF16(qfrag[kc], kb, acc);
+        }
+        s[ct] = acc;
+      }
+
+      // ---- mask + scale + online softmax ----
+      float pmax[4] = {-INFINITY, -INFINITY, -INFINITY, -INFINITY};
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct) {
+        int kcol = kv0 + ct * 16 + ccol;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float val = s[ct][r] * scale;
+          if ((CAUSAL && kcol > my_qrow + r) || kcol >= S) val = -INFINITY;
+          s[ct][r] = val;
+          pmax[r] = fmaxf(pmax[r], val);
+        }
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) pmax[r] = group16_reduce_max(pmax[r]);
+
+      float alpha[4];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float nm = fmaxf(m_r[r], pmax[r]);
+        if (nm == -INFINITY) nm = 0.f;
+        alpha[r] = (m_r[r] == -INFINITY) ? 0.f : __expf(m_r[r] - nm);
+        m_r[r] = (m_r[r] == -INFINITY && pmax[r] == -INFINITY) ? -INFINITY : nm;
+      }
+
+      float psum[4] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float p = (s[ct][r] == -INFINITY) ? 0.f : __expf(s[ct][r] - m_r[r]);
+          s[ct][r] = p;
+          psum[r] += p;
+        }
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        psum[r] = group16_reduce_sum(psum[r]);
+        l_r[r] = l_r[r] * alpha[r] + psum[r];
+      }
+#pragma unroll
+      for (int i = 0; i < NDT; ++i) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) oacc[i][r] *= alpha[r];
+      }
+
+      // ---- P -> per-wave LDS for A-fragments ----
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          myp[(crow4 + r) * PRS + ct * 16 + ccol] = f2bf_raw(s[ct][r]);
+        }
+      }
+
+      // ---- PV ----
+#pragma unroll
+      for (int kc = 0; kc < 2; ++kc) {
+        bf8 pa = read_a_frag_lds(myp, PRS, kc * 32, lane);
+#pragma unroll
+        for (int dt = 0; dt < NDT; ++dt) {
+          bf8 vb = read_b_frag(vb_lds, VRS, dt * 16, kc * 32, lane);
+          oacc[dt] = MFMA_BF16(pa, vb, oacc[dt]);
+        }
+      }
+    }
+
+    // ---- write next tile into the other buffer, one barrier per tile ----
+    if (have_next && st_on) {
+      unsigned short* kd = k_lds + (cur ^ 1) * KSZ;
+      unsigned short* vd = vt_lds + (cur ^ 1) * VSZ;
+      *reinterpret_cast<short8v*>(kd + st_row * KRS + st_col) = kreg[0];
+      *reinterpret_cast<short8v*>(kd + st_row * KRS + st_col + 8) = kreg[1];
+      const unsigned short* vr = (const unsigned short*)&vreg[0];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vd[(st_col + j) * VRS + st_row] = vr[j];
+      vr = (const unsigned short*)&vreg[1];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vd[(st_col + 8 + j) * VRS + st_row] = vr[j];
+    }
     __syncthreads();
-    stage_rowmajor<D>(kp + (long)kv0 * k.rs, k.rs, nvalid, k_lds);
-    stage_transposed<D>(vp + (long)kv0 * v.rs, v.rs, nvalid, vt_lds);
-    __syncthreads();
-
-    // ---- QK^T ----
-    f4 s[4];
-#pragma unroll
-    for (int ct = 0; ct < 4; ++ct) {
-      f4 acc = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-      for (int kc = 0; kc < D / 32; ++kc) {
-        bf8 kb = read_b_frag(k_lds, KRS, ct * 16, kc * 32, lane);
-        acc = MFMA_BF16(qfrag[kc], kb, acc);
-      }
-      s[ct] = acc;
-    }
-
-    // ---- mask + scale + online softmax ----
-    float pmax[4] = {-INFINITY, -INFINITY, -INFINITY, -INFINITY};
-#pragma unroll
-    for (int ct = 0; ct < 4; ++ct) {
-      int kcol = kv0 + ct * 16 + ccol;
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        float val = s[ct][r] * scale;
-        if (kcol > my_qrow + r || kcol >= S) val = -INFINITY;
-        s[ct][r] = val;
-        pmax[r] = fmaxf(pmax[r], val);
-      }
-    }
-#pragma unroll
-    for (int r = 0; r < 4; ++r) pmax[r] = group16_reduce_max(pmax[r]);
-
-    float alpha[4];
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      float nm = fmaxf(m_r[r], pmax[r]);
-      if (nm == -INFINITY) nm = 0.f;
-      alpha[r] = (m_r[r] == -INFINITY) ? 0.f : expf(m_r[r] - nm);
-      m_r[r] = (m_r[r] == -INFINITY && pmax[r] == -INFINITY) ? -INFINITY : nm;
-    }
-
-    float psum[4] = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-    for (int ct = 0; ct < 4; ++ct) {
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        float p = (s[ct][r] == -INFINITY) ? 0.f : expf(s[ct][r] - m_r[r]);
-        s[ct][r] = p;
-        psum[r] += p;
-      }
-    }
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      psum[r] = group16_reduce_sum(psum[r]);
-      l_r[r] = l_r[r] * alpha[r] + psum[r];
-    }
-#pragma unroll
-    for (int i = 0; i < NDT; ++i) {
-#pragma unroll
-      for (int r = 0; r < 4; ++r) oacc[i][r] *= alpha[r];
-    }
-
-    // ---- P -> LDS for A-fragments ----
-#pragma unroll
-    for (int ct = 0; ct < 4; ++ct) {
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        myp[(crow4 + r) * PRS + ct * 16 + ccol] = f2bf_raw(s[ct][r]);
-      }
-    }
-
-    // ---- PV ----
-#pragma unroll
-    for (int kc = 0; kc < 2; ++kc) {
-      bf8 pa = read_a_frag_lds(myp, PRS, kc * 32, lane);
-#pragma unroll
-      for (int dt = 0; dt < NDT; ++dt) {
-        bf8 vb = read_b_frag(vt_lds, VRS, dt * 16, kc * 32, lane);
-        oacc[dt] = MFMA_BF16(pa, vb, oacc[dt]);
-      }
-    }
   }
 
   // ---- epilogue (strided o) ----
@@ -317,10 +391,6 @@ __global__ void attn_bwd_delta_kernel(Strided dout, Strided o,
                                       float* __restrict__ delta, int H, int S,
                                       int D) {
   long row = (long)blockIdx.x * (BLOCKT / WAVE) + threadIdx.x / WAVE;
-  long rows = (long)gridDim.x * (BLOCKT / WAVE);
-  long total = 0;  // computed by caller grid; bounds via S*H*B passed in grid
-  (void)rows; (void)total;
-  // decode (b, h, s)
   long bhs = row;
   int s = (int)(bhs % S);
   long bh = bhs / S;
@@ -341,7 +411,7 @@ __global__ void attn_bwd_delta_kernel(Strided dout, Strided o,
 // ===========================================================================
 // Backward dK/dV (transposed score space S^T = K Q^T)
 // ===========================================================================
-template <int D>
+template <int D, bool CAUSAL>
 __global__ __launch_bounds__(BLOCKT) void attn_bwd_dkv_kernel(
     Strided q, Strided k, Strided v, Strided dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
@@ -386,7 +456,7 @@ __global__ __launch_bounds__(BLOCKT) void attn_bwd_dkv_kernel(
   const int my_kvrow = kvrow0 + crow4;
   unsigned short* myp = p_lds + wid * 16 * TS;
 
-  for (int qt = kt; qt < q_tiles; ++qt) {
+  for (int qt = CAUSAL ? kt : 0; qt < q_tiles; ++qt) {
     const int q0 = qt * TILE;
     const int nvalid = min(TILE, S - q0);
     __syncthreads();
@@ -415,8 +485,8 @@ __global__ __launch_bounds__(BLOCKT) void attn_bwd_dkv_kernel(
       for (int r = 0; r < 4; ++r) {
         int kvr = my_kvrow + r;
         float p = 0.f;
-        if (qcol >= kvr && qcol < S && kvr < S)
-          p = expf(acc[r] * scale - l);
+        if ((!CAUSAL || qcol >= kvr) && qcol < S && kvr < S)
+          p = __expf(acc[r] * scale - l);
         pt[ct][r] = p;
       }
     }
@@ -491,7 +561,7 @@ __global__ __launch_bounds__(BLOCKT) void attn_bwd_dkv_kernel(
 // ===========================================================================
 // Backward dQ
 // ===========================================================================
-template <int D>
+template <int D, bool CAUSAL>
 __global__ __launch_bounds__(BLOCKT) void attn_bwd_dq_kernel(
     Strided q, Strided k, Strided v, Strided dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
@@ -538,7 +608,7 @@ __global__ __launch_bounds__(BLOCKT) void attn_bwd_dq_kernel(
     dlt_r[r] = (qr < S) ? delta[bh * (long)S + qr] : 0.f;
   }
 
-  const int kv_tiles = min(q_tiles, qt + 1);
+  const int kv_tiles = CAUSAL ? min(q_tiles, qt + 1) : q_tiles;
   for (int ktl = 0; ktl < kv_tiles; ++ktl) {
     const int kv0 = ktl * TILE;
     const int nvalid = min(TILE, S - kv0);
@@ -561,8 +631,8 @@ __global__ __launch_bounds__(BLOCKT) void attn_bwd_dq_kernel(
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         float pv = 0.f;
-        if (kcol <= my_qrow + r && kcol < S && my_qrow + r < S)
-          pv = expf(acc[r] * scale - lse_r[r]);
+        if ((!CAUSAL || kcol <= my_qrow + r) && kcol < S && my_qrow + r < S)
+          pv = __expf(acc[r] * scale - lse_r[r]);
         p[ct][r] = pv;
       }
     }
@@ -636,21 +706,23 @@ void check_attn_tensor(const torch::Tensor& t, int d_dim, int D) {
 }
 
 void launch_fwd(Strided q, Strided k, Strided v, StridedMut o, float* lse,
-                int B, int H, int S, int D, float scale) {
-  int q_tiles = (S + TILE - 1) / TILE;
-  dim3 grid(q_tiles, B * H);
+                int B, int H, int S, int D, float scale, bool causal) {
+  int q_blocks = (S + QTILE - 1) / QTILE;
+  int kv_total = (S + TILE - 1) / TILE;
+  dim3 grid(q_blocks, B * H);
   auto stream = at::hip::getCurrentHIPStream();
-  if (D == 128)
-    hipLaunchKernelGGL((attn_fwd_kernel<128>), grid, dim3(BLOCKT), 0, stream,
-                       q, k, v, o, lse, H, S, scale, q_tiles);
-  else
-    hipLaunchKernelGGL((attn_fwd_kernel<64>), grid, dim3(BLOCKT), 0, stream,
-                       q, k, v, o, lse, H, S, scale, q_tiles);
+#define LAUNCH_FWD(DD, CC)                                                  \
+  hipLaunchKernelGGL((attn_fwd_kernel<DD, CC>), grid, dim3(FW_BLOCKT), 0,   \
+                     stream, q, k, v, o, lse, H, S, scale, kv_total)
+  if (D == 128) { if (causal) LAUNCH_FWD(128, true); else LAUNCH_FWD(128, false); }
+  else          { if (causal) LAUNCH_FWD(64, true);  else LAUNCH_FWD(64, false); }
+#undef LAUNCH_FWD
 }
 
 void launch_bwd(Strided q, Strided k, Strided v, Strided dout, Strided o,
                 StridedMut dq, StridedMut dk, StridedMut dv, float* lse,
-                float* delta, int B, int H, int S, int D, float scale) {
+                float* delta, int B, int H, int S, int D, float scale,
+                bool causal) {
   auto stream = at::hip::getCurrentHIPStream();
   long rows = (long)B * H * S;
   hipLaunchKernelGGL(attn_bwd_delta_kernel,
@@ -658,21 +730,18 @@ void launch_bwd(Strided q, Strided k, Strided v, Strided dout, Strided o,
                      dim3(BLOCKT), 0, stream, dout, o, delta, H, S, D);
   int q_tiles = (S + TILE - 1) / TILE;
   dim3 grid(q_tiles, B * H);
-  if (D == 128) {
-    hipLaunchKernelGGL((attn_bwd_dkv_kernel<128>), grid, dim3(BLOCKT), 0,
-                       stream, q, k, v, dout, lse, delta, dk, dv, H, S, scale,
-                       q_tiles);
-    hipLaunchKernelGGL((attn_bwd_dq_kernel<128>), grid, dim3(BLOCKT), 0,
-                       stream, q, k, v, dout, lse, delta, dq, H, S, scale,
-                       q_tiles);
-  } else {
-    hipLaunchKernelGGL((attn_bwd_dkv_kernel<64>), grid, dim3(BLOCKT), 0,
-                       stream, q, k, v, dout, lse, delta, dk, dv, H, S, scale,
-                       q_tiles);
-    hipLaunchKernelGGL((attn_bwd_dq_kernel<64>), grid, dim3(BLOCKT), 0,
-                       stream, q, k, v, dout, lse, delta, dq, H, S, scale,
-                       q_tiles);
-  }
+#define LAUNCH_BWD(DD, CC)                                                   \
+  do {                                                                       \
+    hipLaunchKernelGGL((attn_bwd_dkv_kernel<DD, CC>), grid, dim3(BLOCKT), 0, \
+                       stream, q, k, v, dout, lse, delta, dk, dv, H, S,      \
+                       scale, q_tiles);                                      \
+    hipLaunchKernelGGL((attn_bwd_dq_kernel<DD, CC>), grid, dim3(BLOCKT), 0,  \
+                       stream, q, k, v, dout, lse, delta, dq, H, S, scale,   \
+                       q_tiles);                                             \
+  } while (0)
+  if (D == 128) { if (causal) LAUNCH_BWD(128, true); else LAUNCH_BWD(128, false); }
+  else          { if (causal) LAUNCH_BWD(64, true);  else LAUNCH_BWD(64, false); }
+#undef LAUNCH_BWD
 }
 
 }  // namespace
@@ -686,7 +755,6 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, bool causal,
                                     double scale) {
   TORCH_CHECK(q.dim() == 4 && q.is_contiguous());
-  TORCH_CHECK(causal, "attn_fwd: only causal attention implemented");
   int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
   TORCH_CHECK(k.size(2) == S, "q and k seq length must match");
   TORCH_CHECK(D == 64 || D == 128, "head_dim must be 64 or 128");
@@ -695,7 +763,7 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
   auto lse = torch::empty({B, H, S}, q.options().dtype(torch::kFloat));
   launch_fwd(strided_of(q, 0, 1, 2), strided_of(k, 0, 1, 2),
              strided_of(v, 0, 1, 2), strided_mut_of(o, 0, 1, 2),
-             lse.data_ptr<float>(), B, H, S, D, (float)scale);
+             lse.data_ptr<float>(), B, H, S, D, (float)scale, causal);
   return {o, lse};
 }
 
@@ -703,7 +771,6 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     torch::Tensor k, torch::Tensor v,
                                     torch::Tensor o, torch::Tensor lse,
                                     bool causal, double scale) {
-  TORCH_CHECK(causal);
   int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
   auto dq = torch::empty_like(q);
   auto dk = torch::empty_like(k);
@@ -715,7 +782,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
              strided_of(o, 0, 1, 2), strided_mut_of(dq, 0, 1, 2),
              strided_mut_of(dk, 0, 1, 2), strided_mut_of(dv, 0, 1, 2),
              lse.data_ptr<float>(), delta.data_ptr<float>(), B, H, S, D,
-             (float)scale);
+             (float)scale, causal);
   return {dq, dk, dv};
 }
 
@@ -736,7 +803,8 @@ std::vector<torch::Tensor> attn_fwd_packed(torch::Tensor qkv, long num_heads,
   Strided v = q; v.p += 2 * D;
   StridedMut om{(__hip_bfloat16*)o.data_ptr(), (long)S * H * D, (long)D,
                 (long)H * D};
-  launch_fwd(q, k, v, om, lse.data_ptr<float>(), B, H, S, D, (float)scale);
+  launch_fwd(q, k, v, om, lse.data_ptr<float>(), B, H, S, D, (float)scale,
+             /*causal=*/true);
   return {o, lse};
 }
 
@@ -761,6 +829,7 @@ torch::Tensor attn_bwd_packed(torch::Tensor dout, torch::Tensor qkv,
   StridedMut dkm = dqm; dkm.p += D;
   StridedMut dvm = dqm; dvm.p += 2 * D;
   launch_bwd(q, k, v, dos, os, dqm, dkm, dvm, lse.data_ptr<float>(),
-             delta.data_ptr<float>(), B, H, S, D, (float)scale);
+             delta.data_ptr<float>(), B, H, S, D, (float)scale,
+             /*causal=*/true);
   return dqkv;
 }

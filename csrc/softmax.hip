@@ -32,7 +32,7 @@ __global__ void softmax_causal_fwd_kernel(const T* __restrict__ s,
   for (int i = threadIdx.x; i < valid; i += BLOCK) {
     float v = (sizeof(T) == 2) ? bf_raw2f(((const unsigned short*)sr)[i])
                                : ((const float*)sr)[i];
-    sum += expf(v * scale - mx);
+    sum += __expf(v * scale - mx);
   }
   sum = block_reduce_sum<BLOCK>(sum, sred);
   float inv = 1.f / sum;
@@ -41,7 +41,7 @@ __global__ void softmax_causal_fwd_kernel(const T* __restrict__ s,
     if (i < valid) {
       float v = (sizeof(T) == 2) ? bf_raw2f(((const unsigned short*)sr)[i])
                                  : ((const float*)sr)[i];
-      o = expf(v * scale - mx) * inv;
+      o = __expf(v * scale - mx) * inv;
     }
     if constexpr (sizeof(T) == 2)
       ((unsigned short*)yr)[i] = f2bf_raw(o);
@@ -104,7 +104,7 @@ __global__ void ce_fwd_kernel(const T* __restrict__ logits,
   for (int i = threadIdx.x; i < V; i += BLOCK) {
     float v = (sizeof(T) == 2) ? bf_raw2f(((const unsigned short*)lr)[i])
                                : ((const float*)lr)[i];
-    sum += expf(v - mx);
+    sum += __expf(v - mx);
   }
   sum = block_reduce_sum<BLOCK>(sum, sred);
   if (threadIdx.x == 0) {
@@ -139,7 +139,7 @@ __global__ void ce_bwd_kernel(const float* __restrict__ dloss,
   for (int i = threadIdx.x; i < V; i += BLOCK) {
     float v = (sizeof(T) == 2) ? bf_raw2f(((const unsigned short*)lr)[i])
                                : ((const float*)lr)[i];
-    float p = expf(v - l);
+    float p = __expf(v - l);
     float g = (p - (i == (int)lab ? 1.f : 0.f)) * dl;
     if constexpr (sizeof(T) == 2)
       ((unsigned short*)dr)[i] = f2bf_raw(g);
@@ -177,7 +177,7 @@ __global__ void row_sumexp_kernel(const T* __restrict__ x,
   for (int i = threadIdx.x; i < V; i += BLOCK) {
     float v = (sizeof(T) == 2) ? bf_raw2f(((const unsigned short*)xr)[i])
                                : ((const float*)xr)[i];
-    s += expf(v - m);
+    s += __expf(v - m);
   }
   s = block_reduce_sum<BLOCK>(s, sred);
   if (threadIdx.x == 0) out[row] = s;
@@ -221,7 +221,7 @@ __global__ void vp_ce_bwd_kernel(const float* __restrict__ dloss,
   for (int i = threadIdx.x; i < V; i += BLOCK) {
     float v = (sizeof(T) == 2) ? bf_raw2f(((const unsigned short*)lr)[i])
                                : ((const float*)lr)[i];
-    float p = expf(v - l);
+    float p = __expf(v - l);
     float g = (p - (i == (int)local ? 1.f : 0.f)) * dl;
     if constexpr (sizeof(T) == 2)
       ((unsigned short*)dr)[i] = f2bf_raw(g);

@@ -87,11 +87,19 @@ class ViTAttention(nn.Module):
         q = q.transpose(1, 2)
         k = k.transpose(1, 2)
         v = v.transpose(1, 2)
-        scores = torch.matmul(q, k.transpose(-1, -2)) * self.scale
-        probs = F.softmax(scores.float(), dim=-1).to(x.dtype)
-        if self.attn_drop_p > 0.0 and self.training:
-            probs = F.dropout(probs, p=self.attn_drop_p)
-        o = torch.matmul(probs, v).transpose(1, 2).reshape(B, N, C)
+        use_kernel = (x.is_cuda and x.dtype == torch.bfloat16
+                      and self.head_dim in (64, 128)
+                      and not (self.attn_drop_p > 0.0 and self.training))
+        if use_kernel:
+            from paddlefleetx_amd.ops import flash_attention
+            o = flash_attention(q, k, v, causal=False, scale=self.scale)
+            o = o.transpose(1, 2).reshape(B, N, C)
+        else:
+            scores = torch.matmul(q, k.transpose(-1, -2)) * self.scale
+            probs = F.softmax(scores.float(), dim=-1).to(x.dtype)
+            if self.attn_drop_p > 0.0 and self.training:
+                probs = F.dropout(probs, p=self.attn_drop_p)
+            o = torch.matmul(probs, v).transpose(1, 2).reshape(B, N, C)
         o = self.proj(o)
         if self.proj_drop_p > 0.0 and self.training:
             o = F.dropout(o, p=self.proj_drop_p)

@@ -163,6 +163,30 @@ def test_flash_attention_bwd(dev, cfg):
             f"{name}: max err {err}"
 
 
+@pytest.mark.parametrize("cfg", [
+    dict(B=2, H=4, S=257, D=64),    # ViT-base geometry (bidirectional)
+    dict(B=1, H=2, S=512, D=128),
+])
+def test_flash_attention_noncausal(dev, cfg):
+    B, H, S, D = cfg["B"], cfg["H"], cfg["S"], cfg["D"]
+    q = torch.randn(B, H, S, D, device=dev, dtype=torch.bfloat16) * 0.5
+    k = torch.randn_like(q) * 0.5
+    v = torch.randn_like(q)
+    do = torch.randn_like(q)
+    scale = 1.0 / math.sqrt(D)
+    o, lse = hip_ext().attn_fwd(q, k, v, False, scale)
+    o2, lse2 = ref.attention_fwd(q, k, v, False, scale)
+    assert torch.allclose(lse, lse2, atol=2e-2, rtol=1e-2)
+    assert torch.allclose(o.float(), o2.float(), atol=3e-2, rtol=3e-2), \
+        (o.float() - o2.float()).abs().max()
+    dq, dk, dv = hip_ext().attn_bwd(do, q, k, v, o, lse, False, scale)
+    dq2, dk2, dv2 = ref.attention_bwd(do, q, k, v, o, lse, False, scale)
+    for name, a, b in (("dv", dv, dv2), ("dk", dk, dk2), ("dq", dq, dq2)):
+        err = (a.float() - b.float()).abs().max()
+        assert torch.allclose(a.float(), b.float(), atol=8e-2, rtol=8e-2), \
+            f"{name}: max err {err}"
+
+
 def test_topp_sampling(dev):
     B, V = 8, 50304
     logits = torch.randn(B, V, device=dev)
