@@ -409,10 +409,13 @@ __global__ void attn_bwd_delta_kernel(Strided dout, Strided o,
 }
 
 // ===========================================================================
-// Backward dK/dV (transposed score space S^T = K Q^T)
+// Backward dK/dV v2 (transposed score space S^T = K Q^T).
+// 8 waves, 128 kv rows per block (16/wave); 64-row q tiles streamed with
+// issue-early register staging (guide T14): tile t+1's global loads are in
+// flight during tile t's MFMAs; single LDS buffer set, two barriers/tile.
 // ===========================================================================
 template <int D, bool CAUSAL>
-__global__ __launch_bounds__(BLOCKT) void attn_bwd_dkv_kernel(
+__global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dkv_kernel(
     Strided q, Strided k, Strided v, Strided dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
     StridedMut dk, StridedMut dv, int H, int S, float scale, int q_tiles) {
@@ -424,10 +427,10 @@ __global__ __launch_bounds__(BLOCKT) void attn_bwd_dkv_kernel(
   __shared__ unsigned short qt_lds[D * TS];
   __shared__ unsigned short do_lds[TILE * RS];
   __shared__ unsigned short dot_lds[D * TS];
-  __shared__ unsigned short p_lds[NWAVES * 16 * TS];
+  __shared__ unsigned short p_lds[FW_WAVES * 16 * TS];
   __shared__ float lsed_lds[2 * TILE];
 
-  const int kt = blockIdx.x;
+  const int kt = blockIdx.x;            // 128-row kv block
   const int b = blockIdx.y / H, hh = blockIdx.y % H;
   const long bh = blockIdx.y;
   const int wid = threadIdx.x / WAVE;
@@ -438,8 +441,7 @@ __global__ __launch_bounds__(BLOCKT) void attn_bwd_dkv_kernel(
   const __hip_bfloat16* vp = v.at(b, hh);
   const __hip_bfloat16* dop = dout.at(b, hh);
 
-  const int kv0 = kt * TILE;
-  const int kvrow0 = kv0 + wid * 16;
+  const int kvrow0 = kt * QTILE + wid * 16;
   bf8 kfrag[D / 32], vfrag[D / 32];
   load_a_frags<D>(kp, k.rs, kvrow0, S, lane, kfrag);
   load_a_frags<D>(vp, v.rs, kvrow0, S, lane, vfrag);
@@ -456,89 +458,149 @@ __global__ __launch_bounds__(BLOCKT) void attn_bwd_dkv_kernel(
   const int my_kvrow = kvrow0 + crow4;
   unsigned short* myp = p_lds + wid * 16 * TS;
 
-  for (int qt = CAUSAL ? kt : 0; qt < q_tiles; ++qt) {
+  // staging thread map: D/16 threads per q row, 16 shorts per tensor each
+  constexpr int TPR = D / 16;
+  const int st_row = threadIdx.x / TPR;
+  const int st_col = (threadIdx.x % TPR) * 16;
+  const bool st_on = st_row < TILE;
+  short8v qreg[2], dreg[2];
+  float lse_s = 0.f, dlt_s = 0.f;
+
+  const int qt_first = CAUSAL ? (kt * QTILE) / TILE : 0;
+
+  auto issue_loads = [&](int q0) {
+#pragma unroll
+    for (int h2 = 0; h2 < 2; ++h2) { qreg[h2] = short8v{}; dreg[h2] = short8v{}; }
+    const int row = q0 + st_row;
+    if (st_on && row < S) {
+      const unsigned short* qs = (const unsigned short*)qp + (long)row * q.rs + st_col;
+      const unsigned short* ds = (const unsigned short*)dop + (long)row * dout.rs + st_col;
+      qreg[0] = *reinterpret_cast<const short8v*>(qs);
+      qreg[1] = *reinterpret_cast<const short8v*>(qs + 8);
+      dreg[0] = *reinterpret_cast<const short8v*>(ds);
+      dreg[1] = *reinterpret_cast<const short8v*>(ds + 8);
+    }
+    const int li = threadIdx.x;
+    if (li < TILE) {
+      int qi = q0 + li;
+      lse_s = (qi < S) ? lse[bh * (long)S + qi] : 0.f;
+      dlt_s = (qi < S) ? delta[bh * (long)S + qi] : 0.f;
+    }
+  };
+  auto write_tiles = [&]() {
+    if (st_on) {
+      *reinterpret_cast<short8v*>(q_lds + st_row * RS + st_col) = qreg[0];
+      *reinterpret_cast<short8v*>(q_lds + st_row * RS + st_col + 8) = qreg[1];
+      *reinterpret_cast<short8v*>(do_lds + st_row * RS + st_col) = dreg[0];
+      *reinterpret_cast<short8v*>(do_lds + st_row * RS + st_col + 8) = dreg[1];
+      const unsigned short* qr = (const unsigned short*)&qreg[0];
+      const unsigned short* dr = (const unsigned short*)&dreg[0];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        qt_lds[(st_col + j) * TS + st_row] = qr[j];
+        dot_lds[(st_col + j) * TS + st_row] = dr[j];
+      }
+      qr = (const unsigned short*)&qreg[1];
+      dr = (const unsigned short*)&dreg[1];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        qt_lds[(st_col + 8 + j) * TS + st_row] = qr[j];
+        dot_lds[(st_col + 8 + j) * TS + st_row] = dr[j];
+      }
+    }
+    if (threadIdx.x < TILE) {
+      lsed_lds[threadIdx.x] = lse_s;
+      lsed_lds[TILE + threadIdx.x] = dlt_s;
+    }
+  };
+
+  issue_loads(qt_first * TILE);
+  write_tiles();
+  __syncthreads();
+
+  for (int qt = qt_first; qt < q_tiles; ++qt) {
     const int q0 = qt * TILE;
-    const int nvalid = min(TILE, S - q0);
-    __syncthreads();
-    stage_both<D>(qp + (long)q0 * q.rs, q.rs, nvalid, q_lds, qt_lds);
-    stage_both<D>(dop + (long)q0 * dout.rs, dout.rs, nvalid, do_lds, dot_lds);
-    for (int i = threadIdx.x; i < TILE; i += BLOCKT) {
-      int qi = q0 + i;
-      lsed_lds[i] = (qi < S) ? lse[bh * (long)S + qi] : 0.f;
-      lsed_lds[TILE + i] = (qi < S) ? delta[bh * (long)S + qi] : 0.f;
-    }
-    __syncthreads();
+    const bool have_next = (qt + 1) < q_tiles;
+    if (have_next) issue_loads(q0 + TILE);
 
-    // ---- S^T = K Q^T; P^T = exp(scale*S^T - lse[q]) ----
-    f4 pt[4];
+    // waves whose kv rows are entirely below this q tile have no overlap
+    const bool active = !CAUSAL || (q0 + TILE - 1 >= kvrow0);
+    f4 pt[4], dpt[4];
+    if (active) {
+      // ---- S^T = K Q^T; P^T = exp(scale*S^T - lse[q]) ----
 #pragma unroll
-    for (int ct = 0; ct < 4; ++ct) {
-      f4 acc = {0.f, 0.f, 0.f, 0.f};
+      for (int ct = 0; ct < 4; ++ct) {
+        f4 acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-      for (int kc = 0; kc < D / 32; ++kc) {
-        bf8 qb = read_b_frag(q_lds, RS, ct * 16, kc * 32, lane);
-        acc = MFMA_BF16(kfrag[kc], qb, acc);
+        for (int kc = 0; kc < D / 32; ++kc) {
+          bf8 qb = read_b_frag(q_lds, RS, ct * 16, kc * 32, lane);
+          acc = MFMA_BF16(kfrag[kc], qb, acc);
+        }
+        int qcol = q0 + ct * 16 + ccol;
+        float l = lsed_lds[ct * 16 + ccol];
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int kvr = my_kvrow + r;
+          float p = 0.f;
+          if ((!CAUSAL || qcol >= kvr) && qcol < S && kvr < S)
+            p = __expf(acc[r] * scale - l);
+          pt[ct][r] = p;
+        }
       }
-      int qcol = q0 + ct * 16 + ccol;
-      float l = lsed_lds[ct * 16 + ccol];
+
+      // ---- stage P^T; dV += P^T dO ----
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        int kvr = my_kvrow + r;
-        float p = 0.f;
-        if ((!CAUSAL || qcol >= kvr) && qcol < S && kvr < S)
-          p = __expf(acc[r] * scale - l);
-        pt[ct][r] = p;
+      for (int ct = 0; ct < 4; ++ct)
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          myp[(crow4 + r) * TS + ct * 16 + ccol] = f2bf_raw(pt[ct][r]);
+#pragma unroll
+      for (int kc = 0; kc < 2; ++kc) {
+        bf8 pa = read_a_frag_lds(myp, TS, kc * 32, lane);
+#pragma unroll
+        for (int dt = 0; dt < NDT; ++dt) {
+          bf8 db = read_b_frag(dot_lds, TS, dt * 16, kc * 32, lane);
+          dvacc[dt] = MFMA_BF16(pa, db, dvacc[dt]);
+        }
+      }
+
+      // ---- dP^T = V dO^T ----
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct) {
+        f4 acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int kc = 0; kc < D / 32; ++kc) {
+          bf8 db = read_b_frag(do_lds, RS, ct * 16, kc * 32, lane);
+          acc = MFMA_BF16(vfrag[kc], db, acc);
+        }
+        dpt[ct] = acc;
+      }
+
+      // ---- dS^T -> LDS; dK += dS^T Q ----
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct) {
+        float dlt = lsed_lds[TILE + ct * 16 + ccol];
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float ds = pt[ct][r] * (dpt[ct][r] - dlt) * scale;
+          myp[(crow4 + r) * TS + ct * 16 + ccol] = f2bf_raw(ds);
+        }
+      }
+#pragma unroll
+      for (int kc = 0; kc < 2; ++kc) {
+        bf8 dsa = read_a_frag_lds(myp, TS, kc * 32, lane);
+#pragma unroll
+        for (int dt = 0; dt < NDT; ++dt) {
+          bf8 qb = read_b_frag(qt_lds, TS, dt * 16, kc * 32, lane);
+          dkacc[dt] = MFMA_BF16(dsa, qb, dkacc[dt]);
+        }
       }
     }
 
-    // ---- stage P^T; dV += P^T dO ----
-#pragma unroll
-    for (int ct = 0; ct < 4; ++ct)
-#pragma unroll
-      for (int r = 0; r < 4; ++r)
-        myp[(crow4 + r) * TS + ct * 16 + ccol] = f2bf_raw(pt[ct][r]);
-#pragma unroll
-    for (int kc = 0; kc < 2; ++kc) {
-      bf8 pa = read_a_frag_lds(myp, TS, kc * 32, lane);
-#pragma unroll
-      for (int dt = 0; dt < NDT; ++dt) {
-        bf8 db = read_b_frag(dot_lds, TS, dt * 16, kc * 32, lane);
-        dvacc[dt] = MFMA_BF16(pa, db, dvacc[dt]);
-      }
-    }
-
-    // ---- dP^T = V dO^T ----
-    f4 dpt[4];
-#pragma unroll
-    for (int ct = 0; ct < 4; ++ct) {
-      f4 acc = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-      for (int kc = 0; kc < D / 32; ++kc) {
-        bf8 db = read_b_frag(do_lds, RS, ct * 16, kc * 32, lane);
-        acc = MFMA_BF16(vfrag[kc], db, acc);
-      }
-      dpt[ct] = acc;
-    }
-
-    // ---- dS^T -> LDS; dK += dS^T Q ----
-    __syncthreads();
-#pragma unroll
-    for (int ct = 0; ct < 4; ++ct) {
-      float dlt = lsed_lds[TILE + ct * 16 + ccol];
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        float ds = pt[ct][r] * (dpt[ct][r] - dlt) * scale;
-        myp[(crow4 + r) * TS + ct * 16 + ccol] = f2bf_raw(ds);
-      }
-    }
-#pragma unroll
-    for (int kc = 0; kc < 2; ++kc) {
-      bf8 dsa = read_a_frag_lds(myp, TS, kc * 32, lane);
-#pragma unroll
-      for (int dt = 0; dt < NDT; ++dt) {
-        bf8 qb = read_b_frag(qt_lds, TS, dt * 16, kc * 32, lane);
-        dkacc[dt] = MFMA_BF16(dsa, qb, dkacc[dt]);
-      }
+    if (have_next) {
+      __syncthreads();   // all waves done reading this tile's LDS
+      write_tiles();
+      __syncthreads();   // next tile visible
     }
   }
 
@@ -559,23 +621,28 @@ __global__ __launch_bounds__(BLOCKT) void attn_bwd_dkv_kernel(
 }
 
 // ===========================================================================
-// Backward dQ
+// Backward dQ v2: 8 waves, 128 q rows per block; 64-row kv tiles
+// double-buffered in LDS with issue-early staging, one barrier per tile.
 // ===========================================================================
 template <int D, bool CAUSAL>
-__global__ __launch_bounds__(BLOCKT) void attn_bwd_dq_kernel(
+__global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dq_kernel(
     Strided q, Strided k, Strided v, Strided dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
-    StridedMut dq, int H, int S, float scale, int q_tiles) {
+    StridedMut dq, int H, int S, float scale, int kv_total) {
   constexpr int RS = D + PAD;
   constexpr int TS = TILE + PAD;
   constexpr int NDT = D / 16;
+  constexpr int KSZ = TILE * RS;   // K row-major buffer
+  constexpr int TSZ = D * TS;      // K^T buffer
+  constexpr int VSZ = TILE * RS;   // V row-major buffer
 
-  __shared__ unsigned short k_lds[TILE * RS];
-  __shared__ unsigned short kt_lds[D * TS];
-  __shared__ unsigned short v_lds[TILE * RS];
-  __shared__ unsigned short p_lds[NWAVES * 16 * TS];
+  __shared__ unsigned short smem[2 * (KSZ + TSZ + VSZ) + FW_WAVES * 16 * TS];
+  unsigned short* k_lds = smem;
+  unsigned short* kt_lds = smem + 2 * KSZ;
+  unsigned short* v_lds = smem + 2 * (KSZ + TSZ);
+  unsigned short* p_lds = smem + 2 * (KSZ + TSZ + VSZ);
 
-  const int qt = blockIdx.x;
+  const int qt = blockIdx.x;   // 128-row q block
   const int b = blockIdx.y / H, hh = blockIdx.y % H;
   const long bh = blockIdx.y;
   const int wid = threadIdx.x / WAVE;
@@ -586,7 +653,7 @@ __global__ __launch_bounds__(BLOCKT) void attn_bwd_dq_kernel(
   const __hip_bfloat16* vp = v.at(b, hh);
   const __hip_bfloat16* dop = dout.at(b, hh);
 
-  const int qrow0 = qt * TILE + wid * 16;
+  const int qrow0 = qt * QTILE + wid * 16;
   bf8 qfrag[D / 32], dofrag[D / 32];
   load_a_frags<D>(qp, q.rs, qrow0, S, lane, qfrag);
   load_a_frags<D>(dop, dout.rs, qrow0, S, lane, dofrag);
@@ -598,6 +665,7 @@ __global__ __launch_bounds__(BLOCKT) void attn_bwd_dq_kernel(
   const int ccol = lane & 15;
   const int crow4 = (lane >> 4) * 4;
   const int my_qrow = qrow0 + crow4;
+  const int wave_last_row = min(qrow0 + 15, S - 1);
   unsigned short* myp = p_lds + wid * 16 * TS;
 
   float lse_r[4], dlt_r[4];
@@ -608,65 +676,117 @@ __global__ __launch_bounds__(BLOCKT) void attn_bwd_dq_kernel(
     dlt_r[r] = (qr < S) ? delta[bh * (long)S + qr] : 0.f;
   }
 
-  const int kv_tiles = CAUSAL ? min(q_tiles, qt + 1) : q_tiles;
+  const int kv_tiles = CAUSAL
+      ? min(kv_total, (qt * QTILE + QTILE - 1) / TILE + 1)
+      : kv_total;
+
+  // staging map: D/16 threads per kv row, 16 shorts per tensor each
+  constexpr int TPR = D / 16;
+  const int st_row = threadIdx.x / TPR;
+  const int st_col = (threadIdx.x % TPR) * 16;
+  const bool st_on = st_row < TILE;
+  short8v kreg[2], vreg[2];
+
+  auto issue_loads = [&](int kv0) {
+#pragma unroll
+    for (int h2 = 0; h2 < 2; ++h2) { kreg[h2] = short8v{}; vreg[h2] = short8v{}; }
+    const int row = kv0 + st_row;
+    if (st_on && row < S) {
+      const unsigned short* ks = (const unsigned short*)kp + (long)row * k.rs + st_col;
+      const unsigned short* vs = (const unsigned short*)vp + (long)row * v.rs + st_col;
+      kreg[0] = *reinterpret_cast<const short8v*>(ks);
+      kreg[1] = *reinterpret_cast<const short8v*>(ks + 8);
+      vreg[0] = *reinterpret_cast<const short8v*>(vs);
+      vreg[1] = *reinterpret_cast<const short8v*>(vs + 8);
+    }
+  };
+  auto write_tiles = [&](int buf) {
+    if (!st_on) return;
+    unsigned short* kd = k_lds + buf * KSZ;
+    unsigned short* ktd = kt_lds + buf * TSZ;
+    unsigned short* vd = v_lds + buf * VSZ;
+    *reinterpret_cast<short8v*>(kd + st_row * RS + st_col) = kreg[0];
+    *reinterpret_cast<short8v*>(kd + st_row * RS + st_col + 8) = kreg[1];
+    *reinterpret_cast<short8v*>(vd + st_row * RS + st_col) = vreg[0];
+    *reinterpret_cast<short8v*>(vd + st_row * RS + st_col + 8) = vreg[1];
+    const unsigned short* kr = (const unsigned short*)&kreg[0];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) ktd[(st_col + j) * TS + st_row] = kr[j];
+    kr = (const unsigned short*)&kreg[1];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) ktd[(st_col + 8 + j) * TS + st_row] = kr[j];
+  };
+
+  issue_loads(0);
+  write_tiles(0);
+  __syncthreads();
+
   for (int ktl = 0; ktl < kv_tiles; ++ktl) {
     const int kv0 = ktl * TILE;
-    const int nvalid = min(TILE, S - kv0);
+    const int cur = ktl & 1;
+    const unsigned short* kb_lds = k_lds + cur * KSZ;
+    const unsigned short* ktb_lds = kt_lds + cur * TSZ;
+    const unsigned short* vb_lds = v_lds + cur * VSZ;
+
+    const bool have_next = (ktl + 1) < kv_tiles;
+    if (have_next) issue_loads(kv0 + TILE);
+
+    const bool active = !CAUSAL || (kv0 <= wave_last_row);
+    if (active) {
+      // ---- S = Q K^T; P ----
+      f4 p[4];
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct) {
+        f4 acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int kc = 0; kc < D / 32; ++kc) {
+          bf8 kb = read_b_frag(kb_lds, RS, ct * 16, kc * 32, lane);
+          acc = MFMA_BF16(qfrag[kc], kb, acc);
+        }
+        int kcol = kv0 + ct * 16 + ccol;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float pv = 0.f;
+          if ((!CAUSAL || kcol <= my_qrow + r) && kcol < S && my_qrow + r < S)
+            pv = __expf(acc[r] * scale - lse_r[r]);
+          p[ct][r] = pv;
+        }
+      }
+
+      // ---- dP = dO V^T ----
+      f4 dp[4];
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct) {
+        f4 acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int kc = 0; kc < D / 32; ++kc) {
+          bf8 vb = read_b_frag(vb_lds, RS, ct * 16, kc * 32, lane);
+          acc = MFMA_BF16(dofrag[kc], vb, acc);
+        }
+        dp[ct] = acc;
+      }
+
+      // ---- dS -> LDS; dQ += dS K ----
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float ds = p[ct][r] * (dp[ct][r] - dlt_r[r]) * scale;
+          myp[(crow4 + r) * TS + ct * 16 + ccol] = f2bf_raw(ds);
+        }
+#pragma unroll
+      for (int kc = 0; kc < 2; ++kc) {
+        bf8 dsa = read_a_frag_lds(myp, TS, kc * 32, lane);
+#pragma unroll
+        for (int dt = 0; dt < NDT; ++dt) {
+          bf8 kb = read_b_frag(ktb_lds, TS, dt * 16, kc * 32, lane);
+          dqacc[dt] = MFMA_BF16(dsa, kb, dqacc[dt]);
+        }
+      }
+    }
+
+    if (have_next) write_tiles(cur ^ 1);
     __syncthreads();
-    stage_both<D>(kp + (long)kv0 * k.rs, k.rs, nvalid, k_lds, kt_lds);
-    stage_rowmajor<D>(vp + (long)kv0 * v.rs, v.rs, nvalid, v_lds);
-    __syncthreads();
-
-    // ---- S = Q K^T; P ----
-    f4 p[4];
-#pragma unroll
-    for (int ct = 0; ct < 4; ++ct) {
-      f4 acc = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-      for (int kc = 0; kc < D / 32; ++kc) {
-        bf8 kb = read_b_frag(k_lds, RS, ct * 16, kc * 32, lane);
-        acc = MFMA_BF16(qfrag[kc], kb, acc);
-      }
-      int kcol = kv0 + ct * 16 + ccol;
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        float pv = 0.f;
-        if ((!CAUSAL || kcol <= my_qrow + r) && kcol < S && my_qrow + r < S)
-          pv = __expf(acc[r] * scale - lse_r[r]);
-        p[ct][r] = pv;
-      }
-    }
-
-    // ---- dP = dO V^T ----
-    f4 dp[4];
-#pragma unroll
-    for (int ct = 0; ct < 4; ++ct) {
-      f4 acc = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-      for (int kc = 0; kc < D / 32; ++kc) {
-        bf8 vb = read_b_frag(v_lds, RS, ct * 16, kc * 32, lane);
-        acc = MFMA_BF16(dofrag[kc], vb, acc);
-      }
-      dp[ct] = acc;
-    }
-
-    // ---- dS -> LDS; dQ += dS K ----
-#pragma unroll
-    for (int ct = 0; ct < 4; ++ct)
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        float ds = p[ct][r] * (dp[ct][r] - dlt_r[r]) * scale;
-        myp[(crow4 + r) * TS + ct * 16 + ccol] = f2bf_raw(ds);
-      }
-#pragma unroll
-    for (int kc = 0; kc < 2; ++kc) {
-      bf8 dsa = read_a_frag_lds(myp, TS, kc * 32, lane);
-#pragma unroll
-      for (int dt = 0; dt < NDT; ++dt) {
-        bf8 kb = read_b_frag(kt_lds, TS, dt * 16, kc * 32, lane);
-        dqacc[dt] = MFMA_BF16(dsa, kb, dqacc[dt]);
-      }
-    }
   }
 
   __hip_bfloat16* dqp = dq.at(b, hh);
@@ -728,16 +848,17 @@ void launch_bwd(Strided q, Strided k, Strided v, Strided dout, Strided o,
   hipLaunchKernelGGL(attn_bwd_delta_kernel,
                      dim3((rows + (BLOCKT / WAVE) - 1) / (BLOCKT / WAVE)),
                      dim3(BLOCKT), 0, stream, dout, o, delta, H, S, D);
-  int q_tiles = (S + TILE - 1) / TILE;
-  dim3 grid(q_tiles, B * H);
+  int tiles64 = (S + TILE - 1) / TILE;       // inner streamed tiles
+  int blocks128 = (S + QTILE - 1) / QTILE;   // per-block home tile
+  dim3 grid(blocks128, B * H);
 #define LAUNCH_BWD(DD, CC)                                                   \
   do {                                                                       \
-    hipLaunchKernelGGL((attn_bwd_dkv_kernel<DD, CC>), grid, dim3(BLOCKT), 0, \
-                       stream, q, k, v, dout, lse, delta, dk, dv, H, S,      \
-                       scale, q_tiles);                                      \
-    hipLaunchKernelGGL((attn_bwd_dq_kernel<DD, CC>), grid, dim3(BLOCKT), 0,  \
-                       stream, q, k, v, dout, lse, delta, dq, H, S, scale,   \
-                       q_tiles);                                             \
+    hipLaunchKernelGGL((attn_bwd_dkv_kernel<DD, CC>), grid,                  \
+                       dim3(FW_BLOCKT), 0, stream, q, k, v, dout, lse,       \
+                       delta, dk, dv, H, S, scale, tiles64);                 \
+    hipLaunchKernelGGL((attn_bwd_dq_kernel<DD, CC>), grid, dim3(FW_BLOCKT),  \
+                       0, stream, q, k, v, dout, lse, delta, dq, H, S,       \
+                       scale, tiles64);                                      \
   } while (0)
   if (D == 128) { if (causal) LAUNCH_BWD(128, true); else LAUNCH_BWD(128, false); }
   else          { if (causal) LAUNCH_BWD(64, true);  else LAUNCH_BWD(64, false); }
