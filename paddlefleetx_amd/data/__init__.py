@@ -10,6 +10,7 @@ from torch.utils.data import DataLoader
 
 from paddlefleetx_amd.data.gpt_dataset import GPTDataset, GPTSyntheticDataset
 from paddlefleetx_amd.data.sampler import GPTBatchSampler
+from paddlefleetx_amd.data.ernie_dataset import ErnieSyntheticDataset
 from paddlefleetx_amd.data.vision_dataset import (ImageFolderDataset,
                                                   SyntheticImageNetDataset)
 from paddlefleetx_amd.parallel.env import (get_data_world_rank,
@@ -22,6 +23,7 @@ _DATASETS = {
     "SyntheticImageNetDataset": SyntheticImageNetDataset,
     "ImageFolderDataset": ImageFolderDataset,
     "GeneralClsDataset": ImageFolderDataset,
+    "ErnieSyntheticDataset": ErnieSyntheticDataset,
 }
 
 

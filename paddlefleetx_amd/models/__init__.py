@@ -26,6 +26,11 @@ def build_module(config):
         from paddlefleetx_amd.models.vit_module import GeneralClsModule
         table["ViTModule"] = GeneralClsModule
         table["GeneralClsModule"] = GeneralClsModule
+    if name in ("ErnieModule", "ErnieSeqClsModule"):
+        from paddlefleetx_amd.models.ernie_module import (ErnieModule,
+                                                          ErnieSeqClsModule)
+        table["ErnieModule"] = ErnieModule
+        table["ErnieSeqClsModule"] = ErnieSeqClsModule
     if name not in table:
         raise ValueError(f"unknown module {name}")
     logger.info(f"building module {name}")
