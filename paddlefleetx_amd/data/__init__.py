@@ -27,6 +27,20 @@ _DATASETS = {
 }
 
 
+def _register_lazy():
+    from paddlefleetx_amd.data.eval_dataset import (Lambada_Eval_Dataset,
+                                                    LM_Eval_Dataset)
+    from paddlefleetx_amd.data.glue_dataset import (GLUEDataset,
+                                                    SyntheticGLUEDataset)
+    _DATASETS.setdefault("LM_Eval_Dataset", LM_Eval_Dataset)
+    _DATASETS.setdefault("Lambada_Eval_Dataset", Lambada_Eval_Dataset)
+    _DATASETS.setdefault("GLUEDataset", GLUEDataset)
+    _DATASETS.setdefault("SyntheticGLUEDataset", SyntheticGLUEDataset)
+
+
+_register_lazy()
+
+
 def register_dataset(name, cls):
     _DATASETS[name] = cls
 

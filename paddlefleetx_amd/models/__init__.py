@@ -9,11 +9,13 @@ def build_module(config):
     # vision configs name the module under Model.module (vit base.yaml),
     # language configs under Model.name
     name = config["Model"].get("module") or config["Model"]["name"]
-    from paddlefleetx_amd.models.language_module import (GPTGenerationModule,
+    from paddlefleetx_amd.models.language_module import (GPTFinetuneModule,
+                                                         GPTGenerationModule,
                                                          GPTModule)
     table = {
         "GPTModule": GPTModule,
         "GPTGenerationModule": GPTGenerationModule,
+        "GPTFinetuneModule": GPTFinetuneModule,
     }
     # late registrations to avoid importing every family eagerly
     if name == "MoEModule":

@@ -360,6 +360,29 @@ class GPTForPretraining(nn.Module):
         return logits
 
 
+class GPTForSequenceClassification(nn.Module):
+    """GPT + classification head on the last real token
+    (single_model.py:856-897)."""
+
+    def __init__(self, gpt: GPTModel, num_classes: int = 2):
+        super().__init__()
+        self.gpt = gpt
+        hidden = gpt.final_ln.weight.shape[0]
+        self.score = nn.Linear(hidden, num_classes, bias=False,
+                               dtype=gpt.final_ln.weight.dtype)
+
+    def forward(self, input_ids, attention_mask=None, position_ids=None):
+        hidden = self.gpt(input_ids, position_ids)
+        logits = self.score(hidden)  # [B, S, C]
+        if attention_mask is not None:
+            last = attention_mask.long().sum(dim=-1).clamp(min=1) - 1
+        else:
+            last = torch.full((input_ids.shape[0],), input_ids.shape[1] - 1,
+                              device=input_ids.device, dtype=torch.long)
+        return logits[torch.arange(logits.shape[0], device=logits.device),
+                      last]
+
+
 class GPTPretrainingCriterion(nn.Module):
     """(Parallel) softmax CE with loss mask (hybrid_model.py:943-996)."""
 

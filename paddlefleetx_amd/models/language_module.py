@@ -104,6 +104,70 @@ class GPTModule(LanguageModule):
         return batch
 
 
+class GPTFinetuneModule(LanguageModule):
+    """GLUE finetuning (reference language_module.py:228-488): GPT backbone
+    + sequence-classification head, per-task metric, CE or MSE loss."""
+
+    def __init__(self, configs):
+        from paddlefleetx_amd.data.glue_dataset import GLUE_METRICS
+        self.task = configs["Model"].get("task", "sst2").lower()
+        metric_name = configs["Model"].get("metric", None) or \
+            GLUE_METRICS.get(self.task, "Accuracy")
+        from paddlefleetx_amd.models import metrics as M
+        self.metric = getattr(M, metric_name)()
+        self.regression = self.task == "stsb"
+        super().__init__(configs)
+
+    def get_model(self):
+        cfg = self.configs
+        mcfg = dict(cfg["Model"])
+        for k in ("name", "task", "metric", "num_classes",
+                  "vocab_size_divisible_unit", "moe_configs"):
+            mcfg.pop(k, None)
+        hcg = get_hcg()
+        mcfg["vocab_size"] = vocab_size_with_padding(
+            mcfg.get("vocab_size", 50304),
+            cfg["Model"].get("vocab_size_divisible_unit", 128),
+            hcg.get_model_parallel_world_size())
+        num_classes = int(cfg["Model"].get("num_classes",
+                                           1 if self.regression else 2))
+        from paddlefleetx_amd.models.gpt.model import (
+            GPTForSequenceClassification, GPTModel)
+        gpt = GPTModel(dtype=_model_dtype(cfg), **mcfg)
+        return GPTForSequenceClassification(gpt, num_classes=num_classes)
+
+    def get_loss_fn(self):
+        if self.regression:
+            return torch.nn.MSELoss()
+        return torch.nn.CrossEntropyLoss()
+
+    def training_step(self, batch):
+        ids, mask, labels = batch[:3]
+        logits = self.model(ids, attention_mask=mask)
+        if self.regression:
+            return self.loss_fn(logits.float().squeeze(-1), labels.float())
+        return self.loss_fn(logits.float(), labels)
+
+    def validation_step(self, batch):
+        ids, mask, labels = batch[:3]
+        logits = self.model(ids, attention_mask=mask)
+        if self.regression:
+            self.metric.update(logits.squeeze(-1), labels)
+            return self.loss_fn(logits.float().squeeze(-1), labels.float())
+        self.metric.update(logits, labels)
+        return self.loss_fn(logits.float(), labels)
+
+    def validation_step_end(self, log_dict):
+        vals = self.metric.accumulate()
+        names = self.metric.name()
+        if not isinstance(vals, tuple):
+            vals, names = (vals,), (names,)
+        stats = ", ".join(f"{n}: {v:.4f}" for n, v in zip(names, vals))
+        logger.eval("[eval] epoch: %d, batch: %d, loss: %.9f, %s"
+                    % (log_dict["epoch"], log_dict["batch"],
+                       log_dict["loss"], stats))
+
+
 class GPTGenerationModule(BasicModule):
     """Text generation (reference language_module.py:490-598)."""
 
