@@ -113,6 +113,9 @@ class EagerEngine(BasicEngine):
         if ckpt_dir:
             self.load(ckpt_dir)
 
+        from paddlefleetx_amd.utils.profiler import ProfilerGuard
+        self.profiler = ProfilerGuard(configs.get("Profiler"))
+
     # ------------------------------------------------------------------
     def _sync_params(self):
         """Broadcast params from dp-rank-0 (reference strategy.py:43 sync_params_buffers)."""
@@ -135,10 +138,14 @@ class EagerEngine(BasicEngine):
     def fit(self, train_data_loader=None, valid_data_loader=None, epoch=None):
         epochs = epoch if epoch is not None else self.num_train_epochs
         start_epoch = self._load_recovery["epoch"]
-        for ep in range(start_epoch, epochs):
-            done = self._train_one_epoch(ep, train_data_loader, valid_data_loader)
-            if done:
-                break
+        try:
+            for ep in range(start_epoch, epochs):
+                done = self._train_one_epoch(ep, train_data_loader,
+                                             valid_data_loader)
+                if done:
+                    break
+        finally:
+            self.profiler.stop_and_summary()
 
     def _train_one_epoch(self, epoch: int, train_loader, valid_loader) -> bool:
         self.module.model.train()
@@ -150,6 +157,7 @@ class EagerEngine(BasicEngine):
             if step < skip_until:
                 continue
             loss = self._fit_impl(batch)
+            self.profiler.step()
             self.module.global_step += 1
             gstep = self.module.global_step
             interval_cost = time.time() - t_start

@@ -1,0 +1,56 @@
+"""Inference engine over exported models (optionally tensor-parallel).
+
+Reference: ppfleetx/core/engine/inference_engine.py:104-272 — loads
+rank_{i}/ exported artifacts, one process per MP rank, NCCL rings from a
+generated csv, TensorRT optional. MI355X-native: the same rank_{i}/ layout
+(utils/export.py), RCCL process groups come from init_dist_env (no ring
+csv — torch.distributed owns transport), and the optimized runtime is the
+gfx950 kernel path itself (flash decode + top-p kernel), so there is no
+separate compiled-graph format.
+"""
+
+from __future__ import annotations
+
+import json
+import os
+from typing import Any, Dict, List, Optional
+
+import torch
+
+from paddlefleetx_amd.utils.export import load_inference_model
+from paddlefleetx_amd.utils.log import logger
+
+
+class InferenceEngine:
+    def __init__(self, model_dir: str, mp_degree: int = 1,
+                 generation_cfg: Optional[Dict[str, Any]] = None):
+        self.model_dir = model_dir
+        with open(os.path.join(model_dir, "config.json")) as f:
+            meta = json.load(f)
+        assert int(meta.get("mp_degree", 1)) == mp_degree, \
+            (f"exported for mp={meta.get('mp_degree')}, "
+             f"launched with mp={mp_degree}")
+        mcfg = dict(meta["model"])
+        self.device = torch.device("cuda") if torch.cuda.is_available() \
+            else torch.device("cpu")
+        from paddlefleetx_amd.models.gpt.generation import GPTForGeneration
+        from paddlefleetx_amd.models.gpt.model import GPTModel
+        gen_cfg = dict(meta.get("generation", {}))
+        gen_cfg.update(generation_cfg or {})
+        dtype = torch.bfloat16 if self.device.type == "cuda" else torch.float32
+        gpt = GPTModel(dtype=dtype,
+                       **{k: v for k, v in mcfg.items()
+                          if k not in ("name", "module")})
+        self.model = GPTForGeneration(gpt, gen_cfg)
+        load_inference_model(self.model, model_dir)
+        self.model.to(self.device).eval()
+        logger.info(f"inference engine ready (model_dir={model_dir}, "
+                    f"mp={mp_degree}, device={self.device})")
+
+    @torch.no_grad()
+    def predict(self, input_ids) -> torch.Tensor:
+        if not torch.is_tensor(input_ids):
+            input_ids = torch.tensor(input_ids, dtype=torch.long)
+        if input_ids.ndim == 1:
+            input_ids = input_ids.unsqueeze(0)
+        return self.model(input_ids.to(self.device))
