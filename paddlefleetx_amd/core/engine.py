@@ -98,11 +98,27 @@ class EagerEngine(BasicEngine):
         if mode == "train":
             sharding_group = self.hcg.get_sharding_parallel_group() \
                 if self.sharding_degree > 1 else None
-            self.optimizer = build_optimizer(
-                opt_cfg, self.module.model,
-                lr_value=self.lr_scheduler.get_lr(),
-                sharding_group=sharding_group,
-                sharding_stage=self.sharding_stage)
+            if self.sharding_stage == 3 and self.sharding_degree > 1:
+                # ZeRO-3: wrap the model for parameter sharding
+                # (reference group_sharded_parallel level="p_g_os",
+                # eager_engine.py:281-307)
+                from paddlefleetx_amd.parallel.zero3 import (
+                    GroupShardedStage3, Stage3AdamW)
+                assert not self.is_pipeline, \
+                    "sharding stage 3 excludes pipeline parallel " \
+                    "(reference eager_engine.py:276)"
+                self.module.model = GroupShardedStage3(
+                    self.module.model, group=sharding_group)
+                ocfg = {k: v for k, v in opt_cfg.items()
+                        if k in ("weight_decay", "beta1", "beta2", "epsilon")}
+                self.optimizer = Stage3AdamW(
+                    self.module.model, lr=self.lr_scheduler.get_lr(), **ocfg)
+            else:
+                self.optimizer = build_optimizer(
+                    opt_cfg, self.module.model,
+                    lr_value=self.lr_scheduler.get_lr(),
+                    sharding_group=sharding_group,
+                    sharding_stage=self.sharding_stage)
             # broadcast initial params across dp (and sharding) so replicas agree
             self._sync_params()
         else:
@@ -119,6 +135,16 @@ class EagerEngine(BasicEngine):
     # ------------------------------------------------------------------
     def _sync_params(self):
         """Broadcast params from dp-rank-0 (reference strategy.py:43 sync_params_buffers)."""
+        from paddlefleetx_amd.parallel.zero3 import Stage3AdamW
+        if isinstance(self.optimizer, Stage3AdamW):
+            # stage 3: shards live per rank; sync shards over dp only
+            dp = self.hcg.get_data_parallel_group()
+            if dp.world_size > 1 and dist.is_initialized():
+                for u, st in zip(self.module.model.units,
+                                 self.optimizer.state):
+                    dist.broadcast(u.shard, src=dp.ranks[0], group=dp.group)
+                    st["master"].copy_(u.shard.float())
+            return
         dp = self.hcg.get_data_parallel_group()
         sd = self.hcg.get_sharding_parallel_group()
         for g in (dp, sd):
@@ -241,13 +267,14 @@ class EagerEngine(BasicEngine):
             self.optimizer.step(lr=self.lr_scheduler.get_lr())
             self.optimizer.zero_grad()
         else:
-            # plain torch optimizer path (incl. ShardedOptimizer which mimics it)
+            # plain torch optimizer path (incl. Stage3AdamW which mimics it)
             if hasattr(self.optimizer, "reduce_and_step"):
                 self.lr_scheduler.step()
                 self.optimizer.reduce_and_step(
                     lr=self.lr_scheduler.get_lr(),
                     grad_clip=self.grad_clip_norm,
-                    loss_scale=self.loss_scale)
+                    loss_scale=self.loss_scale,
+                    dp_group=self.hcg.get_data_parallel_group())
             else:
                 self._allreduce_plain_grads()
                 if self.grad_clip_norm:
