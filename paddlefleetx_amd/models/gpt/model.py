@@ -86,6 +86,22 @@ class MultiHeadAttention(nn.Module):
                 use_cache: bool = False
                 ) -> Tuple[torch.Tensor, Optional[KVCache]]:
         qkv = self.qkv(x)
+        cp = get_hcg().get_context_parallel_world_size()
+        if (cp > 1 and cache is None and not use_cache
+                and not self.sequence_parallel):
+            # Ulysses CP: x is the [B, S/cp, H] sequence shard; all-to-all
+            # inside attention trades seq for heads (parallel/cp.py)
+            assert self.attn_dropout_p == 0.0, \
+                "attention dropout unsupported under context parallel"
+            assert self.num_heads_local % cp == 0, \
+                f"heads/mp {self.num_heads_local} not divisible by cp {cp}"
+            from paddlefleetx_amd.parallel.cp import UlyssesAttention
+            B, Sl, _ = qkv.shape
+            qkv = qkv.view(B, Sl, self.num_heads_local, 3 * self.head_dim)
+            q, k, v = qkv.split(self.head_dim, dim=-1)  # [B, S/cp, h, D]
+            ua = UlyssesAttention(scale=self.scale, causal=True)
+            o = ua(q, k, v).reshape(B, Sl, -1)
+            return self.out_proj(o), None
         if (self.fused_attn and self.attn_dropout_p == 0.0 and cache is None
                 and not use_cache and not self.sequence_parallel):
             # packed fast path: kernel reads the fused-QKV linear output
@@ -384,7 +400,9 @@ class GPTForSequenceClassification(nn.Module):
 
 
 class GPTPretrainingCriterion(nn.Module):
-    """(Parallel) softmax CE with loss mask (hybrid_model.py:943-996)."""
+    """(Parallel) softmax CE with loss mask (hybrid_model.py:943-996).
+    Under context parallel each rank scores its sequence shard and the
+    masked sum/count are all-reduced so every rank sees the global loss."""
 
     def __init__(self, sequence_parallel: bool = False):
         super().__init__()
@@ -392,7 +410,15 @@ class GPTPretrainingCriterion(nn.Module):
 
     def forward(self, logits, labels, loss_mask=None):
         losses = self.ce(logits, labels)  # [N] fp32
+        cp = get_hcg().get_context_parallel_world_size()
         if loss_mask is not None:
             m = loss_mask.reshape(-1).float()
-            return (losses * m).sum() / m.sum().clamp(min=1.0)
-        return losses.mean()
+            num, den = (losses * m).sum(), m.sum()
+        else:
+            num = losses.sum()
+            den = torch.tensor(float(losses.numel()), device=losses.device)
+        if cp > 1:
+            from paddlefleetx_amd.parallel.cp import cp_allreduce_sum
+            num = cp_allreduce_sum(num)
+            den = cp_allreduce_sum(den)
+        return num / den.clamp(min=1.0)

@@ -101,6 +101,15 @@ class GPTModule(LanguageModule):
         return GPTPretrainingCriterion()
 
     def pretreating_batch(self, batch):
+        hcg = get_hcg()
+        cp = hcg.get_context_parallel_world_size()
+        if cp > 1:
+            # each cp rank takes its sequence chunk (Ulysses sharding)
+            r = hcg.get_context_parallel_rank()
+            batch = tuple(
+                torch.chunk(t, cp, dim=1)[r].contiguous()
+                if torch.is_tensor(t) and t.ndim >= 2 else t
+                for t in batch)
         return batch
 
 

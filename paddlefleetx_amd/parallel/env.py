@@ -72,6 +72,7 @@ def init_dist_env(config) -> HybridTopology:
         mp=int(d.get("mp_degree", 1)),
         pp=int(d.get("pp_degree", 1)),
         sharding=int(d.get("sharding", {}).get("sharding_degree", 1)),
+        cp=int(d.get("cp_degree", 1) or 1),
         moe_expert_parallel=moe,
     )
     set_hcg(hcg)
