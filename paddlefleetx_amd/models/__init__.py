@@ -28,6 +28,9 @@ def build_module(config):
         from paddlefleetx_amd.models.vit_module import GeneralClsModule
         table["ViTModule"] = GeneralClsModule
         table["GeneralClsModule"] = GeneralClsModule
+    if name == "MOCOModule":
+        from paddlefleetx_amd.models.moco import MOCOModule
+        table["MOCOModule"] = MOCOModule
     if name in ("ErnieModule", "ErnieSeqClsModule"):
         from paddlefleetx_amd.models.ernie_module import (ErnieModule,
                                                           ErnieSeqClsModule)
