@@ -269,9 +269,11 @@ class ErnieForPretraining(nn.Module):
     def __init__(self, ernie: ErnieModel):
         super().__init__()
         self.ernie = ernie
+        dtype = ernie.embeddings.word_embeddings.weight.dtype
         self.cls = ErniePretrainingHeads(
             ernie.hidden_size, ernie.vocab_size, ernie.hidden_act,
-            embedding_weights=ernie.embeddings.word_embeddings.weight)
+            embedding_weights=ernie.embeddings.word_embeddings.weight,
+            dtype=dtype)
 
     def forward(self, input_ids, token_type_ids=None, position_ids=None,
                 attention_mask=None, masked_positions=None):
@@ -305,7 +307,9 @@ class ErnieForSequenceClassification(nn.Module):
         super().__init__()
         self.ernie = ernie
         self.dropout_p = dropout if dropout is not None else 0.1
-        self.classifier = nn.Linear(ernie.hidden_size, num_classes)
+        self.classifier = nn.Linear(
+            ernie.hidden_size, num_classes,
+            dtype=ernie.embeddings.word_embeddings.weight.dtype)
 
     def forward(self, input_ids, token_type_ids=None, position_ids=None,
                 attention_mask=None):
