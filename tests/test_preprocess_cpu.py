@@ -1,0 +1,61 @@
+"""Offline preprocessing pipeline: text -> jsonl -> token bins -> GPTDataset."""
+
+import json
+import os
+import subprocess
+import sys
+
+import numpy as np
+import pytest
+
+REPO = os.path.abspath(os.path.join(os.path.dirname(__file__), ".."))
+
+
+def _make_vocab_dir(tmp_path):
+    from paddlefleetx_amd.data.tokenizers.gpt_tokenizer import bytes_to_unicode
+    b2u = bytes_to_unicode()
+    vocab = {c: i for i, c in enumerate(b2u.values())}
+    vocab["<|endoftext|>"] = len(vocab)
+    d = tmp_path / "vocab"
+    d.mkdir()
+    with open(d / "vocab.json", "w", encoding="utf-8") as f:
+        json.dump(vocab, f, ensure_ascii=False)
+    with open(d / "merges.txt", "w") as f:
+        f.write("#version: 0.2\n")
+    return str(d)
+
+
+def test_full_preprocess_pipeline(tmp_path):
+    vocab_dir = _make_vocab_dir(tmp_path)
+    # raw text -> jsonl
+    raw = tmp_path / "raw.txt"
+    raw.write_text("\n".join(f"document number {i} with some text words"
+                             for i in range(30)))
+    jsonl = tmp_path / "corpus.jsonl"
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "tools/raw_trans_to_json.py"),
+         "--input_path", str(raw), "--output_path", str(jsonl)],
+        capture_output=True, text=True, cwd=REPO)
+    assert r.returncode == 0, r.stderr
+    assert len(jsonl.read_text().splitlines()) == 30
+
+    # jsonl -> token bins
+    prefix = tmp_path / "bins" / "corpus"
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "tools/preprocess_data.py"),
+         "--input_path", str(jsonl), "--output_prefix", str(prefix),
+         "--vocab_dir", vocab_dir, "--workers", "1"],
+        capture_output=True, text=True, cwd=REPO)
+    assert r.returncode == 0, r.stderr
+    ids = np.load(str(prefix) + "_ids.npy")
+    idx = np.load(str(prefix) + "_idx.npz")
+    assert idx["lens"].sum() == len(ids)
+    assert len(idx["lens"]) == 30
+
+    # token bins -> GPTDataset samples
+    from paddlefleetx_amd.data.gpt_dataset import GPTDataset
+    ds = GPTDataset(str(tmp_path / "bins"), mode="Train", max_seq_len=32,
+                    num_samples=20)
+    tokens, pos, labels, mask = ds[0]
+    assert tokens.shape == (32,)
+    assert (labels[:-1] == tokens[1:]).all()  # packed shift-by-one
