@@ -1,0 +1,160 @@
+"""Imagen: continuous-time Gaussian diffusion + text-conditioned U-Net.
+
+Reference: ppfleetx/models/multimodal_model/imagen/modeling.py —
+ImagenCriterion :94 (p2-weighted eps-MSE), ImagenModel :138 (T5/DebertaV2
+text encoder + unet + GaussianDiffusionContinuousTimes from utils.py:384).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from paddlefleetx_amd.models.imagen.unet import Unet
+
+
+def log_snr_cosine(t: torch.Tensor, s: float = 0.008) -> torch.Tensor:
+    """Cosine noise schedule in log-SNR form (utils.py:384
+    GaussianDiffusionContinuousTimes, alpha_cosine_log_snr)."""
+    return -torch.log(
+        torch.clamp(torch.tan((t + s) / (1 + s) * math.pi / 2) ** 2,
+                    min=1e-8, max=1e8))
+
+
+class GaussianDiffusionContinuousTimes(nn.Module):
+    def __init__(self, noise_schedule: str = "cosine", timesteps: int = 1000):
+        super().__init__()
+        assert noise_schedule == "cosine"
+        self.num_timesteps = timesteps
+
+    def sample_random_times(self, batch: int, device) -> torch.Tensor:
+        return torch.rand(batch, device=device)
+
+    def log_snr(self, t):
+        return log_snr_cosine(t)
+
+    def alpha_sigma(self, t):
+        log_snr = self.log_snr(t)
+        alpha = torch.sqrt(torch.sigmoid(log_snr))
+        sigma = torch.sqrt(torch.sigmoid(-log_snr))
+        return alpha, sigma
+
+    def q_sample(self, x0, t, noise=None):
+        noise = noise if noise is not None else torch.randn_like(x0)
+        alpha, sigma = self.alpha_sigma(t)
+        a = alpha[:, None, None, None]
+        s = sigma[:, None, None, None]
+        return a * x0 + s * noise, noise
+
+    def predict_start_from_noise(self, x_t, t, noise):
+        alpha, sigma = self.alpha_sigma(t)
+        a = alpha[:, None, None, None]
+        s = sigma[:, None, None, None]
+        return (x_t - s * noise) / a.clamp(min=1e-8)
+
+    def q_posterior_mean(self, x0, x_t, t, t_next):
+        """DDPM-style posterior mean for the t -> t_next step."""
+        log_snr_t = self.log_snr(t)
+        log_snr_n = self.log_snr(t_next)
+        alpha_t = torch.sqrt(torch.sigmoid(log_snr_t))
+        alpha_n = torch.sqrt(torch.sigmoid(log_snr_n))
+        sigma_t2 = torch.sigmoid(-log_snr_t)
+        sigma_n2 = torch.sigmoid(-log_snr_n)
+        c = -torch.expm1(log_snr_t - log_snr_n)
+        mean = alpha_n[:, None, None, None] * (
+            x_t * (1 - c)[:, None, None, None] /
+            alpha_t[:, None, None, None].clamp(min=1e-8)
+            + c[:, None, None, None] * x0)
+        var = sigma_n2 * c
+        return mean, var
+
+
+class ImagenCriterion(nn.Module):
+    """p2-weighted eps-prediction MSE (modeling.py:94-137)."""
+
+    def __init__(self, p2_loss_weight_gamma: float = 0.5,
+                 p2_loss_weight_k: float = 1.0):
+        super().__init__()
+        self.gamma = p2_loss_weight_gamma
+        self.k = p2_loss_weight_k
+
+    def forward(self, pred, target, log_snr=None):
+        loss = F.mse_loss(pred.float(), target.float(), reduction="none")
+        loss = loss.mean(dim=tuple(range(1, loss.ndim)))
+        if self.gamma > 0 and log_snr is not None:
+            w = (self.k + log_snr.exp()) ** -self.gamma
+            loss = loss * w
+        return loss.mean()
+
+
+class ImagenModel(nn.Module):
+    """One Imagen stage: text encoder + denoising U-Net (modeling.py:138)."""
+
+    def __init__(self, unet: Optional[Unet] = None, image_size: int = 64,
+                 text_encoder_name: str = "t5", text_embed_dim: int = 512,
+                 timesteps: int = 1000, cond_drop_prob: float = 0.1,
+                 text_encoder_kwargs: Optional[dict] = None,
+                 unet_kwargs: Optional[dict] = None,
+                 freeze_text_encoder: bool = True, **unused):
+        super().__init__()
+        self.image_size = image_size
+        self.cond_drop_prob = cond_drop_prob
+        if unet is None:
+            unet = Unet(text_embed_dim=text_embed_dim, **(unet_kwargs or {}))
+        self.unet = unet
+        from paddlefleetx_amd.models.t5 import T5EncoderModel
+        self.text_encoder = T5EncoderModel(
+            d_model=text_embed_dim, **(text_encoder_kwargs or {}))
+        if freeze_text_encoder:
+            for p in self.text_encoder.parameters():
+                p.requires_grad = False
+        self.scheduler = GaussianDiffusionContinuousTimes(
+            timesteps=timesteps)
+        self.criterion = ImagenCriterion()
+
+    def encode_text(self, text_ids, text_mask=None):
+        with torch.no_grad():
+            return self.text_encoder(text_ids, attention_mask=text_mask)
+
+    def forward(self, images, text_ids=None, text_mask=None,
+                text_embeds=None):
+        """Training loss for one denoising step."""
+        B = images.shape[0]
+        device = images.device
+        if text_embeds is None and text_ids is not None:
+            text_embeds = self.encode_text(text_ids, text_mask)
+        if text_embeds is not None and self.training and \
+                self.cond_drop_prob > 0:
+            keep = (torch.rand(B, device=device) >
+                    self.cond_drop_prob)[:, None, None]
+            text_embeds = text_embeds * keep
+        t = self.scheduler.sample_random_times(B, device)
+        x_t, noise = self.scheduler.q_sample(images, t)
+        pred = self.unet(x_t.to(images.dtype), t, text_embeds=text_embeds,
+                         text_mask=text_mask)
+        return self.criterion(pred, noise, log_snr=self.scheduler.log_snr(t))
+
+    @torch.no_grad()
+    def sample(self, text_ids=None, text_mask=None, batch_size: int = 1,
+               steps: int = 50, device=None):
+        device = device or next(self.unet.parameters()).device
+        text_embeds = self.encode_text(text_ids.to(device), text_mask) \
+            if text_ids is not None else None
+        x = torch.randn(batch_size, self.unet.channels, self.image_size,
+                        self.image_size, device=device)
+        times = torch.linspace(1.0, 0.0, steps + 1, device=device)
+        for i in range(steps):
+            t = times[i].repeat(batch_size)
+            t_next = times[i + 1].repeat(batch_size)
+            eps = self.unet(x, t, text_embeds=text_embeds,
+                            text_mask=text_mask)
+            x0 = self.scheduler.predict_start_from_noise(x, t, eps)
+            x0 = x0.clamp(-1, 1)
+            mean, var = self.scheduler.q_posterior_mean(x0, x, t, t_next)
+            noise = torch.randn_like(x) if i < steps - 1 else 0
+            x = mean + var.sqrt()[:, None, None, None] * noise
+        return x
