@@ -28,6 +28,9 @@ def build_module(config):
         from paddlefleetx_amd.models.vit_module import GeneralClsModule
         table["ViTModule"] = GeneralClsModule
         table["GeneralClsModule"] = GeneralClsModule
+    if name == "ImagenModule":
+        from paddlefleetx_amd.models.imagen_module import ImagenModule
+        table["ImagenModule"] = ImagenModule
     if name == "MOCOModule":
         from paddlefleetx_amd.models.moco import MOCOModule
         table["MOCOModule"] = MOCOModule
