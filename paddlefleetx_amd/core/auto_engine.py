@@ -1,0 +1,167 @@
+"""AutoEngine: automatic parallel-strategy planning + tuning.
+
+Reference: ppfleetx/core/engine/auto_engine.py:39-209 delegates to
+paddle.distributed.fleet.auto (graph compiler plans DP/MP/PP from
+shard_tensor annotations; `tune()` searches strategies). The MI355X-native
+equivalent is an explicit planner: it sizes the model against 288 GB HBM
+per GPU and the xGMI topology and picks (dp, mp, pp, sharding) — TP only
+as needed to fit (latency-bound allreduces on 7-link xGMI), then DP for
+throughput; ZeRO before PP intra-node. `tune()` measures candidate
+topologies with short timed runs.
+"""
+
+from __future__ import annotations
+
+import copy
+import math
+import time
+from typing import Dict, List, Optional, Tuple
+
+from paddlefleetx_amd.utils.log import logger
+
+HBM_BYTES = 288e9
+HBM_USABLE = 0.90  # leave headroom for activations spikes / RCCL buffers
+
+
+def estimate_param_count(mcfg: Dict) -> float:
+    """Decoder-LM parameter estimate from the Model section."""
+    h = int(mcfg.get("hidden_size", 1024))
+    L = int(mcfg.get("num_layers", mcfg.get("num_hidden_layers", 24)))
+    v = int(mcfg.get("vocab_size", 50304))
+    ffn = int(mcfg.get("ffn_hidden_size") or
+              mcfg.get("intermediate_size") or 4 * h)
+    per_layer = 4 * h * h + 2 * h * ffn + 9 * h  # qkv+out + up/down + norms
+    emb = v * h + int(mcfg.get("max_position_embeddings", 1024)) * h
+    return L * per_layer + emb
+
+
+def memory_per_gpu(params: float, dp: int, mp: int, pp: int, sharding: int,
+                   stage: int, micro_batch: int, seq: int, hidden: int,
+                   layers: int, use_recompute: bool,
+                   opt_bytes: int = 12) -> float:
+    """Rough bytes/GPU: bf16 params + grads + adam states + activations.
+    opt_bytes: 12 = fp32 master+m+v; 4 = bf16 moments (low-precision
+    optimizer fallback for 175B-scale single-node fits)."""
+    shard_params = params / (mp * pp)
+    opt_div = sharding if stage >= 1 and sharding > 1 else 1
+    param_b = shard_params * 2
+    if stage >= 3 and sharding > 1:
+        param_b = shard_params * 2 / sharding
+    grad_b = shard_params * 2 / (sharding if stage >= 2 else 1)
+    adam_b = shard_params * opt_bytes / opt_div
+    layers_local = max(1, layers // pp)
+    act_per_layer = micro_batch * seq * hidden * 2 * (2 if use_recompute
+                                                      else 16) / mp
+    act_b = act_per_layer * (1 if use_recompute else layers_local)
+    return param_b + grad_b + adam_b + act_b
+
+
+def plan_strategy(mcfg: Dict, world_size: int, micro_batch: int = 1,
+                  seq: Optional[int] = None,
+                  use_recompute: Optional[bool] = None) -> Dict[str, int]:
+    """Pick (dp, mp, pp, sharding, stage) for one 8xMI355X-style node."""
+    params = estimate_param_count(mcfg)
+    h = int(mcfg.get("hidden_size", 1024))
+    L = int(mcfg.get("num_layers", mcfg.get("num_hidden_layers", 24)))
+    seq = seq or int(mcfg.get("max_position_embeddings", 1024))
+    rec = bool(mcfg.get("use_recompute", False)) if use_recompute is None \
+        else use_recompute
+    budget = HBM_BYTES * HBM_USABLE
+
+    candidates: List[Tuple[float, Dict[str, int]]] = []
+    deg = [1, 2, 4, 8, 16]
+    for opt_bytes in (12, 4):  # fp32 adam first; bf16-moment fallback
+        for mp in [d for d in deg if d <= world_size]:
+            for pp in [d for d in deg if mp * d <= world_size]:
+                rest = world_size // (mp * pp)
+                if mp * pp * rest != world_size:
+                    continue
+                for sharding, stage in ((1, 1), (rest, 2), (rest, 3)):
+                    if sharding > rest or sharding < 1 or \
+                            (sharding > 1 and pp > 1):
+                        continue
+                    dp = rest // sharding
+                    mem = memory_per_gpu(params, dp, mp, pp, sharding, stage,
+                                         micro_batch, seq, h, L, rec,
+                                         opt_bytes)
+                    if mem > budget:
+                        continue
+                    # cost model: prefer pure DP; penalize mp (per-layer
+                    # latency-bound allreduce on xGMI), then pp (bubble),
+                    # then sharding stage (collective volume)
+                    cost = (mp - 1) * 1.0 + (pp - 1) * 0.6 + \
+                        (stage - 1) * 0.1 + (8 / max(1, dp)) * 0.01
+                    candidates.append((cost, {
+                        "dp_degree": dp, "mp_degree": mp, "pp_degree": pp,
+                        "sharding_degree": sharding, "sharding_stage": stage,
+                        "optimizer_dtype": "float32" if opt_bytes == 12
+                        else "bfloat16",
+                        "est_mem_gb": round(mem / 1e9, 1)}))
+        if candidates:
+            break
+    if not candidates:
+        raise ValueError(
+            f"model ({params/1e9:.1f}B params) does not fit on "
+            f"{world_size} GPUs with any supported strategy — add nodes or "
+            "enable recompute")
+    candidates.sort(key=lambda c: c[0])
+    best = candidates[0][1]
+    logger.info(f"auto plan for {params/1e9:.2f}B params on "
+                f"{world_size} GPUs: {best}")
+    return best
+
+
+class AutoEngine:
+    """Plans the strategy, then behaves like EagerEngine
+    (auto_engine.py:39-145 surface: fit/evaluate/tune/save/load)."""
+
+    def __init__(self, configs, module=None, mode: str = "train"):
+        self.configs = copy.deepcopy(configs)
+        import torch.distributed as torch_dist
+        world = torch_dist.get_world_size() if torch_dist.is_initialized() \
+            else 1
+        plan = plan_strategy(
+            dict(configs["Model"]), world,
+            micro_batch=int(configs.get("Global", {})
+                            .get("micro_batch_size", 1)))
+        d = self.configs.setdefault("Distributed", {})
+        d["dp_degree"] = plan["dp_degree"]
+        d["mp_degree"] = plan["mp_degree"]
+        d["pp_degree"] = plan["pp_degree"]
+        d.setdefault("sharding", {})
+        d["sharding"]["sharding_degree"] = plan["sharding_degree"]
+        d["sharding"]["sharding_stage"] = plan["sharding_stage"]
+        self.plan = plan
+
+        from paddlefleetx_amd.parallel.env import init_dist_env
+        init_dist_env(self.configs)
+        if module is None:
+            from paddlefleetx_amd.models import build_module
+            module = build_module(self.configs)
+        from paddlefleetx_amd.core.engine import EagerEngine
+        self._engine = EagerEngine(self.configs, module, mode=mode)
+
+    def __getattr__(self, name):
+        return getattr(self._engine, name)
+
+    def tune(self, train_loader, candidates: Optional[List[Dict]] = None,
+             steps: int = 5) -> Dict:
+        """Measure a few steps per candidate batch size; returns timings
+        (auto_engine.py:146 tune — strategy search surface)."""
+        import torch
+        timings = {}
+        it = iter(train_loader)
+        batch = next(it)
+        for cand in (candidates or [{"accumulate_steps": a}
+                                    for a in (1, 2, 4)]):
+            acc = cand.get("accumulate_steps", 1)
+            self._engine.accumulate_steps = acc
+            t0 = time.time()
+            for _ in range(steps):
+                self._engine._fit_impl(batch)
+            if torch.cuda.is_available():
+                torch.cuda.synchronize()
+            timings[str(cand)] = (time.time() - t0) / steps
+        best = min(timings, key=timings.get)
+        logger.info(f"tune results: {timings} -> best {best}")
+        return timings
