@@ -161,7 +161,10 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
     Strided q, Strided k, Strided v, StridedMut o, float* __restrict__ lse_out,
     int H, int S, float scale, int kv_total) {
   constexpr int KRS = D + PAD;          // K row-major row stride
-  constexpr int VRS = TILE + PAD;       // V^T row stride
+  // V^T row stride: 68 shorts (136 B) keeps b64 writes/reads 8B-aligned
+  // while avoiding the 0-mod-128B d-stride that made the transpose writes
+  // 16-way bank conflicted (measured SQ_LDS_BANK_CONFLICT 15% of cycles)
+  constexpr int VRS = TILE + 4;
   constexpr int PRS = TILE + PAD;       // per-wave P row stride
   constexpr int NDT = D / 16;
   constexpr int KSZ = TILE * KRS;       // one K buffer (shorts)
@@ -204,41 +207,60 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
       : kv_total;
   unsigned short* myp = p_lds + wid * 16 * PRS;
 
-  // ---- staging thread map: D/16 threads per row, 16 shorts each ----
-  constexpr int TPR = D / 16;  // threads per kv row
+  // ---- staging thread maps ----
+  // K (row-major image): D/16 threads per row, 16 shorts each
+  constexpr int TPR = D / 16;
   const int st_row = threadIdx.x / TPR;          // > TILE-1 threads idle (D=64)
   const int st_col = (threadIdx.x % TPR) * 16;
   const bool st_on = st_row < TILE;
-  short8v kreg[2], vreg[2];
+  // V (transposed image): each thread owns a 4x4 block, transposes it in
+  // registers and writes 4 b64 rows of the [d][kv] image — no b16 scatter
+  constexpr int VCB = D / 4;                      // 4-col blocks per row
+  const int vb_row = (threadIdx.x / VCB) * 4;
+  const int vb_col = (threadIdx.x % VCB) * 4;
+  const bool vb_on = vb_row < TILE;
+  short8v kreg[2];
+  short4v vblk[4];
 
-  // prologue: load tile 0 and write buffer 0
-  {
-    const int nvalid = min(TILE, S);
+  auto load_kv = [&](int kv0) {
 #pragma unroll
-    for (int h2 = 0; h2 < 2; ++h2) { kreg[h2] = short8v{}; vreg[h2] = short8v{}; }
-    if (st_on && st_row < nvalid) {
+    for (int h2 = 0; h2 < 2; ++h2) kreg[h2] = short8v{};
+#pragma unroll
+    for (int r = 0; r < 4; ++r) vblk[r] = short4v{};
+    const int krow = kv0 + st_row;
+    if (st_on && krow < S) {
       const unsigned short* ks = (const unsigned short*)kp +
-                                 (long)st_row * k.rs + st_col;
-      const unsigned short* vs = (const unsigned short*)vp +
-                                 (long)st_row * v.rs + st_col;
+                                 (long)krow * k.rs + st_col;
       kreg[0] = *reinterpret_cast<const short8v*>(ks);
       kreg[1] = *reinterpret_cast<const short8v*>(ks + 8);
-      vreg[0] = *reinterpret_cast<const short8v*>(vs);
-      vreg[1] = *reinterpret_cast<const short8v*>(vs + 8);
     }
+    if (vb_on) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int vrow = kv0 + vb_row + r;
+        if (vrow < S)
+          vblk[r] = *reinterpret_cast<const short4v*>(
+              (const unsigned short*)vp + (long)vrow * v.rs + vb_col);
+      }
+    }
+  };
+  auto write_kv = [&](unsigned short* kd, unsigned short* vd) {
     if (st_on) {
-      *reinterpret_cast<short8v*>(k_lds + st_row * KRS + st_col) = kreg[0];
-      *reinterpret_cast<short8v*>(k_lds + st_row * KRS + st_col + 8) = kreg[1];
-      const unsigned short* vr = (const unsigned short*)&vreg[0];
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        vt_lds[(st_col + j) * VRS + st_row] = vr[j];
-      vr = (const unsigned short*)&vreg[1];
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        vt_lds[(st_col + 8 + j) * VRS + st_row] = vr[j];
+      *reinterpret_cast<short8v*>(kd + st_row * KRS + st_col) = kreg[0];
+      *reinterpret_cast<short8v*>(kd + st_row * KRS + st_col + 8) = kreg[1];
     }
-  }
+    if (vb_on) {
+#pragma unroll
+      for (int c = 0; c < 4; ++c) {
+        short4v t = {vblk[0][c], vblk[1][c], vblk[2][c], vblk[3][c]};
+        *reinterpret_cast<short4v*>(vd + (vb_col + c) * VRS + vb_row) = t;
+      }
+    }
+  };
+
+  // prologue: load tile 0 and write buffer 0
+  load_kv(0);
+  write_kv(k_lds, vt_lds);
   __syncthreads();
 
   for (int kt = 0; kt < kv_tiles; ++kt) {
@@ -249,21 +271,7 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
 
     // ---- issue next tile's global loads (land during this tile's MFMAs) --
     const bool have_next = (kt + 1) < kv_tiles;
-    if (have_next) {
-      const int nrow = (kt + 1) * TILE + st_row;
-#pragma unroll
-      for (int h2 = 0; h2 < 2; ++h2) { kreg[h2] = short8v{}; vreg[h2] = short8v{}; }
-      if (st_on && nrow < S) {
-        const unsigned short* ks = (const unsigned short*)kp +
-                                   (long)nrow * k.rs + st_col;
-        const unsigned short* vs = (const unsigned short*)vp +
-                                   (long)nrow * v.rs + st_col;
-        kreg[0] = *reinterpret_cast<const short8v*>(ks);
-        kreg[1] = *reinterpret_cast<const short8v*>(ks + 8);
-        vreg[0] = *reinterpret_cast<const short8v*>(vs);
-        vreg[1] = *reinterpret_cast<const short8v*>(vs + 8);
-      }
-    }
+    if (have_next) load_kv((kt + 1) * TILE);
 
     // waves whose rows are entirely above this kv tile skip compute
     // (they still stage and hit the barrier)
@@ -350,18 +358,8 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
     }
 
     // ---- write next tile into the other buffer, one barrier per tile ----
-    if (have_next && st_on) {
-      unsigned short* kd = k_lds + (cur ^ 1) * KSZ;
-      unsigned short* vd = vt_lds + (cur ^ 1) * VSZ;
-      *reinterpret_cast<short8v*>(kd + st_row * KRS + st_col) = kreg[0];
-      *reinterpret_cast<short8v*>(kd + st_row * KRS + st_col + 8) = kreg[1];
-      const unsigned short* vr = (const unsigned short*)&vreg[0];
-#pragma unroll
-      for (int j = 0; j < 8; ++j) vd[(st_col + j) * VRS + st_row] = vr[j];
-      vr = (const unsigned short*)&vreg[1];
-#pragma unroll
-      for (int j = 0; j < 8; ++j) vd[(st_col + 8 + j) * VRS + st_row] = vr[j];
-    }
+    if (have_next)
+      write_kv(k_lds + (cur ^ 1) * KSZ, vt_lds + (cur ^ 1) * VSZ);
     __syncthreads();
   }
 
@@ -420,13 +418,14 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dkv_kernel(
     const float* __restrict__ lse, const float* __restrict__ delta,
     StridedMut dk, StridedMut dv, int H, int S, float scale, int q_tiles) {
   constexpr int RS = D + PAD;
-  constexpr int TS = TILE + PAD;
+  constexpr int TS = TILE + PAD;       // P image stride (b128-aligned)
+  constexpr int TRS = TILE + 4;        // transposed q/dO stride (see VRS)
   constexpr int NDT = D / 16;
 
   __shared__ unsigned short q_lds[TILE * RS];
-  __shared__ unsigned short qt_lds[D * TS];
+  __shared__ unsigned short qt_lds[D * TRS];
   __shared__ unsigned short do_lds[TILE * RS];
-  __shared__ unsigned short dot_lds[D * TS];
+  __shared__ unsigned short dot_lds[D * TRS];
   __shared__ unsigned short p_lds[FW_WAVES * 16 * TS];
   __shared__ float lsed_lds[2 * TILE];
 
@@ -458,12 +457,18 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dkv_kernel(
   const int my_kvrow = kvrow0 + crow4;
   unsigned short* myp = p_lds + wid * 16 * TS;
 
-  // staging thread map: D/16 threads per q row, 16 shorts per tensor each
+  // staging maps: row-major via D/16 threads per q row; transposed images
+  // via per-thread 4x4 register transpose (b64 writes — see fwd VRS note)
   constexpr int TPR = D / 16;
   const int st_row = threadIdx.x / TPR;
   const int st_col = (threadIdx.x % TPR) * 16;
   const bool st_on = st_row < TILE;
+  constexpr int VCB = D / 4;
+  const int vb_row = (threadIdx.x / VCB) * 4;
+  const int vb_col = (threadIdx.x % VCB) * 4;
+  const bool vb_on = vb_row < TILE;
   short8v qreg[2], dreg[2];
+  short4v qblk[4], dblk[4];
   float lse_s = 0.f, dlt_s = 0.f;
 
   const int qt_first = CAUSAL ? (kt * QTILE) / TILE : 0;
@@ -471,6 +476,8 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dkv_kernel(
   auto issue_loads = [&](int q0) {
 #pragma unroll
     for (int h2 = 0; h2 < 2; ++h2) { qreg[h2] = short8v{}; dreg[h2] = short8v{}; }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) { qblk[r] = short4v{}; dblk[r] = short4v{}; }
     const int row = q0 + st_row;
     if (st_on && row < S) {
       const unsigned short* qs = (const unsigned short*)qp + (long)row * q.rs + st_col;
@@ -479,6 +486,18 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dkv_kernel(
       qreg[1] = *reinterpret_cast<const short8v*>(qs + 8);
       dreg[0] = *reinterpret_cast<const short8v*>(ds);
       dreg[1] = *reinterpret_cast<const short8v*>(ds + 8);
+    }
+    if (vb_on) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int vrow = q0 + vb_row + r;
+        if (vrow < S) {
+          qblk[r] = *reinterpret_cast<const short4v*>(
+              (const unsigned short*)qp + (long)vrow * q.rs + vb_col);
+          dblk[r] = *reinterpret_cast<const short4v*>(
+              (const unsigned short*)dop + (long)vrow * dout.rs + vb_col);
+        }
+      }
     }
     const int li = threadIdx.x;
     if (li < TILE) {
@@ -493,19 +512,14 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dkv_kernel(
       *reinterpret_cast<short8v*>(q_lds + st_row * RS + st_col + 8) = qreg[1];
       *reinterpret_cast<short8v*>(do_lds + st_row * RS + st_col) = dreg[0];
       *reinterpret_cast<short8v*>(do_lds + st_row * RS + st_col + 8) = dreg[1];
-      const unsigned short* qr = (const unsigned short*)&qreg[0];
-      const unsigned short* dr = (const unsigned short*)&dreg[0];
+    }
+    if (vb_on) {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        qt_lds[(st_col + j) * TS + st_row] = qr[j];
-        dot_lds[(st_col + j) * TS + st_row] = dr[j];
-      }
-      qr = (const unsigned short*)&qreg[1];
-      dr = (const unsigned short*)&dreg[1];
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        qt_lds[(st_col + 8 + j) * TS + st_row] = qr[j];
-        dot_lds[(st_col + 8 + j) * TS + st_row] = dr[j];
+      for (int c = 0; c < 4; ++c) {
+        short4v tq = {qblk[0][c], qblk[1][c], qblk[2][c], qblk[3][c]};
+        short4v td = {dblk[0][c], dblk[1][c], dblk[2][c], dblk[3][c]};
+        *reinterpret_cast<short4v*>(qt_lds + (vb_col + c) * TRS + vb_row) = tq;
+        *reinterpret_cast<short4v*>(dot_lds + (vb_col + c) * TRS + vb_row) = td;
       }
     }
     if (threadIdx.x < TILE) {
@@ -559,7 +573,7 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dkv_kernel(
         bf8 pa = read_a_frag_lds(myp, TS, kc * 32, lane);
 #pragma unroll
         for (int dt = 0; dt < NDT; ++dt) {
-          bf8 db = read_b_frag(dot_lds, TS, dt * 16, kc * 32, lane);
+          bf8 db = read_b_frag(dot_lds, TRS, dt * 16, kc * 32, lane);
           dvacc[dt] = MFMA_BF16(pa, db, dvacc[dt]);
         }
       }
@@ -591,7 +605,7 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dkv_kernel(
         bf8 dsa = read_a_frag_lds(myp, TS, kc * 32, lane);
 #pragma unroll
         for (int dt = 0; dt < NDT; ++dt) {
-          bf8 qb = read_b_frag(qt_lds, TS, dt * 16, kc * 32, lane);
+          bf8 qb = read_b_frag(qt_lds, TRS, dt * 16, kc * 32, lane);
           dkacc[dt] = MFMA_BF16(dsa, qb, dkacc[dt]);
         }
       }
@@ -630,10 +644,11 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dq_kernel(
     const float* __restrict__ lse, const float* __restrict__ delta,
     StridedMut dq, int H, int S, float scale, int kv_total) {
   constexpr int RS = D + PAD;
-  constexpr int TS = TILE + PAD;
+  constexpr int TS = TILE + PAD;   // P image stride
+  constexpr int TRS = TILE + 4;    // K^T stride (b64 images, see fwd VRS)
   constexpr int NDT = D / 16;
   constexpr int KSZ = TILE * RS;   // K row-major buffer
-  constexpr int TSZ = D * TS;      // K^T buffer
+  constexpr int TSZ = D * TRS;     // K^T buffer
   constexpr int VSZ = TILE * RS;   // V row-major buffer
 
   __shared__ unsigned short smem[2 * (KSZ + TSZ + VSZ) + FW_WAVES * 16 * TS];
@@ -680,16 +695,24 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dq_kernel(
       ? min(kv_total, (qt * QTILE + QTILE - 1) / TILE + 1)
       : kv_total;
 
-  // staging map: D/16 threads per kv row, 16 shorts per tensor each
+  // staging maps: D/16 threads per kv row (row-major K/V); per-thread
+  // 4x4 register transpose for the K^T image (b64 writes, see fwd)
   constexpr int TPR = D / 16;
   const int st_row = threadIdx.x / TPR;
   const int st_col = (threadIdx.x % TPR) * 16;
   const bool st_on = st_row < TILE;
+  constexpr int VCB = D / 4;
+  const int vb_row = (threadIdx.x / VCB) * 4;
+  const int vb_col = (threadIdx.x % VCB) * 4;
+  const bool vb_on = vb_row < TILE;
   short8v kreg[2], vreg[2];
+  short4v kblk[4];
 
   auto issue_loads = [&](int kv0) {
 #pragma unroll
     for (int h2 = 0; h2 < 2; ++h2) { kreg[h2] = short8v{}; vreg[h2] = short8v{}; }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) kblk[r] = short4v{};
     const int row = kv0 + st_row;
     if (st_on && row < S) {
       const unsigned short* ks = (const unsigned short*)kp + (long)row * k.rs + st_col;
@@ -699,22 +722,33 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dq_kernel(
       vreg[0] = *reinterpret_cast<const short8v*>(vs);
       vreg[1] = *reinterpret_cast<const short8v*>(vs + 8);
     }
+    if (vb_on) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int vrow = kv0 + vb_row + r;
+        if (vrow < S)
+          kblk[r] = *reinterpret_cast<const short4v*>(
+              (const unsigned short*)kp + (long)vrow * k.rs + vb_col);
+      }
+    }
   };
   auto write_tiles = [&](int buf) {
-    if (!st_on) return;
     unsigned short* kd = k_lds + buf * KSZ;
     unsigned short* ktd = kt_lds + buf * TSZ;
     unsigned short* vd = v_lds + buf * VSZ;
-    *reinterpret_cast<short8v*>(kd + st_row * RS + st_col) = kreg[0];
-    *reinterpret_cast<short8v*>(kd + st_row * RS + st_col + 8) = kreg[1];
-    *reinterpret_cast<short8v*>(vd + st_row * RS + st_col) = vreg[0];
-    *reinterpret_cast<short8v*>(vd + st_row * RS + st_col + 8) = vreg[1];
-    const unsigned short* kr = (const unsigned short*)&kreg[0];
+    if (st_on) {
+      *reinterpret_cast<short8v*>(kd + st_row * RS + st_col) = kreg[0];
+      *reinterpret_cast<short8v*>(kd + st_row * RS + st_col + 8) = kreg[1];
+      *reinterpret_cast<short8v*>(vd + st_row * RS + st_col) = vreg[0];
+      *reinterpret_cast<short8v*>(vd + st_row * RS + st_col + 8) = vreg[1];
+    }
+    if (vb_on) {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) ktd[(st_col + j) * TS + st_row] = kr[j];
-    kr = (const unsigned short*)&kreg[1];
-#pragma unroll
-    for (int j = 0; j < 8; ++j) ktd[(st_col + 8 + j) * TS + st_row] = kr[j];
+      for (int c = 0; c < 4; ++c) {
+        short4v t = {kblk[0][c], kblk[1][c], kblk[2][c], kblk[3][c]};
+        *reinterpret_cast<short4v*>(ktd + (vb_col + c) * TRS + vb_row) = t;
+      }
+    }
   };
 
   issue_loads(0);
@@ -779,7 +813,7 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dq_kernel(
         bf8 dsa = read_a_frag_lds(myp, TS, kc * 32, lane);
 #pragma unroll
         for (int dt = 0; dt < NDT; ++dt) {
-          bf8 kb = read_b_frag(ktb_lds, TS, dt * 16, kc * 32, lane);
+          bf8 kb = read_b_frag(ktb_lds, TRS, dt * 16, kc * 32, lane);
           dqacc[dt] = MFMA_BF16(dsa, kb, dqacc[dt]);
         }
       }
