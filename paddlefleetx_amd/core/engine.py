@@ -121,8 +121,18 @@ class EagerEngine(BasicEngine):
                     sharding_stage=self.sharding_stage)
             # broadcast initial params across dp (and sharding) so replicas agree
             self._sync_params()
+            # DP grad allreduce overlapped with the last backward
+            # (reference reduce_overlap knob, eager_engine.py:303-307)
+            self._overlap_reduce = False
+            want_overlap = bool(configs.get("Distributed", {})
+                                .get("reduce_overlap", True))
+            if want_overlap and isinstance(self.optimizer, FusedAdamW) \
+                    and not self.is_pipeline:
+                self._overlap_reduce = self.optimizer.enable_overlap(
+                    self.hcg.get_data_parallel_group())
         else:
             self.optimizer = None
+            self._overlap_reduce = False
 
         self._load_recovery = {"step": 0, "epoch": 0}
         ckpt_dir = sl.get("ckpt_dir") if isinstance(sl, dict) else None
@@ -224,9 +234,11 @@ class EagerEngine(BasicEngine):
     def _model_forward_backward(self, batch) -> torch.Tensor:
         micros = _split_micro(batch, self.accumulate_steps)
         total = 0.0
-        for mb in micros:
+        for i, mb in enumerate(micros):
             loss = self.module.training_step(mb)
             scaled = loss * (self.loss_scale / self.accumulate_steps)
+            if self._overlap_reduce and i == len(micros) - 1:
+                self.optimizer.begin_overlap_reduce()
             self.module.backward(scaled)
             total += float(loss.detach())
         return torch.tensor(total / self.accumulate_steps)
@@ -243,7 +255,10 @@ class EagerEngine(BasicEngine):
             # ZeRO reduce-scatter over sharding + DP allreduce on fused buffers
             dp = self.hcg.get_data_parallel_group()
             n_replicas = dp.world_size * self.sharding_degree
-            self.optimizer.reduce_gradients(dp, avg_factor=1.0)
+            if getattr(self, "_overlap_reduce", False):
+                self.optimizer.finish_overlap_reduce()  # issued in backward
+            else:
+                self.optimizer.reduce_gradients(dp, avg_factor=1.0)
             inv = 1.0 / (n_replicas * self.loss_scale)
             self.optimizer.scale_grads(inv)
             if self.loss_scale != 1.0:

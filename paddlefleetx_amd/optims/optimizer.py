@@ -173,6 +173,62 @@ class FusedAdamW(torch.optim.Optimizer):
     def grad_buffers(self) -> List[torch.Tensor]:
         return [b.grad_flat for b in self.buckets]
 
+    # --- DP allreduce overlapped with the last micro-batch's backward ------
+    # (reference knob reduce_overlap, eager_engine.py:303-307; here the
+    # bucket's allreduce is issued by the last param's post-accumulate hook
+    # so RCCL traffic rides the xGMI links while backward keeps computing)
+    def enable_overlap(self, dp_group):
+        if self.sharding_group is not None or self._fp32_main_grad:
+            return False  # ZeRO path keeps the fused reduce-scatter
+        if dp_group is None or getattr(dp_group, "world_size", 1) <= 1:
+            return False
+        self._ov_group = dp_group.group if hasattr(dp_group, "group") \
+            else dp_group
+        self._ov_active = False
+        self._ov_handles = []
+        self._param_bucket = {}
+        self._bucket_pending: Dict[int, int] = {}
+        for bi, b in enumerate(self.buckets):
+            for p in b.params:
+                self._param_bucket[id(p)] = bi
+
+        def make_hook():
+            def hook(p):
+                if not self._ov_active:
+                    return
+                bi = self._param_bucket[id(p)]
+                self._bucket_pending[bi] -= 1
+                if self._bucket_pending[bi] == 0:
+                    h = dist.all_reduce(self.buckets[bi].grad_flat,
+                                        group=self._ov_group, async_op=True)
+                    self._ov_handles.append(h)
+            return hook
+
+        for b in self.buckets:
+            for p in b.params:
+                self._hooks.append(
+                    p.register_post_accumulate_grad_hook(make_hook()))
+        return True
+
+    def begin_overlap_reduce(self):
+        """Arm the hooks for the LAST micro-batch's backward."""
+        self._ov_active = True
+        self._ov_handles = []
+        self._bucket_pending = {i: len(b.params)
+                                for i, b in enumerate(self.buckets)}
+
+    def finish_overlap_reduce(self):
+        """Wait for in-flight allreduces; sync-reduce any bucket whose hooks
+        never all fired (e.g. params without grads this step)."""
+        self._ov_active = False
+        for h in self._ov_handles:
+            h.wait()
+        for bi, left in self._bucket_pending.items():
+            if left > 0:
+                dist.all_reduce(self.buckets[bi].grad_flat,
+                                group=self._ov_group)
+        self._ov_handles = []
+
     def reduce_gradients(self, group, avg_factor: Optional[float] = None):
         """Reduce fused grad buffers: reduce-scatter over the sharding group
         (ZeRO) then allreduce own shard over DP."""
