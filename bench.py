@@ -45,7 +45,9 @@ MODELS = {
 
 
 def topology_for(n):
-    return {1: (1, 1, 1), 2: (1, 2, 1), 4: (1, 2, 2), 8: (2, 2, 2)}.get(
+    """(dp, tp, pp). N=8 is the named DP2xTP2xPP2 config; intermediate N
+    scale out over DP (weak scaling: per-GPU work fixed)."""
+    return {1: (1, 1, 1), 2: (2, 1, 1), 4: (4, 1, 1), 8: (2, 2, 2)}.get(
         n, (n, 1, 1))  # fallback: pure DP
 
 
@@ -62,9 +64,15 @@ def main():
     from paddlefleetx_amd.core import EagerEngine
 
     # defaults from the measured 1-GPU sweep (profiles/): micro=8 keeps the
-    # hipBLASLt GEMMs at M=8192 where MFMA efficiency is ~20% higher than M=2048
-    micro = args.micro_batch or 8
-    acc = args.acc_steps or 4
+    # hipBLASLt GEMMs at M=8192 where MFMA efficiency is ~20% higher than
+    # M=2048. Under pipeline parallel trade a little GEMM width for more
+    # micro-batches (1F1B bubble = (pp-1)/(acc+pp-1)).
+    if pp > 1:
+        micro = args.micro_batch or 4
+        acc = args.acc_steps or 8
+    else:
+        micro = args.micro_batch or 8
+        acc = args.acc_steps or 4
     local_bs = micro * acc
     seq = args.seq_len
     cfg_path = os.path.join(os.path.dirname(os.path.abspath(__file__)),
