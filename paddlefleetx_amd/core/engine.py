@@ -344,6 +344,29 @@ class EagerEngine(BasicEngine):
             self.module.validation_step_end({
                 "epoch": epoch, "batch": i, "loss": float(loss),
                 "eval_cost": (time.time() - t0) / (i + 1)})
+        if hasattr(self.module, "validation_epoch_end"):
+            try:
+                self.module.validation_epoch_end()
+            except TypeError:
+                pass  # modules with a log_dict-taking signature
+
+    def compress_model(self):
+        """Apply the Compress config section (reference eager_engine.py
+        :757-774 via utils/compression_helper.py)."""
+        ccfg = self.configs.get("Compress", {})
+        if not ccfg:
+            return
+        from paddlefleetx_amd.utils.compression_helper import (prune_model,
+                                                               quant_model)
+        if "Prune" in ccfg:
+            p = ccfg["Prune"] or {}
+            prune_model(self.module.model,
+                        ratio=float(p.get("ratio", 0.125)),
+                        structured=p.get("criterion", "l1_norm") != "unstructured",
+                        include=p.get("include"))
+        if "Quantization" in ccfg:
+            q = ccfg["Quantization"] or {}
+            quant_model(self.module.model, include=q.get("include"))
 
     def evaluate(self, valid_data_loader=None, epoch: int = 0):
         self._evaluate_impl(epoch, valid_data_loader)
