@@ -350,6 +350,25 @@ class EagerEngine(BasicEngine):
             except TypeError:
                 pass  # modules with a log_dict-taking signature
 
+    def export(self, output_dir: str = "./exported_model"):
+        """Export the trained model for inference (reference
+        eager_engine.py:832 + utils/export.py)."""
+        from paddlefleetx_amd.utils.export import export_inference_model
+        extra = {}
+        if "Generation" in self.configs:
+            extra["generation"] = dict(self.configs["Generation"])
+        return export_inference_model(self.module.model,
+                                      dict(self.configs["Model"]),
+                                      output_dir, extra=extra)
+
+    def inference(self, data, model_dir: str = "./exported_model"):
+        """Run the exported model (reference eager_engine.py:852)."""
+        from paddlefleetx_amd.core.inference_engine import InferenceEngine
+        if not hasattr(self, "_infer_engine"):
+            mp = self.hcg.get_model_parallel_world_size()
+            self._infer_engine = InferenceEngine(model_dir, mp_degree=mp)
+        return self._infer_engine.predict(data)
+
     def compress_model(self):
         """Apply the Compress config section (reference eager_engine.py
         :757-774 via utils/compression_helper.py)."""

@@ -165,8 +165,8 @@ class MOCOModule(BasicModule):
         return nn.CrossEntropyLoss()
 
     def training_step(self, batch):
-        (x1, x2), = (batch[:1] if isinstance(batch[0], (tuple, list))
-                     else [(batch[0], batch[0])])
+        # batch: ((view1, view2), _) from a two-crop transform, or a single
+        # image tensor (both views identical — plumbing tests)
         if isinstance(batch[0], (tuple, list)):
             x1, x2 = batch[0]
         else:
