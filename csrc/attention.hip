@@ -186,24 +186,31 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
   const __hip_bfloat16* kp = k.at(b, hh);
   const __hip_bfloat16* vp = v.at(b, hh);
 
-  const int qrow0 = qt * QTILE + wid * 16;
-  bf8 qfrag[D / 32];
-  load_a_frags<D>(qp, q.rs, qrow0, S, lane, qfrag);
+  // RB 16-row fragments per wave (32 q rows/wave, 256/block): every K/V
+  // B-fragment LDS read feeds RB MFMAs, and barriers amortize over 2x rows
+  constexpr int RB = 2;
+  const int qrow0 = qt * (QTILE * RB) + wid * (16 * RB);
+  bf8 qfrag[RB][D / 32];
+#pragma unroll
+  for (int rb = 0; rb < RB; ++rb)
+    load_a_frags<D>(qp, q.rs, qrow0 + rb * 16, S, lane, qfrag[rb]);
 
-  f4 oacc[NDT];
+  f4 oacc[RB][NDT];
 #pragma unroll
-  for (int i = 0; i < NDT; ++i) oacc[i] = f4{0.f, 0.f, 0.f, 0.f};
-  float m_r[4], l_r[4];
+  for (int rb = 0; rb < RB; ++rb)
 #pragma unroll
-  for (int r = 0; r < 4; ++r) { m_r[r] = -INFINITY; l_r[r] = 0.f; }
+    for (int i = 0; i < NDT; ++i) oacc[rb][i] = f4{0.f, 0.f, 0.f, 0.f};
+  float m_r[RB][4], l_r[RB][4];
+#pragma unroll
+  for (int rb = 0; rb < RB; ++rb)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) { m_r[rb][r] = -INFINITY; l_r[rb][r] = 0.f; }
 
   const int ccol = lane & 15;
   const int crow4 = (lane >> 4) * 4;
-  const int my_qrow = qrow0 + crow4;
-  const int wave_last_row = min(qrow0 + 15, S - 1);
 
   const int kv_tiles = CAUSAL
-      ? min(kv_total, (qt * QTILE + QTILE - 1) / TILE + 1)
+      ? min(kv_total, (qt * QTILE * RB + QTILE * RB - 1) / TILE + 1)
       : kv_total;
   unsigned short* myp = p_lds + wid * 16 * PRS;
 
@@ -273,10 +280,14 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
     const bool have_next = (kt + 1) < kv_tiles;
     if (have_next) load_kv((kt + 1) * TILE);
 
-    // waves whose rows are entirely above this kv tile skip compute
-    // (they still stage and hit the barrier)
-    const bool active = !CAUSAL || (kv0 <= wave_last_row);
-    if (active) {
+#pragma unroll
+    for (int rb = 0; rb < RB; ++rb) {
+      const int my_qrow = qrow0 + rb * 16 + crow4;
+      // waves whose rows are entirely above this kv tile skip compute
+      // (they still stage and hit the barrier)
+      const bool active = !CAUSAL ||
+          (kv0 <= min(qrow0 + rb * 16 + 15, S - 1));
+      if (!active) continue;
       // ---- QK^T ----
       f4 s[4];
 #pragma unroll
@@ -285,7 +296,7 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
 #pragma unroll
         for (int kc = 0; kc < D / 32; ++kc) {
           bf8 kb = read_b_frag(kb_lds, KRS, ct * 16, kc * 32, lane);
-          acc = MFMA_BF16(qfrag[kc], kb, acc);
+          acc = MFMA_BF16(qfrag[rb][kc], kb, acc);
         }
         s[ct] = acc;
       }
@@ -309,10 +320,11 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
       float alpha[4];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        float nm = fmaxf(m_r[r], pmax[r]);
+        float nm = fmaxf(m_r[rb][r], pmax[r]);
         if (nm == -INFINITY) nm = 0.f;
-        alpha[r] = (m_r[r] == -INFINITY) ? 0.f : __expf(m_r[r] - nm);
-        m_r[r] = (m_r[r] == -INFINITY && pmax[r] == -INFINITY) ? -INFINITY : nm;
+        alpha[r] = (m_r[rb][r] == -INFINITY) ? 0.f : __expf(m_r[rb][r] - nm);
+        m_r[rb][r] = (m_r[rb][r] == -INFINITY && pmax[r] == -INFINITY)
+            ? -INFINITY : nm;
       }
 
       float psum[4] = {0.f, 0.f, 0.f, 0.f};
@@ -320,7 +332,8 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
       for (int ct = 0; ct < 4; ++ct) {
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
-          float p = (s[ct][r] == -INFINITY) ? 0.f : __expf(s[ct][r] - m_r[r]);
+          float p = (s[ct][r] == -INFINITY) ? 0.f
+              : __expf(s[ct][r] - m_r[rb][r]);
           s[ct][r] = p;
           psum[r] += p;
         }
@@ -328,12 +341,12 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         psum[r] = group16_reduce_sum(psum[r]);
-        l_r[r] = l_r[r] * alpha[r] + psum[r];
+        l_r[rb][r] = l_r[rb][r] * alpha[r] + psum[r];
       }
 #pragma unroll
       for (int i = 0; i < NDT; ++i) {
 #pragma unroll
-        for (int r = 0; r < 4; ++r) oacc[i][r] *= alpha[r];
+        for (int r = 0; r < 4; ++r) oacc[rb][i][r] *= alpha[r];
       }
 
       // ---- P -> per-wave LDS for A-fragments ----
@@ -352,7 +365,7 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
 #pragma unroll
         for (int dt = 0; dt < NDT; ++dt) {
           bf8 vb = read_b_frag(vb_lds, VRS, dt * 16, kc * 32, lane);
-          oacc[dt] = MFMA_BF16(pa, vb, oacc[dt]);
+          oacc[rb][dt] = MFMA_BF16(pa, vb, oacc[rb][dt]);
         }
       }
     }
@@ -366,20 +379,23 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
   // ---- epilogue (strided o) ----
   __hip_bfloat16* op = o.at(b, hh);
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    int qrow = my_qrow + r;
-    if (qrow >= S) continue;
-    float inv = (l_r[r] > 0.f) ? 1.f / l_r[r] : 0.f;
-    unsigned short* orow = (unsigned short*)op + (long)qrow * o.rs;
+  for (int rb = 0; rb < RB; ++rb)
 #pragma unroll
-    for (int dt = 0; dt < NDT; ++dt) {
-      orow[dt * 16 + ccol] = f2bf_raw(oacc[dt][r] * inv);
+    for (int r = 0; r < 4; ++r) {
+      int qrow = qrow0 + rb * 16 + crow4 + r;
+      if (qrow >= S) continue;
+      float inv = (l_r[rb][r] > 0.f) ? 1.f / l_r[rb][r] : 0.f;
+      unsigned short* orow = (unsigned short*)op + (long)qrow * o.rs;
+#pragma unroll
+      for (int dt = 0; dt < NDT; ++dt) {
+        orow[dt * 16 + ccol] = f2bf_raw(oacc[rb][dt][r] * inv);
+      }
+      if (ccol == 0 && lse_out) {
+        float lv = (l_r[rb][r] > 0.f) ? m_r[rb][r] + logf(l_r[rb][r])
+                                      : -INFINITY;
+        lse_out[bh * (long)S + qrow] = lv;
+      }
     }
-    if (ccol == 0 && lse_out) {
-      float lv = (l_r[r] > 0.f) ? m_r[r] + logf(l_r[r]) : -INFINITY;
-      lse_out[bh * (long)S + qrow] = lv;
-    }
-  }
 }
 
 // ===========================================================================
@@ -861,7 +877,7 @@ void check_attn_tensor(const torch::Tensor& t, int d_dim, int D) {
 
 void launch_fwd(Strided q, Strided k, Strided v, StridedMut o, float* lse,
                 int B, int H, int S, int D, float scale, bool causal) {
-  int q_blocks = (S + QTILE - 1) / QTILE;
+  int q_blocks = (S + QTILE * 2 - 1) / (QTILE * 2);  // RB=2 home tiles
   int kv_total = (S + TILE - 1) / TILE;
   dim3 grid(q_blocks, B * H);
   auto stream = at::hip::getCurrentHIPStream();
