@@ -56,6 +56,7 @@ class GPTForPretrainingPipe(PipelineModule):
                  recompute_granularity: str = "full",
                  sequence_parallel: bool = False,
                  initializer_range: float = 0.02,
+                 virtual_pp_degree: int = 1,
                  dtype: Optional[torch.dtype] = None, **unused: Any):
         ffn_hidden_size = ffn_hidden_size or 4 * hidden_size
         if sequence_parallel:
@@ -86,5 +87,6 @@ class GPTForPretrainingPipe(PipelineModule):
                                      hidden_size=hidden_size, dtype=dtype,
                                      init_std=initializer_range))
         super().__init__(descs, seg_method="layer:TransformerDecoderLayer",
-                         act_dtype=dtype or torch.float32)
+                         act_dtype=dtype or torch.float32,
+                         num_virtual_stages=virtual_pp_degree)
         self.hidden_size = hidden_size

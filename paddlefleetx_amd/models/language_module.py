@@ -94,7 +94,10 @@ class GPTModule(LanguageModule):
         if hcg.get_pipe_parallel_world_size() > 1:
             from paddlefleetx_amd.models.gpt.pipeline_model import \
                 GPTForPretrainingPipe
-            return GPTForPretrainingPipe(dtype=dtype, **mcfg)
+            vpp = int(cfg.get("Distributed", {}).get("pipeline", {})
+                      .get("virtual_pp_degree", 1) or 1)
+            return GPTForPretrainingPipe(dtype=dtype, virtual_pp_degree=vpp,
+                                         **mcfg)
         return GPTForPretraining(GPTModel(dtype=dtype, **mcfg))
 
     def get_loss_fn(self):

@@ -9,6 +9,15 @@ eager_engine.py:507-517). This is a from-scratch implementation:
   layers pinned to the edge stages).
 - Non-interleaved 1F1B schedule: warmup fwd, steady 1F1B, cooldown bwd;
   micro-batch count = accumulate_steps.
+- Interleaved (virtual-stage) 1F1B: `num_virtual_stages` splits each
+  rank's layers into V model chunks placed round-robin over the pp ring
+  (global chunk g on rank g%P), shrinking the pipeline bubble by ~V
+  (reference `num_virtual_pipeline_stages`, hybrid_model.py:1084 /
+  models/language_model/utils.py:88-119 divisibility rules). Schedule
+  and comm-slot structure follow the standard interleaved 1F1B: each
+  slot's sends and next-slot receives are fused in one
+  batch_isend_irecv so p2p ops can never deadlock on a stream-blocking
+  backend.
 - Activation/grad exchange via dist.batch_isend_irecv on the pp axis
   (xGMI p2p on a single node), static [micro_b, s, h] bf16 buffers.
 - Tied first/last embedding: weight broadcast at init + grad allreduce
@@ -87,7 +96,8 @@ class PipelineModule(nn.Module):
     """Builds only this pp stage's layers from the full desc list."""
 
     def __init__(self, descs: List[LayerDesc], loss_shape_hint=None,
-                 seg_method: str = "uniform", act_dtype=torch.bfloat16):
+                 seg_method: str = "uniform", act_dtype=torch.bfloat16,
+                 num_virtual_stages: int = 1):
         super().__init__()
         hcg = get_hcg()
         self.hcg = hcg
@@ -95,56 +105,77 @@ class PipelineModule(nn.Module):
         self.pp_size = hcg.get_pipe_parallel_world_size()
         self.act_dtype = act_dtype
         self.descs = descs
+        self.num_virtual = int(num_virtual_stages or 1)
+        if self.num_virtual > 1 and self.pp_size == 1:
+            # virtual stages only make sense with a real pipeline
+            # (reference utils.py:96-99)
+            self.num_virtual = 1
 
-        start, end = self._partition(descs, seg_method)
-        self.stage_start, self.stage_end = start, end
+        bounds = self._partition_chunks(
+            descs, seg_method, self.pp_size * self.num_virtual)
         self.layers = nn.ModuleList()
+        self._chunk_bounds: List[Tuple[int, int]] = []
+        self._layer_desc_idx: List[int] = []  # local layer -> desc index
         self._shared_keys: Dict[str, Tuple[nn.Module, str]] = {}
-        for i in range(start, end):
-            d = descs[i]
-            m = d.build()
-            if isinstance(d, SharedLayerDesc):
-                if d.key in self._shared_keys:
-                    # reuse the first-built module's weight (same stage)
-                    first, attr = self._shared_keys[d.key]
-                    _set_dotted(m, attr, _get_dotted(first, attr))
-                else:
-                    self._shared_keys[d.key] = (m, d.shared_weight_attr)
-                if d.forward_func is not None:
-                    m = _FuncLayer(m, d.forward_func)
-            self.layers.append(m)
-        logger.info(f"pp stage {self.pp_rank}/{self.pp_size}: layers "
-                    f"[{start}, {end}) of {len(descs)}")
+        for v in range(self.num_virtual):
+            g = v * self.pp_size + self.pp_rank
+            lo, hi = bounds[g]
+            a = len(self.layers)
+            for i in range(lo, hi):
+                d = descs[i]
+                m = d.build()
+                if isinstance(d, SharedLayerDesc):
+                    if d.key in self._shared_keys:
+                        # reuse the first-built module's weight (same stage)
+                        first, attr = self._shared_keys[d.key]
+                        _set_dotted(m, attr, _get_dotted(first, attr))
+                    else:
+                        self._shared_keys[d.key] = (m, d.shared_weight_attr)
+                    if d.forward_func is not None:
+                        m = _FuncLayer(m, d.forward_func)
+                self.layers.append(m)
+                self._layer_desc_idx.append(i)
+            self._chunk_bounds.append((a, len(self.layers)))
+        self.stage_start, self.stage_end = bounds[self.pp_rank]
+        logger.info(
+            f"pp stage {self.pp_rank}/{self.pp_size} "
+            f"(V={self.num_virtual}): desc slices "
+            + str([bounds[v * self.pp_size + self.pp_rank]
+                   for v in range(self.num_virtual)]) + f" of {len(descs)}")
         self._shared_groups = None  # built lazily after dist init
 
     # ------------------------------------------------------------------
-    def _partition(self, descs, seg_method) -> Tuple[int, int]:
+    def _partition_chunks(self, descs, seg_method,
+                          nchunks: int) -> List[Tuple[int, int]]:
+        """Split the desc list into `nchunks` contiguous slices. With
+        `layer:` segmentation the matched layers are balanced across the
+        chunks; leading descs (embedding) join global chunk 0 and
+        trailing descs (final norm, tied head) join the last chunk."""
         n = len(descs)
         if seg_method.startswith("layer:"):
             pat = seg_method.split(":", 1)[1]
             idx = [i for i, d in enumerate(descs)
                    if re.match(pat, d.name)]
-            per = len(idx) // self.pp_size
-            rem = len(idx) % self.pp_size
-            # distribute matched layers; leading descs join stage 0,
-            # trailing descs join the last stage
-            counts = [per + (1 if s < rem else 0) for s in range(self.pp_size)]
-            bounds = [0]
+            per = len(idx) // nchunks
+            rem = len(idx) % nchunks
+            counts = [per + (1 if s < rem else 0) for s in range(nchunks)]
+            cum = [0]
             for c in counts:
-                bounds.append(bounds[-1] + c)
+                cum.append(cum[-1] + c)
             first_matched = idx[0]
-            lo = first_matched + bounds[self.pp_rank]
-            hi = first_matched + bounds[self.pp_rank + 1]
-            if self.pp_rank == 0:
-                lo = 0
-            if self.pp_rank == self.pp_size - 1:
-                hi = n
-            return lo, hi
-        per = n // self.pp_size
-        rem = n % self.pp_size
-        counts = [per + (1 if s < rem else 0) for s in range(self.pp_size)]
-        lo = sum(counts[:self.pp_rank])
-        return lo, lo + counts[self.pp_rank]
+            bounds = [(first_matched + cum[g], first_matched + cum[g + 1])
+                      for g in range(nchunks)]
+            bounds[0] = (0, bounds[0][1])
+            bounds[-1] = (bounds[-1][0], n)
+            return bounds
+        per = n // nchunks
+        rem = n % nchunks
+        counts = [per + (1 if s < rem else 0) for s in range(nchunks)]
+        bounds, lo = [], 0
+        for c in counts:
+            bounds.append((lo, lo + c))
+            lo += c
+        return bounds
 
     # ------------------------------------------------------------------
     def _build_shared_groups(self):
@@ -199,6 +230,15 @@ class PipelineModule(nn.Module):
             x = layer(*x) if isinstance(x, tuple) else layer(x)
         return x
 
+    def chunk_forward(self, v: int, *inputs):
+        """Run only virtual chunk v's layers."""
+        a, b = self._chunk_bounds[v]
+        x = inputs
+        for i in range(a, b):
+            layer = self.layers[i]
+            x = layer(*x) if isinstance(x, tuple) else layer(x)
+        return x
+
     # ------------------------------------------------------------------
     # 1F1B schedule
     # ------------------------------------------------------------------
@@ -213,6 +253,9 @@ class PipelineModule(nn.Module):
     def forward_backward_pipeline(self, batch, loss_fn, accumulate_steps: int,
                                   scale: float = 1.0) -> torch.Tensor:
         """batch = (tokens, position_ids, labels, loss_mask) full local batch."""
+        if self.num_virtual > 1:
+            return self._fb_interleaved(batch, loss_fn, accumulate_steps,
+                                        scale)
         self._build_shared_groups()
         hcg = self.hcg
         pp_group = hcg.get_pipe_parallel_group()
@@ -350,9 +393,271 @@ class PipelineModule(nn.Module):
             dist.broadcast(loss, src=src, group=pp_group.group)
         return loss.cpu()
 
+    # ------------------------------------------------------------------
+    # Interleaved (virtual-stage) 1F1B schedule
+    # ------------------------------------------------------------------
+    def _fb_interleaved(self, batch, loss_fn, accumulate_steps: int,
+                        scale: float = 1.0) -> torch.Tensor:
+        """Interleaved 1F1B over V virtual chunks per rank. Comm is always
+        with the ring neighbours; each schedule slot fuses its sends with
+        the next slot's receives in one batch_isend_irecv (deadlock-free
+        on stream-blocking backends). Mirrors the standard interleaved
+        schedule bookkeeping (per-chunk FIFO stores, slot-indexed chunk
+        ids)."""
+        self._build_shared_groups()
+        hcg = self.hcg
+        P, V, r = self.pp_size, self.num_virtual, self.pp_rank
+        M = accumulate_steps
+        assert M % P == 0, (
+            f"virtual pipeline requires accumulate_steps ({M}) divisible "
+            f"by pp_degree ({P})")
+        prev = hcg._pp_global_rank((r - 1) % P)
+        nxt = hcg._pp_global_rank((r + 1) % P)
+        device = next(self.parameters()).device
+
+        tokens, position_ids, labels, loss_mask = batch
+        micro_b = tokens.shape[0] // M
+        seq = tokens.shape[1]
+        shape = (micro_b, seq, self.hidden_size)
+        total = M * V
+
+        def micro(t, i):
+            return t[i * micro_b:(i + 1) * micro_b]
+
+        def fwd_chunk(k):
+            return (k % (P * V)) // P
+
+        def bwd_chunk(k):
+            return V - 1 - (k % (P * V)) // P
+
+        first_global = r == 0
+        last_global = r == P - 1
+
+        def is_first_chunk(v):
+            return first_global and v == 0
+
+        def is_last_chunk(v):
+            return last_global and v == V - 1
+
+        input_store: List[List] = [[] for _ in range(V)]
+        output_store: List[List] = [[] for _ in range(V)]
+        grad_store: List[List] = [[] for _ in range(V)]
+        fwd_micro = [0] * V
+        bwd_micro = [0] * V
+        losses: List[torch.Tensor] = []
+
+        def comm(send_fwd=None, send_bwd=None, recv_prev=False,
+                 recv_next=False):
+            """One fused p2p slot: act send to nxt / grad send to prev /
+            act recv from prev / grad recv from nxt."""
+            ops = []
+            new_in = new_grad = None
+            if send_fwd is not None:
+                ops.append(dist.P2POp(dist.isend,
+                                      send_fwd.detach().contiguous(), nxt))
+            if send_bwd is not None:
+                ops.append(dist.P2POp(dist.isend, send_bwd.contiguous(),
+                                      prev))
+            if recv_prev:
+                new_in = torch.empty(shape, dtype=self.act_dtype,
+                                     device=device)
+                ops.append(dist.P2POp(dist.irecv, new_in, prev))
+            if recv_next:
+                new_grad = torch.empty(shape, dtype=self.act_dtype,
+                                       device=device)
+                ops.append(dist.P2POp(dist.irecv, new_grad, nxt))
+            if ops:
+                self._p2p(ops)
+            if new_in is not None:
+                new_in.requires_grad_(True)
+            return new_in, new_grad
+
+        def forward_helper(k):
+            v = fwd_chunk(k)
+            if is_first_chunk(v) and \
+                    len(input_store[v]) == len(output_store[v]):
+                input_store[v].append(None)
+            inp = input_store[v][-1]
+            i = fwd_micro[v]
+            fwd_micro[v] += 1
+            if is_first_chunk(v):
+                out = self.chunk_forward(v, micro(tokens, i),
+                                         micro(position_ids, i))
+            else:
+                out = self.chunk_forward(v, inp)
+            if is_last_chunk(v):
+                out = loss_fn(out, micro(labels, i), micro(loss_mask, i))
+                losses.append(out.detach())
+            output_store[v].append(out)
+            return out
+
+        def backward_helper(k):
+            v = bwd_chunk(k)
+            if is_last_chunk(v) and len(grad_store[v]) == 0:
+                grad_store[v].append(None)
+            inp = input_store[v].pop(0)
+            out = output_store[v].pop(0)
+            g = grad_store[v].pop(0)
+            bwd_micro[v] += 1
+            if is_last_chunk(v):
+                (out * (scale / M)).backward()
+            else:
+                out.backward(gradient=g)
+            return inp.grad if inp is not None else None
+
+        if M == P:
+            num_warmup = total
+            all_warmup = True
+        else:
+            num_warmup = min(total, (P - r - 1) * 2 + (V - 1) * P)
+            all_warmup = False
+        remaining = total - num_warmup
+
+        # prime the first input for chunk 0
+        if first_global:
+            pass  # forward_helper feeds chunk 0 from the batch
+        else:
+            ni, _ = comm(recv_prev=True)
+            input_store[0].append(ni)
+
+        # ---- warmup: forwards only -----------------------------------
+        for k in range(num_warmup):
+            v = fwd_chunk(k)
+            out = forward_helper(k)
+            send_out = None if is_last_chunk(v) else out
+            next_v = fwd_chunk(k + 1)
+            recv_prev = True
+            if first_global and next_v == 0:
+                recv_prev = False
+            if k == total - 1:
+                recv_prev = False
+            if k == num_warmup - 1 and not all_warmup:
+                # transition into steady 1F1B: also pull the first grad
+                recv_next = not last_global
+                ni, ng = comm(send_fwd=send_out, recv_prev=recv_prev,
+                              recv_next=recv_next)
+                grad_store[V - 1].append(ng)
+            else:
+                ni, _ = comm(send_fwd=send_out, recv_prev=recv_prev)
+            if recv_prev:
+                input_store[next_v].append(ni)
+
+        # ---- steady 1F1B ---------------------------------------------
+        for k in range(remaining):
+            forward_k = k + num_warmup
+            fv = fwd_chunk(forward_k)
+            out = forward_helper(forward_k)
+            in_grad = backward_helper(k)
+            bv = bwd_chunk(k)
+
+            send_out = None if is_last_chunk(fv) else out
+            send_grad = None if is_first_chunk(bv) else in_grad
+
+            recv_prev = True
+            if first_global:
+                # the act arriving now was produced by the last rank
+                # (P-1) slots ago; if that was the final chunk it went to
+                # the loss instead
+                nf = fwd_chunk(forward_k - (P - 1))
+                if nf == V - 1:
+                    recv_prev = False
+                store_f = nf + 1
+            else:
+                store_f = fwd_chunk(forward_k + 1)
+            if k == remaining - 1:
+                recv_prev = False
+
+            recv_next = True
+            if last_global:
+                nb = bwd_chunk(k - (P - 1))
+                if nb == 0:
+                    recv_next = False
+                store_b = nb - 1
+            else:
+                store_b = bwd_chunk(k + 1)
+
+            ni, ng = comm(send_fwd=send_out, send_bwd=send_grad,
+                          recv_prev=recv_prev, recv_next=recv_next)
+            if recv_prev:
+                input_store[store_f].append(ni)
+            if recv_next:
+                grad_store[store_b].append(ng)
+
+        # ---- cooldown: backwards only --------------------------------
+        if all_warmup:
+            if last_global:
+                grad_store[V - 1].append(None)
+            else:
+                _, ng = comm(recv_next=True)
+                grad_store[V - 1].append(ng)
+        for k in range(remaining, total):
+            in_grad = backward_helper(k)
+            bv = bwd_chunk(k)
+            store_b = bwd_chunk(k + 1)
+            recv_next = True
+            if last_global and store_b == V - 1:
+                recv_next = False
+            if k == total - 1:
+                recv_next = False
+            send_grad = None if is_first_chunk(bv) else in_grad
+            _, ng = comm(send_bwd=send_grad, recv_next=recv_next)
+            if recv_next:
+                grad_store[store_b].append(ng)
+
+        self.sync_shared_grads()
+
+        if last_global and losses:
+            loss = torch.stack(losses).mean()
+        else:
+            loss = torch.zeros((), device=device)
+        if dist.is_initialized() and P > 1:
+            dist.broadcast(loss, src=hcg._pp_global_rank(P - 1),
+                           group=hcg.get_pipe_parallel_group().group)
+        return loss.cpu()
+
+    @torch.no_grad()
+    def _eval_interleaved(self, batch, loss_fn, accumulate_steps: int):
+        """Forward-only walk of the virtual chunks, one micro-batch at a
+        time (depth-first; eval is not throughput-critical)."""
+        self._build_shared_groups()
+        hcg = self.hcg
+        P, V, r = self.pp_size, self.num_virtual, self.pp_rank
+        prev = hcg._pp_global_rank((r - 1) % P)
+        nxt = hcg._pp_global_rank((r + 1) % P)
+        M = accumulate_steps
+        device = next(self.parameters()).device
+        tokens, position_ids, labels, loss_mask = batch
+        micro_b = tokens.shape[0] // M
+        shape = (micro_b, tokens.shape[1], self.hidden_size)
+        losses = []
+        for i in range(M):
+            sl = slice(i * micro_b, (i + 1) * micro_b)
+            for v in range(V):
+                if r == 0 and v == 0:
+                    out = self.chunk_forward(v, tokens[sl], position_ids[sl])
+                else:
+                    buf = torch.empty(shape, dtype=self.act_dtype,
+                                      device=device)
+                    self._p2p([dist.P2POp(dist.irecv, buf, prev)])
+                    out = self.chunk_forward(v, buf)
+                if r == P - 1 and v == V - 1:
+                    losses.append(loss_fn(out, labels[sl], loss_mask[sl]))
+                else:
+                    self._p2p([dist.P2POp(dist.isend, out.contiguous(), nxt)])
+        if r == P - 1 and losses:
+            loss = torch.stack(losses).mean()
+        else:
+            loss = torch.zeros((), device=device)
+        if dist.is_initialized() and P > 1:
+            dist.broadcast(loss, src=hcg._pp_global_rank(P - 1),
+                           group=hcg.get_pipe_parallel_group().group)
+        return loss.cpu()
+
     @torch.no_grad()
     def eval_pipeline(self, batch, loss_fn, accumulate_steps: int):
         """Forward-only pipeline for evaluation."""
+        if self.num_virtual > 1:
+            return self._eval_interleaved(batch, loss_fn, accumulate_steps)
         self._build_shared_groups()
         hcg = self.hcg
         prev, nxt = hcg.pp_prev_rank(), hcg.pp_next_rank()
