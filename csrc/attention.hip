@@ -170,8 +170,9 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
   constexpr int KSZ = TILE * KRS;       // one K buffer (shorts)
   constexpr int VSZ = D * VRS;          // one V^T buffer
 
-  // single __shared__ object (guide §5 trap 4a)
-  __shared__ unsigned short smem[2 * KSZ + 2 * VSZ + FW_WAVES * 16 * PRS];
+  // single __shared__ object (guide §5 trap 4a); P buffer holds BOTH
+  // 16-row groups of the wave (RB=2) so PV can share each V read
+  __shared__ unsigned short smem[2 * KSZ + 2 * VSZ + FW_WAVES * 32 * PRS];
   unsigned short* k_lds = smem;                  // [2][KSZ]
   unsigned short* vt_lds = smem + 2 * KSZ;       // [2][VSZ]
   unsigned short* p_lds = smem + 2 * KSZ + 2 * VSZ;
@@ -212,7 +213,7 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
   const int kv_tiles = CAUSAL
       ? min(kv_total, (qt * QTILE * RB + QTILE * RB - 1) / TILE + 1)
       : kv_total;
-  unsigned short* myp = p_lds + wid * 16 * PRS;
+  unsigned short* myp = p_lds + wid * 32 * PRS;
 
   // ---- staging thread maps ----
   // K (row-major image): D/16 threads per row, 16 shorts each
@@ -269,6 +270,11 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
   load_kv(0);
   write_kv(k_lds, vt_lds);
   __syncthreads();
+  // T5 static form: the younger dispatch half loses VALU arbitration to
+  // the older half; one wave-uniform s_setprio(1) before the loop levels
+  // it (guide T5 static form; readfirstlane keeps it scalar-branch).
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= FW_BLOCKT / 2)
+    __builtin_amdgcn_s_setprio(1);
 
   for (int kt = 0; kt < kv_tiles; ++kt) {
     const int kv0 = kt * TILE;
@@ -280,93 +286,120 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
     const bool have_next = (kt + 1) < kv_tiles;
     if (have_next) load_kv((kt + 1) * TILE);
 
+    // waves whose rows are entirely above this kv tile skip compute
+    // (they still stage and hit the barrier)
+    const bool active = !CAUSAL ||
+        (kv0 <= min(qrow0 + RB * 16 - 1, S - 1));
+    if (active) {
+      // ---- QK^T: both 16-row groups share each K B-fragment read; the
+      // (ct,kc) stream is software-pipelined 2 deep so ds_read latency
+      // hides under the previous fragment's 2 MFMAs ----
+      f4 s[RB][4];
+      bf8 kbuf[3];
+      kbuf[0] = read_b_frag(kb_lds, KRS, 0, 0, lane);
+      kbuf[1] = read_b_frag(kb_lds, KRS, 0, 32, lane);
 #pragma unroll
-    for (int rb = 0; rb < RB; ++rb) {
-      const int my_qrow = qrow0 + rb * 16 + crow4;
-      // waves whose rows are entirely above this kv tile skip compute
-      // (they still stage and hit the barrier)
-      const bool active = !CAUSAL ||
-          (kv0 <= min(qrow0 + rb * 16 + 15, S - 1));
-      if (!active) continue;
-      // ---- QK^T ----
-      f4 s[4];
-#pragma unroll
-      for (int ct = 0; ct < 4; ++ct) {
-        f4 acc = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-        for (int kc = 0; kc < D / 32; ++kc) {
-          bf8 kb = read_b_frag(kb_lds, KRS, ct * 16, kc * 32, lane);
-          acc = MFMA_BF16(qfrag[rb][kc], kb, acc);
+      for (int i = 0; i < 4 * (D / 32); ++i) {
+        const int ct = i / (D / 32), kc = i % (D / 32);
+        if (i + 2 < 4 * (D / 32)) {
+          const int j = i + 2;
+          kbuf[j % 3] = read_b_frag(kb_lds, KRS, (j / (D / 32)) * 16,
+                                    (j % (D / 32)) * 32, lane);
         }
-        s[ct] = acc;
+        const bf8 kb = kbuf[i % 3];
+#pragma unroll
+        for (int rb = 0; rb < RB; ++rb) {
+          f4 acc = (kc == 0) ? f4{0.f, 0.f, 0.f, 0.f} : s[rb][ct];
+          s[rb][ct] = MFMA_BF16(qfrag[rb][kc], kb, acc);
+        }
       }
 
-      // ---- mask + scale + online softmax ----
-      float pmax[4] = {-INFINITY, -INFINITY, -INFINITY, -INFINITY};
+      // ---- mask + scale + online softmax (both row groups: 2x VALU ILP) --
 #pragma unroll
-      for (int ct = 0; ct < 4; ++ct) {
-        int kcol = kv0 + ct * 16 + ccol;
+      for (int rb = 0; rb < RB; ++rb) {
+        const int my_qrow = qrow0 + rb * 16 + crow4;
+        float pmax[4] = {-INFINITY, -INFINITY, -INFINITY, -INFINITY};
+#pragma unroll
+        for (int ct = 0; ct < 4; ++ct) {
+          int kcol = kv0 + ct * 16 + ccol;
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            float val = s[rb][ct][r] * scale;
+            if ((CAUSAL && kcol > my_qrow + r) || kcol >= S) val = -INFINITY;
+            s[rb][ct][r] = val;
+            pmax[r] = fmaxf(pmax[r], val);
+          }
+        }
+#pragma unroll
+        for (int r = 0; r < 4; ++r) pmax[r] = group16_reduce_max(pmax[r]);
+
+        float alpha[4];
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
-          float val = s[ct][r] * scale;
-          if ((CAUSAL && kcol > my_qrow + r) || kcol >= S) val = -INFINITY;
-          s[ct][r] = val;
-          pmax[r] = fmaxf(pmax[r], val);
+          float nm = fmaxf(m_r[rb][r], pmax[r]);
+          if (nm == -INFINITY) nm = 0.f;
+          alpha[r] = (m_r[rb][r] == -INFINITY) ? 0.f : __expf(m_r[rb][r] - nm);
+          m_r[rb][r] = (m_r[rb][r] == -INFINITY && pmax[r] == -INFINITY)
+              ? -INFINITY : nm;
         }
-      }
-#pragma unroll
-      for (int r = 0; r < 4; ++r) pmax[r] = group16_reduce_max(pmax[r]);
 
-      float alpha[4];
+        float psum[4] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        float nm = fmaxf(m_r[rb][r], pmax[r]);
-        if (nm == -INFINITY) nm = 0.f;
-        alpha[r] = (m_r[rb][r] == -INFINITY) ? 0.f : __expf(m_r[rb][r] - nm);
-        m_r[rb][r] = (m_r[rb][r] == -INFINITY && pmax[r] == -INFINITY)
-            ? -INFINITY : nm;
-      }
-
-      float psum[4] = {0.f, 0.f, 0.f, 0.f};
+        for (int ct = 0; ct < 4; ++ct) {
 #pragma unroll
-      for (int ct = 0; ct < 4; ++ct) {
+          for (int r = 0; r < 4; ++r) {
+            float p = (s[rb][ct][r] == -INFINITY) ? 0.f
+                : __expf(s[rb][ct][r] - m_r[rb][r]);
+            s[rb][ct][r] = p;
+            psum[r] += p;
+          }
+        }
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
-          float p = (s[ct][r] == -INFINITY) ? 0.f
-              : __expf(s[ct][r] - m_r[rb][r]);
-          s[ct][r] = p;
-          psum[r] += p;
+          psum[r] = group16_reduce_sum(psum[r]);
+          l_r[rb][r] = l_r[rb][r] * alpha[r] + psum[r];
         }
-      }
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        psum[r] = group16_reduce_sum(psum[r]);
-        l_r[rb][r] = l_r[rb][r] * alpha[r] + psum[r];
-      }
+        for (int i = 0; i < NDT; ++i) {
 #pragma unroll
-      for (int i = 0; i < NDT; ++i) {
-#pragma unroll
-        for (int r = 0; r < 4; ++r) oacc[rb][i][r] *= alpha[r];
-      }
+          for (int r = 0; r < 4; ++r) oacc[rb][i][r] *= alpha[r];
+        }
 
-      // ---- P -> per-wave LDS for A-fragments ----
+        // ---- P -> per-wave LDS for A-fragments ----
 #pragma unroll
-      for (int ct = 0; ct < 4; ++ct) {
+        for (int ct = 0; ct < 4; ++ct) {
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          myp[(crow4 + r) * PRS + ct * 16 + ccol] = f2bf_raw(s[ct][r]);
+          for (int r = 0; r < 4; ++r) {
+            myp[(rb * 16 + crow4 + r) * PRS + ct * 16 + ccol] =
+                f2bf_raw(s[rb][ct][r]);
+          }
         }
       }
 
-      // ---- PV ----
+      // ---- PV: both row groups share each V B-fragment read, 2-deep
+      // prefetch over the (kc,dt) stream ----
+      bf8 pa[RB][2];
 #pragma unroll
-      for (int kc = 0; kc < 2; ++kc) {
-        bf8 pa = read_a_frag_lds(myp, PRS, kc * 32, lane);
+      for (int rb = 0; rb < RB; ++rb)
 #pragma unroll
-        for (int dt = 0; dt < NDT; ++dt) {
-          bf8 vb = read_b_frag(vb_lds, VRS, dt * 16, kc * 32, lane);
-          oacc[rb][dt] = MFMA_BF16(pa, vb, oacc[rb][dt]);
+        for (int kc = 0; kc < 2; ++kc)
+          pa[rb][kc] = read_a_frag_lds(myp + rb * 16 * PRS, PRS,
+                                       kc * 32, lane);
+      bf8 vbuf[3];
+      vbuf[0] = read_b_frag(vb_lds, VRS, 0, 0, lane);
+      vbuf[1] = read_b_frag(vb_lds, VRS, 16, 0, lane);
+#pragma unroll
+      for (int i = 0; i < 2 * NDT; ++i) {
+        const int kc = i / NDT, dt = i % NDT;
+        if (i + 2 < 2 * NDT) {
+          const int j = i + 2;
+          vbuf[j % 3] = read_b_frag(vb_lds, VRS, (j % NDT) * 16,
+                                    (j / NDT) * 32, lane);
         }
+        const bf8 vb = vbuf[i % 3];
+#pragma unroll
+        for (int rb = 0; rb < RB; ++rb)
+          oacc[rb][dt] = MFMA_BF16(pa[rb][kc], vb, oacc[rb][dt]);
       }
     }
 

@@ -11,6 +11,14 @@ Reference: ppfleetx/models/language_model/ernie/dygraph/single_model.py
 MI355X-native: fused LayerNorm + bias-gelu kernels; bidirectional flash
 attention (gfx950) when there is no padding mask, additive-mask GEMM path
 otherwise.
+
+Tensor parallelism (reference ErnieModelHybrid, ernie/dygraph/
+hybrid_model.py:167 + ernie/layers/distributed_transformer.py): when the
+mp degree is >1 the QKV / out-proj / FFN projections become column/row
+parallel over RCCL, the word embedding is vocab-sharded, and the tied MLM
+decoder computes vocab-parallel logits consumed by ParallelCrossEntropy.
+The per-rank qkv weight rows are laid out [q_shard; k_shard; v_shard] so
+the single-card `view(B, S, 3, h_local, D)` split stays valid.
 """
 
 from __future__ import annotations
@@ -23,6 +31,16 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from paddlefleetx_amd.ops import FusedLayerNorm, bias_gelu
+from paddlefleetx_amd.parallel.env import get_hcg
+from paddlefleetx_amd.parallel.tp import (ColumnParallelLinear,
+                                          ParallelCrossEntropy,
+                                          RowParallelLinear,
+                                          VocabParallelEmbedding,
+                                          parallel_matmul)
+
+
+def _mp_degree() -> int:
+    return get_hcg().get_model_parallel_world_size()
 
 
 class ErnieEmbeddings(nn.Module):
@@ -34,9 +52,14 @@ class ErnieEmbeddings(nn.Module):
                  use_task_id: bool = False,
                  dtype: Optional[torch.dtype] = None):
         super().__init__()
-        self.word_embeddings = nn.Embedding(vocab_size, hidden_size,
-                                            padding_idx=pad_token_id,
-                                            dtype=dtype)
+        if _mp_degree() > 1:
+            # vocab-sharded table; tied MLM decoder uses the same shard
+            self.word_embeddings = VocabParallelEmbedding(
+                vocab_size, hidden_size, dtype=dtype)
+        else:
+            self.word_embeddings = nn.Embedding(vocab_size, hidden_size,
+                                                padding_idx=pad_token_id,
+                                                dtype=dtype)
         self.position_embeddings = nn.Embedding(max_position_embeddings,
                                                 hidden_size, dtype=dtype)
         self.type_vocab_size = type_vocab_size
@@ -81,16 +104,26 @@ class ErnieSelfAttention(nn.Module):
                  dtype: Optional[torch.dtype] = None):
         super().__init__()
         assert hidden_size % num_heads == 0
+        mp = _mp_degree()
+        assert num_heads % mp == 0
         self.num_heads = num_heads
+        self.nh_local = num_heads // mp
         self.head_dim = hidden_size // num_heads
         self.scale = 1.0 / math.sqrt(self.head_dim)
-        self.qkv = nn.Linear(hidden_size, 3 * hidden_size, dtype=dtype)
-        self.out_proj = nn.Linear(hidden_size, hidden_size, dtype=dtype)
+        if mp > 1:
+            # per-rank rows are [q_shard; k_shard; v_shard]
+            self.qkv = ColumnParallelLinear(hidden_size, 3 * hidden_size,
+                                            bias=True, dtype=dtype)
+            self.out_proj = RowParallelLinear(hidden_size, hidden_size,
+                                              bias=True, dtype=dtype)
+        else:
+            self.qkv = nn.Linear(hidden_size, 3 * hidden_size, dtype=dtype)
+            self.out_proj = nn.Linear(hidden_size, hidden_size, dtype=dtype)
         self.attn_dropout_p = attn_dropout
 
     def forward(self, x, attn_mask=None):
         B, S, C = x.shape
-        qkv = self.qkv(x).view(B, S, 3, self.num_heads, self.head_dim)
+        qkv = self.qkv(x).view(B, S, 3, self.nh_local, self.head_dim)
         q, k, v = qkv.unbind(dim=2)
         q, k, v = (t.transpose(1, 2) for t in (q, k, v))
         use_kernel = (x.is_cuda and x.dtype == torch.bfloat16
@@ -107,7 +140,7 @@ class ErnieSelfAttention(nn.Module):
             if self.attn_dropout_p > 0.0 and self.training:
                 probs = F.dropout(probs, p=self.attn_dropout_p)
             o = torch.matmul(probs, v)
-        o = o.transpose(1, 2).reshape(B, S, C)
+        o = o.transpose(1, 2).reshape(B, S, -1)  # [B, S, h_local*D]
         return self.out_proj(o)
 
 
@@ -127,6 +160,17 @@ class ErnieEncoderLayer(nn.Module):
         self.ln2 = FusedLayerNorm(hidden_size, eps=1e-12, dtype=dtype)
         if expert_module is not None:
             self.ffn = expert_module
+        elif _mp_degree() > 1:
+            mp = _mp_degree()
+            self.fc1 = ColumnParallelLinear(hidden_size, intermediate_size,
+                                            bias=False, dtype=dtype)
+            self.fc1_bias = nn.Parameter(
+                torch.zeros(intermediate_size // mp, dtype=dtype))
+            self.fc1_bias.is_mp = True
+            self.fc1_bias.partition_dim = 0
+            self.fc2 = RowParallelLinear(intermediate_size, hidden_size,
+                                         bias=True, dtype=dtype)
+            self.ffn = None
         else:
             self.fc1 = nn.Linear(hidden_size, intermediate_size, bias=False,
                                  dtype=dtype)
@@ -233,19 +277,28 @@ class ErnieLMPredictionHead(nn.Module):
         self.transform = nn.Linear(hidden_size, hidden_size, dtype=dtype)
         self.activation = getattr(F, activation)
         self.layer_norm = FusedLayerNorm(hidden_size, eps=1e-12, dtype=dtype)
+        self.mp = _mp_degree()
         if embedding_weights is not None:
             self.decoder_weight = embedding_weights
         else:
             self.decoder_weight = nn.Parameter(
                 torch.empty(vocab_size, hidden_size, dtype=dtype))
             nn.init.normal_(self.decoder_weight, std=0.02)
-        self.decoder_bias = nn.Parameter(torch.zeros(vocab_size, dtype=dtype))
+        bias_n = vocab_size // self.mp if self.mp > 1 else vocab_size
+        self.decoder_bias = nn.Parameter(torch.zeros(bias_n, dtype=dtype))
+        if self.mp > 1:
+            self.decoder_bias.is_mp = True
+            self.decoder_bias.partition_dim = 0
 
     def forward(self, hidden_states, masked_positions=None):
         if masked_positions is not None:
             hs = hidden_states.reshape(-1, hidden_states.shape[-1])
             hidden_states = hs.index_select(0, masked_positions)
         h = self.layer_norm(self.activation(self.transform(hidden_states)))
+        if self.mp > 1:
+            # vocab-parallel logits [.., V/mp] for ParallelCrossEntropy
+            return parallel_matmul(h, self.decoder_weight,
+                                   parallel_output=True) + self.decoder_bias
         return F.linear(h, self.decoder_weight) + self.decoder_bias
 
 
@@ -288,12 +341,23 @@ class ErniePretrainingCriterion(nn.Module):
     def __init__(self, with_nsp_loss: bool = True):
         super().__init__()
         self.with_nsp_loss = with_nsp_loss
+        self.parallel_ce = ParallelCrossEntropy(ignore_index=-1) \
+            if _mp_degree() > 1 else None
 
     def forward(self, prediction_scores, seq_relationship_score,
                 masked_lm_labels, next_sentence_labels=None):
-        mlm = F.cross_entropy(
-            prediction_scores.float().reshape(-1, prediction_scores.shape[-1]),
-            masked_lm_labels.reshape(-1), ignore_index=-1)
+        if self.parallel_ce is not None:
+            flat_labels = masked_lm_labels.reshape(-1)
+            loss_vec = self.parallel_ce(
+                prediction_scores.reshape(-1, prediction_scores.shape[-1]),
+                flat_labels)
+            nvalid = (flat_labels != -1).sum().clamp(min=1)
+            mlm = loss_vec.float().sum() / nvalid.float()
+        else:
+            mlm = F.cross_entropy(
+                prediction_scores.float().reshape(
+                    -1, prediction_scores.shape[-1]),
+                masked_lm_labels.reshape(-1), ignore_index=-1)
         if not self.with_nsp_loss:
             return mlm
         nsp = F.cross_entropy(seq_relationship_score.float().reshape(-1, 2),

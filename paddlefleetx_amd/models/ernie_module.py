@@ -19,10 +19,22 @@ from paddlefleetx_amd.models.language_module import (LanguageModule,
 from paddlefleetx_amd.parallel.env import get_hcg
 
 
-def _ernie_model(cfg) -> ErnieModel:
+def _ernie_cfg(cfg) -> dict:
     mcfg = dict(cfg["Model"])
     for k in ("name", "module", "num_classes"):
         mcfg.pop(k, None)
+    mp = get_hcg().get_model_parallel_world_size()
+    if mp > 1:
+        from paddlefleetx_amd.models.language_module import \
+            vocab_size_with_padding
+        mcfg["vocab_size"] = vocab_size_with_padding(
+            mcfg.get("vocab_size", 18000),
+            mcfg.pop("vocab_size_divisible_unit", 128), mp)
+    return mcfg
+
+
+def _ernie_model(cfg) -> ErnieModel:
+    mcfg = _ernie_cfg(cfg)
     moe_configs = mcfg.pop("moe_configs", None)
     if moe_configs:
         hcg = get_hcg()
@@ -34,9 +46,24 @@ def _ernie_model(cfg) -> ErnieModel:
 
 class ErnieModule(LanguageModule):
     def get_model(self):
+        hcg = get_hcg()
+        if hcg.get_pipe_parallel_world_size() > 1:
+            from paddlefleetx_amd.models.ernie.pipeline_model import \
+                ErnieForPretrainingPipe
+            cfg = self.configs
+            vpp = int(cfg.get("Distributed", {}).get("pipeline", {})
+                      .get("virtual_pp_degree", 1) or 1)
+            mcfg = _ernie_cfg(cfg)
+            mcfg.pop("moe_configs", None)
+            return ErnieForPretrainingPipe(dtype=_model_dtype(cfg),
+                                           virtual_pp_degree=vpp, **mcfg)
         return ErnieForPretraining(_ernie_model(self.configs))
 
     def get_loss_fn(self):
+        if get_hcg().get_pipe_parallel_world_size() > 1:
+            from paddlefleetx_amd.models.ernie.pipeline_model import \
+                ErniePipeCriterion
+            return ErniePipeCriterion()
         return ErniePretrainingCriterion(with_nsp_loss=True)
 
     def training_step(self, batch):
