@@ -99,10 +99,15 @@ class ImagenModel(nn.Module):
                  timesteps: int = 1000, cond_drop_prob: float = 0.1,
                  text_encoder_kwargs: Optional[dict] = None,
                  unet_kwargs: Optional[dict] = None,
-                 freeze_text_encoder: bool = True, **unused):
+                 freeze_text_encoder: bool = True,
+                 dynamic_thresholding: bool = True,
+                 dynamic_thresholding_percentile: float = 0.95, **unused):
         super().__init__()
         self.image_size = image_size
         self.cond_drop_prob = cond_drop_prob
+        # reference modeling.py:289-292, 378-383
+        self.dynamic_thresholding = dynamic_thresholding
+        self.dynamic_thresholding_percentile = dynamic_thresholding_percentile
         if unet is None:
             unet = Unet(text_embed_dim=text_embed_dim, **(unet_kwargs or {}))
         self.unet = unet
@@ -153,7 +158,16 @@ class ImagenModel(nn.Module):
             eps = self.unet(x, t, text_embeds=text_embeds,
                             text_mask=text_mask)
             x0 = self.scheduler.predict_start_from_noise(x, t, eps)
-            x0 = x0.clamp(-1, 1)
+            if self.dynamic_thresholding:
+                # per-sample percentile of |x0|, clamp + renormalize
+                # (reference modeling.py:378-383)
+                s = torch.quantile(
+                    x0.reshape(x0.shape[0], -1).abs().float(),
+                    self.dynamic_thresholding_percentile, dim=-1)
+                s = s.clamp(min=1.0)[:, None, None, None].to(x0.dtype)
+                x0 = x0.clamp(-s, s) / s
+            else:
+                x0 = x0.clamp(-1, 1)
             mean, var = self.scheduler.q_posterior_mean(x0, x, t, t_next)
             noise = torch.randn_like(x) if i < steps - 1 else 0
             x = mean + var.sqrt()[:, None, None, None] * noise
