@@ -74,6 +74,9 @@ class EagerEngine(BasicEngine):
 
         self.device = torch.device("cuda") if torch.cuda.is_available() \
             else torch.device("cpu")
+        if self.device.type == "cuda":
+            from paddlefleetx_amd.utils.tunable import enable_tuned_gemms
+            enable_tuned_gemms()
         self.module.to(self.device)
 
         self.is_pipeline = self.hcg.get_pipe_parallel_world_size() > 1
