@@ -147,7 +147,22 @@ class EagerEngine(BasicEngine):
 
     # ------------------------------------------------------------------
     def _sync_params(self):
-        """Broadcast params from dp-rank-0 (reference strategy.py:43 sync_params_buffers)."""
+        """Broadcast params from dp-rank-0 (reference strategy.py:43 sync_params_buffers).
+        Also broadcasts mp-REPLICATED params (no `is_mp` mark: LN, biases,
+        position tables) over the mp group so all TP ranks start from one
+        copy regardless of init-stream discipline."""
+        mp_g = self.hcg.get_model_parallel_group()
+        if mp_g.world_size > 1 and dist.is_initialized():
+            from paddlefleetx_amd.models.moe.moe_layer import MoELayer
+            expert_params = set()
+            for m in self.module.model.modules():
+                if isinstance(m, MoELayer):
+                    for p in m.experts.parameters():
+                        expert_params.add(id(p))
+            for p in self.module.model.parameters():
+                if getattr(p, "is_mp", False) or id(p) in expert_params:
+                    continue
+                dist.broadcast(p.data, src=mp_g.ranks[0], group=mp_g.group)
         from paddlefleetx_amd.parallel.zero3 import Stage3AdamW
         if isinstance(self.optimizer, Stage3AdamW):
             # stage 3: shards live per rank; sync shards over dp only

@@ -258,8 +258,10 @@ class GPTEmbeddings(nn.Module):
                                                       init_std=init_std)
         self.position_embeddings = nn.Embedding(max_position_embeddings,
                                                 hidden_size, dtype=dtype)
-        with model_parallel_rng():
-            nn.init.normal_(self.position_embeddings.weight, std=init_std)
+        # REPLICATED across mp ranks -> must init from the global stream,
+        # not the per-mp-rank local_seed (else SP's seq allgather mixes
+        # inconsistent copies; reference env.py:34-98 seed discipline)
+        nn.init.normal_(self.position_embeddings.weight, std=init_std)
         self.dropout_p = dropout
         self.sequence_parallel = sequence_parallel
 

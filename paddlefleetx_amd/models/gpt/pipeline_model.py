@@ -25,22 +25,34 @@ class EmbeddingPipe(GPTEmbeddings):
 
 
 class LayerNormPipe(FusedLayerNorm):
-    pass
+    def __init__(self, *args, sequence_parallel: bool = False, **kwargs):
+        super().__init__(*args, **kwargs)
+        if sequence_parallel:
+            from paddlefleetx_amd.parallel import sp as sp_ops
+            sp_ops.mark_as_sp_param(self.weight)
+            sp_ops.mark_as_sp_param(self.bias)
 
 
 class TiedLogitsPipe(nn.Module):
     """Last-stage copy of the word-embedding table used as the LM head;
-    weight synced with stage 0 by PipelineModule's shared-key machinery."""
+    weight synced with stage 0 by PipelineModule's shared-key machinery.
+    Under SP the [s/mp, b, h] activation is gathered back to the full
+    sequence before the logits matmul (reference hybrid_model.py:891-892
+    final GatherOp)."""
 
     def __init__(self, vocab_size: int, hidden_size: int,
                  dtype: Optional[torch.dtype] = None, init_std: float = 0.02,
-                 **unused):
+                 sequence_parallel: bool = False, **unused):
         super().__init__()
         from paddlefleetx_amd.parallel.tp import VocabParallelEmbedding
         self.word_embeddings = VocabParallelEmbedding(
             vocab_size, hidden_size, dtype=dtype, init_std=init_std)
+        self.sequence_parallel = sequence_parallel
 
     def forward(self, x):
+        if self.sequence_parallel:
+            from paddlefleetx_amd.parallel import sp as sp_ops
+            x = sp_ops.gather_from_sp_region(x)  # [s/mp, b, h] -> [b, S, h]
         return parallel_matmul(x, self.word_embeddings.weight,
                                parallel_output=True)
 
@@ -59,12 +71,14 @@ class GPTForPretrainingPipe(PipelineModule):
                  virtual_pp_degree: int = 1,
                  dtype: Optional[torch.dtype] = None, **unused: Any):
         ffn_hidden_size = ffn_hidden_size or 4 * hidden_size
+        from paddlefleetx_amd.parallel.env import get_hcg
+        mp = get_hcg().get_model_parallel_world_size()
         if sequence_parallel:
-            raise NotImplementedError(
-                "sequence_parallel with pipeline: round-2 item")
+            assert mp > 1, "sequence_parallel requires mp_degree > 1"
         embed_kwargs = dict(vocab_size=vocab_size, hidden_size=hidden_size,
                             max_position_embeddings=max_position_embeddings,
                             dropout=hidden_dropout_prob, dtype=dtype,
+                            sequence_parallel=sequence_parallel,
                             init_std=initializer_range)
         descs = [
             SharedLayerDesc("embed", EmbeddingPipe,
@@ -77,16 +91,29 @@ class GPTForPretrainingPipe(PipelineModule):
                 ffn_hidden_size, hidden_dropout=hidden_dropout_prob,
                 attn_dropout=attention_probs_dropout_prob,
                 fused_attn=fused_attn, dtype=dtype,
+                sequence_parallel=sequence_parallel,
                 init_std=initializer_range, num_layers_for_scale=num_layers,
                 use_recompute=use_recompute,
                 recompute_granularity=recompute_granularity))
-        descs.append(LayerDesc(LayerNormPipe, hidden_size, dtype=dtype))
+        descs.append(LayerDesc(LayerNormPipe, hidden_size, dtype=dtype,
+                               sequence_parallel=sequence_parallel))
         descs.append(SharedLayerDesc("embed", TiedLogitsPipe,
                                      shared_weight_attr="word_embeddings.weight",
                                      vocab_size=vocab_size,
                                      hidden_size=hidden_size, dtype=dtype,
+                                     sequence_parallel=sequence_parallel,
                                      init_std=initializer_range))
         super().__init__(descs, seg_method="layer:TransformerDecoderLayer",
                          act_dtype=dtype or torch.float32,
                          num_virtual_stages=virtual_pp_degree)
         self.hidden_size = hidden_size
+        self.sequence_parallel = sequence_parallel
+        self._sp_degree = mp
+
+    def _comm_shape(self, micro_b, seq, hidden):
+        """Stage-boundary activation shape: [s/mp, b, h] under SP
+        (sequence stays sharded across the whole stack)."""
+        if self.sequence_parallel:
+            assert seq % self._sp_degree == 0
+            return (seq // self._sp_degree, micro_b, hidden)
+        return (micro_b, seq, hidden)

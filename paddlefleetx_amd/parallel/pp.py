@@ -418,7 +418,7 @@ class PipelineModule(nn.Module):
         tokens, position_ids, labels, loss_mask = batch
         micro_b = tokens.shape[0] // M
         seq = tokens.shape[1]
-        shape = (micro_b, seq, self.hidden_size)
+        shape = self._comm_shape(micro_b, seq, self.hidden_size)
         total = M * V
 
         def micro(t, i):
