@@ -684,8 +684,10 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dkv_kernel(
 }
 
 // ===========================================================================
-// Backward dQ v2: 8 waves, 128 q rows per block; 64-row kv tiles
-// double-buffered in LDS with issue-early staging, one barrier per tile.
+// Backward dQ v3: 8 waves, 256 q rows per block (RB=2 16-row groups per
+// wave — every K/V/K^T B-fragment LDS read feeds 2 MFMAs, like fwd);
+// 64-row kv tiles double-buffered in LDS with issue-early staging, one
+// barrier per tile.
 // ===========================================================================
 template <int D, bool CAUSAL>
 __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dq_kernel(
@@ -699,14 +701,15 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dq_kernel(
   constexpr int KSZ = TILE * RS;   // K row-major buffer
   constexpr int TSZ = D * TRS;     // K^T buffer
   constexpr int VSZ = TILE * RS;   // V row-major buffer
+  constexpr int RB = 2;
 
-  __shared__ unsigned short smem[2 * (KSZ + TSZ + VSZ) + FW_WAVES * 16 * TS];
+  __shared__ unsigned short smem[2 * (KSZ + TSZ + VSZ) + FW_WAVES * 32 * TS];
   unsigned short* k_lds = smem;
   unsigned short* kt_lds = smem + 2 * KSZ;
   unsigned short* v_lds = smem + 2 * (KSZ + TSZ);
   unsigned short* p_lds = smem + 2 * (KSZ + TSZ + VSZ);
 
-  const int qt = blockIdx.x;   // 128-row q block
+  const int qt = blockIdx.x;   // 256-row q block
   const int b = blockIdx.y / H, hh = blockIdx.y % H;
   const long bh = blockIdx.y;
   const int wid = threadIdx.x / WAVE;
@@ -717,32 +720,38 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dq_kernel(
   const __hip_bfloat16* vp = v.at(b, hh);
   const __hip_bfloat16* dop = dout.at(b, hh);
 
-  const int qrow0 = qt * QTILE + wid * 16;
-  bf8 qfrag[D / 32], dofrag[D / 32];
-  load_a_frags<D>(qp, q.rs, qrow0, S, lane, qfrag);
-  load_a_frags<D>(dop, dout.rs, qrow0, S, lane, dofrag);
-
-  f4 dqacc[NDT];
+  const int qrow0 = qt * (QTILE * RB) + wid * (16 * RB);
+  bf8 qfrag[RB][D / 32], dofrag[RB][D / 32];
 #pragma unroll
-  for (int i = 0; i < NDT; ++i) dqacc[i] = f4{0.f, 0.f, 0.f, 0.f};
+  for (int rb = 0; rb < RB; ++rb) {
+    load_a_frags<D>(qp, q.rs, qrow0 + rb * 16, S, lane, qfrag[rb]);
+    load_a_frags<D>(dop, dout.rs, qrow0 + rb * 16, S, lane, dofrag[rb]);
+  }
+
+  f4 dqacc[RB][NDT];
+#pragma unroll
+  for (int rb = 0; rb < RB; ++rb)
+#pragma unroll
+    for (int i = 0; i < NDT; ++i) dqacc[rb][i] = f4{0.f, 0.f, 0.f, 0.f};
 
   const int ccol = lane & 15;
   const int crow4 = (lane >> 4) * 4;
-  const int my_qrow = qrow0 + crow4;
-  const int wave_last_row = min(qrow0 + 15, S - 1);
-  unsigned short* myp = p_lds + wid * 16 * TS;
+  unsigned short* myp = p_lds + wid * 32 * TS;
 
-  float lse_r[4], dlt_r[4];
+  float lse_r[RB][4], dlt_r[RB][4];
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    int qr = my_qrow + r;
-    lse_r[r] = (qr < S) ? lse[bh * (long)S + qr] : 0.f;
-    dlt_r[r] = (qr < S) ? delta[bh * (long)S + qr] : 0.f;
-  }
+  for (int rb = 0; rb < RB; ++rb)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int qr = qrow0 + rb * 16 + crow4 + r;
+      lse_r[rb][r] = (qr < S) ? lse[bh * (long)S + qr] : 0.f;
+      dlt_r[rb][r] = (qr < S) ? delta[bh * (long)S + qr] : 0.f;
+    }
 
   const int kv_tiles = CAUSAL
-      ? min(kv_total, (qt * QTILE + QTILE - 1) / TILE + 1)
+      ? min(kv_total, (qt * QTILE * RB + QTILE * RB - 1) / TILE + 1)
       : kv_total;
+  const int wave_last_row = min(qrow0 + RB * 16 - 1, S - 1);
 
   // staging maps: D/16 threads per kv row (row-major K/V); per-thread
   // 4x4 register transpose for the K^T image (b64 writes, see fwd)
@@ -816,54 +825,62 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dq_kernel(
 
     const bool active = !CAUSAL || (kv0 <= wave_last_row);
     if (active) {
-      // ---- S = Q K^T; P ----
-      f4 p[4];
+      // ---- per ct: S = Q K^T and dP = dO V^T for BOTH row groups
+      // (each kb/vb read feeds 2 MFMAs), then dS -> LDS ----
 #pragma unroll
       for (int ct = 0; ct < 4; ++ct) {
-        f4 acc = {0.f, 0.f, 0.f, 0.f};
+        f4 sacc[RB], dpacc[RB];
+#pragma unroll
+        for (int rb = 0; rb < RB; ++rb) {
+          sacc[rb] = f4{0.f, 0.f, 0.f, 0.f};
+          dpacc[rb] = f4{0.f, 0.f, 0.f, 0.f};
+        }
 #pragma unroll
         for (int kc = 0; kc < D / 32; ++kc) {
           bf8 kb = read_b_frag(kb_lds, RS, ct * 16, kc * 32, lane);
-          acc = MFMA_BF16(qfrag[kc], kb, acc);
-        }
-        int kcol = kv0 + ct * 16 + ccol;
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          float pv = 0.f;
-          if ((!CAUSAL || kcol <= my_qrow + r) && kcol < S && my_qrow + r < S)
-            pv = __expf(acc[r] * scale - lse_r[r]);
-          p[ct][r] = pv;
+          for (int rb = 0; rb < RB; ++rb)
+            sacc[rb] = MFMA_BF16(qfrag[rb][kc], kb, sacc[rb]);
         }
-      }
-
-      // ---- dP = dO V^T ----
-      f4 dp[4];
-#pragma unroll
-      for (int ct = 0; ct < 4; ++ct) {
-        f4 acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
         for (int kc = 0; kc < D / 32; ++kc) {
           bf8 vb = read_b_frag(vb_lds, RS, ct * 16, kc * 32, lane);
-          acc = MFMA_BF16(dofrag[kc], vb, acc);
+#pragma unroll
+          for (int rb = 0; rb < RB; ++rb)
+            dpacc[rb] = MFMA_BF16(dofrag[rb][kc], vb, dpacc[rb]);
         }
-        dp[ct] = acc;
+        int kcol = kv0 + ct * 16 + ccol;
+#pragma unroll
+        for (int rb = 0; rb < RB; ++rb) {
+          const int row_base = qrow0 + rb * 16 + crow4;
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            float pv = 0.f;
+            if ((!CAUSAL || kcol <= row_base + r) && kcol < S &&
+                row_base + r < S)
+              pv = __expf(sacc[rb][r] * scale - lse_r[rb][r]);
+            float ds = pv * (dpacc[rb][r] - dlt_r[rb][r]) * scale;
+            myp[(rb * 16 + crow4 + r) * TS + ct * 16 + ccol] = f2bf_raw(ds);
+          }
+        }
       }
 
-      // ---- dS -> LDS; dQ += dS K ----
+      // ---- dQ += dS K (each K^T read feeds both row groups) ----
+      bf8 dsa[RB][2];
 #pragma unroll
-      for (int ct = 0; ct < 4; ++ct)
+      for (int rb = 0; rb < RB; ++rb)
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          float ds = p[ct][r] * (dp[ct][r] - dlt_r[r]) * scale;
-          myp[(crow4 + r) * TS + ct * 16 + ccol] = f2bf_raw(ds);
-        }
+        for (int kc = 0; kc < 2; ++kc)
+          dsa[rb][kc] = read_a_frag_lds(myp + rb * 16 * TS, TS, kc * 32,
+                                        lane);
 #pragma unroll
       for (int kc = 0; kc < 2; ++kc) {
-        bf8 dsa = read_a_frag_lds(myp, TS, kc * 32, lane);
 #pragma unroll
         for (int dt = 0; dt < NDT; ++dt) {
           bf8 kb = read_b_frag(ktb_lds, TRS, dt * 16, kc * 32, lane);
-          dqacc[dt] = MFMA_BF16(dsa, kb, dqacc[dt]);
+#pragma unroll
+          for (int rb = 0; rb < RB; ++rb)
+            dqacc[rb][dt] = MFMA_BF16(dsa[rb][kc], kb, dqacc[rb][dt]);
         }
       }
     }
@@ -874,15 +891,17 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dq_kernel(
 
   __hip_bfloat16* dqp = dq.at(b, hh);
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    int qr = my_qrow + r;
-    if (qr >= S) continue;
-    unsigned short* qrow_p = (unsigned short*)dqp + (long)qr * dq.rs;
+  for (int rb = 0; rb < RB; ++rb)
 #pragma unroll
-    for (int dt = 0; dt < NDT; ++dt) {
-      qrow_p[dt * 16 + ccol] = f2bf_raw(dqacc[dt][r]);
+    for (int r = 0; r < 4; ++r) {
+      int qr = qrow0 + rb * 16 + crow4 + r;
+      if (qr >= S) continue;
+      unsigned short* qrow_p = (unsigned short*)dqp + (long)qr * dq.rs;
+#pragma unroll
+      for (int dt = 0; dt < NDT; ++dt) {
+        qrow_p[dt * 16 + ccol] = f2bf_raw(dqacc[rb][dt][r]);
+      }
     }
-  }
 }
 
 Strided strided_of(const torch::Tensor& t, int b_dim, int h_dim, int s_dim) {
@@ -932,16 +951,18 @@ void launch_bwd(Strided q, Strided k, Strided v, Strided dout, Strided o,
                      dim3((rows + (BLOCKT / WAVE) - 1) / (BLOCKT / WAVE)),
                      dim3(BLOCKT), 0, stream, dout, o, delta, H, S, D);
   int tiles64 = (S + TILE - 1) / TILE;       // inner streamed tiles
-  int blocks128 = (S + QTILE - 1) / QTILE;   // per-block home tile
+  int blocks128 = (S + QTILE - 1) / QTILE;   // dkv per-block home tile
+  int blocks256 = (S + QTILE * 2 - 1) / (QTILE * 2);  // dq RB=2 home tiles
   dim3 grid(blocks128, B * H);
+  dim3 grid_dq(blocks256, B * H);
 #define LAUNCH_BWD(DD, CC)                                                   \
   do {                                                                       \
     hipLaunchKernelGGL((attn_bwd_dkv_kernel<DD, CC>), grid,                  \
                        dim3(FW_BLOCKT), 0, stream, q, k, v, dout, lse,       \
                        delta, dk, dv, H, S, scale, tiles64);                 \
-    hipLaunchKernelGGL((attn_bwd_dq_kernel<DD, CC>), grid, dim3(FW_BLOCKT),  \
-                       0, stream, q, k, v, dout, lse, delta, dq, H, S,       \
-                       scale, tiles64);                                      \
+    hipLaunchKernelGGL((attn_bwd_dq_kernel<DD, CC>), grid_dq,                \
+                       dim3(FW_BLOCKT), 0, stream, q, k, v, dout, lse,       \
+                       delta, dq, H, S, scale, tiles64);                     \
   } while (0)
   if (D == 128) { if (causal) LAUNCH_BWD(128, true); else LAUNCH_BWD(128, false); }
   else          { if (causal) LAUNCH_BWD(64, true);  else LAUNCH_BWD(64, false); }
