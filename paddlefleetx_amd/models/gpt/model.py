@@ -54,8 +54,11 @@ class MultiHeadAttention(nn.Module):
                  attn_dropout: float = 0.0, fused_attn: bool = True,
                  dtype: Optional[torch.dtype] = None,
                  sequence_parallel: bool = False,
-                 init_std: float = 0.02, output_layer_init_std: float = 0.02):
+                 init_std: float = 0.02, output_layer_init_std: float = 0.02,
+                 cp_backend: str = "ulysses"):
         super().__init__()
+        assert cp_backend in ("ulysses", "ring")
+        self.cp_backend = cp_backend
         mp = get_hcg().get_model_parallel_world_size()
         assert num_heads % mp == 0, f"heads {num_heads} not divisible by mp {mp}"
         assert hidden_size % num_heads == 0
@@ -93,14 +96,20 @@ class MultiHeadAttention(nn.Module):
             # inside attention trades seq for heads (parallel/cp.py)
             assert self.attn_dropout_p == 0.0, \
                 "attention dropout unsupported under context parallel"
-            assert self.num_heads_local % cp == 0, \
-                f"heads/mp {self.num_heads_local} not divisible by cp {cp}"
-            from paddlefleetx_amd.parallel.cp import UlyssesAttention
+            if self.cp_backend != "ring":
+                # ring keeps all heads local; only the a2a path shards them
+                assert self.num_heads_local % cp == 0, \
+                    f"heads/mp {self.num_heads_local} not divisible by cp {cp}"
             B, Sl, _ = qkv.shape
             qkv = qkv.view(B, Sl, self.num_heads_local, 3 * self.head_dim)
             q, k, v = qkv.split(self.head_dim, dim=-1)  # [B, S/cp, h, D]
-            ua = UlyssesAttention(scale=self.scale, causal=True)
-            o = ua(q, k, v).reshape(B, Sl, -1)
+            if self.cp_backend == "ring":
+                from paddlefleetx_amd.parallel.ring import RingAttention
+                ca = RingAttention(scale=self.scale, causal=True)
+            else:
+                from paddlefleetx_amd.parallel.cp import UlyssesAttention
+                ca = UlyssesAttention(scale=self.scale, causal=True)
+            o = ca(q, k, v).reshape(B, Sl, -1)
             return self.out_proj(o), None
         if (self.fused_attn and self.attn_dropout_p == 0.0 and cache is None
                 and not use_cache and not self.sequence_parallel):
@@ -192,7 +201,8 @@ class TransformerDecoderLayer(nn.Module):
                  fused_attn: bool = True, dtype: Optional[torch.dtype] = None,
                  sequence_parallel: bool = False, init_std: float = 0.02,
                  num_layers_for_scale: int = 1, expert_module: Optional[nn.Module] = None,
-                 recompute_granularity: str = "full", use_recompute: bool = False):
+                 recompute_granularity: str = "full", use_recompute: bool = False,
+                 cp_backend: str = "ulysses"):
         super().__init__()
         out_std = init_std / math.sqrt(2.0 * num_layers_for_scale)
         self.ln1 = FusedLayerNorm(hidden_size, dtype=dtype)
@@ -211,7 +221,8 @@ class TransformerDecoderLayer(nn.Module):
                                        fused_attn=fused_attn, dtype=dtype,
                                        sequence_parallel=sequence_parallel,
                                        init_std=init_std,
-                                       output_layer_init_std=out_std)
+                                       output_layer_init_std=out_std,
+                                       cp_backend=cp_backend)
         if expert_module is not None:
             self.ffn = expert_module
         else:
@@ -291,6 +302,7 @@ class GPTModel(nn.Module):
                  sequence_parallel: bool = False,
                  initializer_range: float = 0.02,
                  moe_configs: Optional[Dict[str, Any]] = None,
+                 cp_backend: str = "ulysses",
                  dtype: Optional[torch.dtype] = None, **unused: Any):
         super().__init__()
         ffn_hidden_size = ffn_hidden_size or 4 * hidden_size
@@ -326,7 +338,8 @@ class GPTModel(nn.Module):
                                     num_layers_for_scale=num_layers,
                                     use_recompute=use_recompute,
                                     recompute_granularity=recompute_granularity,
-                                    expert_module=_make_expert_module())
+                                    expert_module=_make_expert_module(),
+                                    cp_backend=cp_backend)
             for _ in range(num_layers)])
         self.final_ln = FusedLayerNorm(hidden_size, dtype=dtype)
         if sequence_parallel:

@@ -1,0 +1,157 @@
+"""Ring attention over the context-parallel group (long-context path).
+
+Alternative CP backend to Ulysses (parallel/cp.py): instead of trading
+sequence for heads with an all-to-all, each rank keeps its sequence
+shard of Q resident and the K/V shards travel around the ring
+(cp p2p hops on xGMI), merged with a numerically stable
+log-sum-exp combine. Per-step traffic is 2·B·Sl·H·D bytes to ONE
+neighbour — on the 8×MI355X mesh each hop has a dedicated xGMI link, and
+compute of block j overlaps the in-flight transfer of block j+1.
+
+The reference has no ring attention (SURVEY §5 long-context: "No ring
+attention ... max context bounded by max_position_embeddings") — this is
+an MI355X-native extension: 288 GB HBM + ring K/V makes 32k+ context
+feasible at 6.7B.
+
+Causality with sequential sharding: rank r's Q rows are globally after
+every row held by ranks < r, so a visiting block from rank s needs
+  s < r : full (non-causal) attention
+  s == r: causal attention (the local diagonal block)
+  s > r : skipped entirely (still rotated to keep the ring in step).
+
+Block math reuses the gfx950 flash kernels unchanged: forward merges
+per-block (o_j, lse_j); backward recomputes each block's P from the
+GLOBAL lse (so per-block dq/dk/dv sum to the exact full-attention
+grads), with the (dk, dv) accumulators travelling with their K/V block —
+after cp rotations they arrive back at the owning rank.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+from paddlefleetx_amd.parallel.env import get_hcg
+
+__all__ = ["RingAttention", "ring_attention"]
+
+
+def _block_fwd(q, k, v, causal, scale):
+    from paddlefleetx_amd.ops import _reference as ref
+    from paddlefleetx_amd.ops import hip_ext, use_hip
+    if use_hip(q):
+        return hip_ext().attn_fwd(q, k, v, causal, scale)
+    return ref.attention_fwd(q, k, v, causal, scale)
+
+
+def _block_bwd(do, q, k, v, o, lse, causal, scale):
+    from paddlefleetx_amd.ops import _reference as ref
+    from paddlefleetx_amd.ops import hip_ext, use_hip
+    if use_hip(q):
+        return hip_ext().attn_bwd(do, q, k, v, o, lse, causal, scale)
+    return ref.attention_bwd(do, q, k, v, o, lse, causal, scale)
+
+
+def _rotate(tensors, g):
+    """Send each tensor to the next ring rank, receive from the previous.
+    Returns the received tensors (blocking, one fused p2p batch)."""
+    ranks = g.ranks
+    me = g.rank
+    nxt = ranks[(me + 1) % g.world_size]
+    prv = ranks[(me - 1) % g.world_size]
+    outs = [torch.empty_like(t) for t in tensors]
+    ops = []
+    for t, o in zip(tensors, outs):
+        ops.append(dist.P2POp(dist.isend, t.contiguous(), nxt))
+        ops.append(dist.P2POp(dist.irecv, o, prv))
+    for r in dist.batch_isend_irecv(ops):
+        r.wait()
+    return outs
+
+
+class _RingAttnFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, scale):
+        g = get_hcg().get_context_parallel_group()
+        cp = g.world_size
+        r = g.rank if cp > 1 else 0
+
+        o_acc = None
+        lse_acc = None
+        cur_k, cur_v = k, v
+        for j in range(cp):
+            src = (r - j) % cp
+            if j < cp - 1:
+                nk, nv = _rotate([cur_k, cur_v], g)
+            if src <= r:
+                o_j, lse_j = _block_fwd(q, cur_k, cur_v, src == r, scale)
+                if o_acc is None:
+                    o_acc, lse_acc = o_j.float(), lse_j
+                else:
+                    lse_new = torch.logaddexp(lse_acc, lse_j)
+                    w_old = torch.exp(lse_acc - lse_new)[..., None]
+                    w_new = torch.exp(lse_j - lse_new)[..., None]
+                    o_acc = o_acc * w_old + o_j.float() * w_new
+                    lse_acc = lse_new
+            if j < cp - 1:
+                cur_k, cur_v = nk, nv
+        out = o_acc.to(q.dtype)
+        ctx.save_for_backward(q, k, v, out, lse_acc)
+        ctx.scale = scale
+        return out
+
+    @staticmethod
+    def backward(ctx, do):
+        q, k, v, o, lse = ctx.saved_tensors
+        g = get_hcg().get_context_parallel_group()
+        cp = g.world_size
+        r = g.rank if cp > 1 else 0
+        do = do.contiguous()
+
+        dq_acc = torch.zeros_like(q, dtype=torch.float32)
+        cur_k, cur_v = k, v
+        cur_dk = torch.zeros_like(k, dtype=torch.float32)
+        cur_dv = torch.zeros_like(v, dtype=torch.float32)
+        for j in range(cp):
+            src = (r - j) % cp
+            if src <= r:
+                dq_j, dk_j, dv_j = _block_bwd(do, q, cur_k, cur_v, o, lse,
+                                              src == r, ctx.scale)
+                dq_acc += dq_j.float()
+                cur_dk += dk_j.float()
+                cur_dv += dv_j.float()
+            if cp > 1:
+                # rotate every step: after cp hops each (k, v, dk, dv)
+                # quartet is back at its owning rank
+                cur_k, cur_v, cur_dk, cur_dv = _rotate(
+                    [cur_k, cur_v, cur_dk, cur_dv], g)
+        return (dq_acc.to(q.dtype), cur_dk.to(k.dtype), cur_dv.to(v.dtype),
+                None)
+
+
+def ring_attention(q, k, v, scale: Optional[float] = None):
+    """Causal ring attention; q,k,v [B, H, S/cp, D] sequence shards
+    (sequential sharding, same slicing as `pretreating_batch`)."""
+    scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
+    return _RingAttnFn.apply(q.contiguous(), k.contiguous(), v.contiguous(),
+                             scale)
+
+
+class RingAttention(torch.nn.Module):
+    """Drop-in alternative to UlyssesAttention: q,k,v [B, S/cp, h, D] ->
+    o [B, S/cp, h, D] (same interface/layout as parallel/cp.py)."""
+
+    def __init__(self, scale: Optional[float] = None, causal: bool = True):
+        super().__init__()
+        assert causal, "ring attention: causal only (GPT pretraining path)"
+        self.scale = scale
+
+    def forward(self, q, k, v):
+        q = q.permute(0, 2, 1, 3)  # [B, h, Sl, D]
+        k = k.permute(0, 2, 1, 3)
+        v = v.permute(0, 2, 1, 3)
+        o = ring_attention(q, k, v, self.scale)
+        return o.permute(0, 2, 1, 3)
