@@ -456,10 +456,12 @@ __global__ void attn_bwd_delta_kernel(Strided dout, Strided o,
 }
 
 // ===========================================================================
-// Backward dK/dV v2 (transposed score space S^T = K Q^T).
+// Backward dK/dV v3 (transposed score space S^T = K Q^T).
 // 8 waves, 128 kv rows per block (16/wave); 64-row q tiles streamed with
-// issue-early register staging (guide T14): tile t+1's global loads are in
-// flight during tile t's MFMAs; single LDS buffer set, two barriers/tile.
+// issue-early register staging (guide T14): tile t+1's global loads are
+// in flight during tile t's MFMAs; the four q/dO LDS images are
+// DOUBLE-BUFFERED (158.7 KB of the CU's 160 KB LDS) so each tile costs
+// one barrier instead of two.
 // ===========================================================================
 template <int D, bool CAUSAL>
 __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dkv_kernel(
@@ -470,13 +472,15 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dkv_kernel(
   constexpr int TS = TILE + PAD;       // P image stride (b128-aligned)
   constexpr int TRS = TILE + 4;        // transposed q/dO stride (see VRS)
   constexpr int NDT = D / 16;
+  constexpr int QSZ = TILE * RS;       // q / dO row-major buffer
+  constexpr int TSZ = D * TRS;         // q^T / dO^T buffer
 
-  __shared__ unsigned short q_lds[TILE * RS];
-  __shared__ unsigned short qt_lds[D * TRS];
-  __shared__ unsigned short do_lds[TILE * RS];
-  __shared__ unsigned short dot_lds[D * TRS];
+  __shared__ unsigned short q_lds[2 * QSZ];
+  __shared__ unsigned short qt_lds[2 * TSZ];
+  __shared__ unsigned short do_lds[2 * QSZ];
+  __shared__ unsigned short dot_lds[2 * TSZ];
   __shared__ unsigned short p_lds[FW_WAVES * 16 * TS];
-  __shared__ float lsed_lds[2 * TILE];
+  __shared__ float lsed_lds[2][2 * TILE];
 
   const int kt = blockIdx.x;            // 128-row kv block
   const int b = blockIdx.y / H, hh = blockIdx.y % H;
@@ -555,34 +559,44 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dkv_kernel(
       dlt_s = (qi < S) ? delta[bh * (long)S + qi] : 0.f;
     }
   };
-  auto write_tiles = [&]() {
+  auto write_tiles = [&](int buf) {
+    unsigned short* qd = q_lds + buf * QSZ;
+    unsigned short* dd = do_lds + buf * QSZ;
+    unsigned short* qtd = qt_lds + buf * TSZ;
+    unsigned short* dtd = dot_lds + buf * TSZ;
     if (st_on) {
-      *reinterpret_cast<short8v*>(q_lds + st_row * RS + st_col) = qreg[0];
-      *reinterpret_cast<short8v*>(q_lds + st_row * RS + st_col + 8) = qreg[1];
-      *reinterpret_cast<short8v*>(do_lds + st_row * RS + st_col) = dreg[0];
-      *reinterpret_cast<short8v*>(do_lds + st_row * RS + st_col + 8) = dreg[1];
+      *reinterpret_cast<short8v*>(qd + st_row * RS + st_col) = qreg[0];
+      *reinterpret_cast<short8v*>(qd + st_row * RS + st_col + 8) = qreg[1];
+      *reinterpret_cast<short8v*>(dd + st_row * RS + st_col) = dreg[0];
+      *reinterpret_cast<short8v*>(dd + st_row * RS + st_col + 8) = dreg[1];
     }
     if (vb_on) {
 #pragma unroll
       for (int c = 0; c < 4; ++c) {
         short4v tq = {qblk[0][c], qblk[1][c], qblk[2][c], qblk[3][c]};
         short4v td = {dblk[0][c], dblk[1][c], dblk[2][c], dblk[3][c]};
-        *reinterpret_cast<short4v*>(qt_lds + (vb_col + c) * TRS + vb_row) = tq;
-        *reinterpret_cast<short4v*>(dot_lds + (vb_col + c) * TRS + vb_row) = td;
+        *reinterpret_cast<short4v*>(qtd + (vb_col + c) * TRS + vb_row) = tq;
+        *reinterpret_cast<short4v*>(dtd + (vb_col + c) * TRS + vb_row) = td;
       }
     }
     if (threadIdx.x < TILE) {
-      lsed_lds[threadIdx.x] = lse_s;
-      lsed_lds[TILE + threadIdx.x] = dlt_s;
+      lsed_lds[buf][threadIdx.x] = lse_s;
+      lsed_lds[buf][TILE + threadIdx.x] = dlt_s;
     }
   };
 
   issue_loads(qt_first * TILE);
-  write_tiles();
+  write_tiles(0);
   __syncthreads();
 
   for (int qt = qt_first; qt < q_tiles; ++qt) {
     const int q0 = qt * TILE;
+    const int cur = (qt - qt_first) & 1;
+    const unsigned short* qb_lds = q_lds + cur * QSZ;
+    const unsigned short* db_lds = do_lds + cur * QSZ;
+    const unsigned short* qtb_lds = qt_lds + cur * TSZ;
+    const unsigned short* dtb_lds = dot_lds + cur * TSZ;
+    const float* lsed_b = lsed_lds[cur];
     const bool have_next = (qt + 1) < q_tiles;
     if (have_next) issue_loads(q0 + TILE);
 
@@ -596,11 +610,11 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dkv_kernel(
         f4 acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
         for (int kc = 0; kc < D / 32; ++kc) {
-          bf8 qb = read_b_frag(q_lds, RS, ct * 16, kc * 32, lane);
+          bf8 qb = read_b_frag(qb_lds, RS, ct * 16, kc * 32, lane);
           acc = MFMA_BF16(kfrag[kc], qb, acc);
         }
         int qcol = q0 + ct * 16 + ccol;
-        float l = lsed_lds[ct * 16 + ccol];
+        float l = lsed_b[ct * 16 + ccol];
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           int kvr = my_kvrow + r;
@@ -622,7 +636,7 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dkv_kernel(
         bf8 pa = read_a_frag_lds(myp, TS, kc * 32, lane);
 #pragma unroll
         for (int dt = 0; dt < NDT; ++dt) {
-          bf8 db = read_b_frag(dot_lds, TRS, dt * 16, kc * 32, lane);
+          bf8 db = read_b_frag(dtb_lds, TRS, dt * 16, kc * 32, lane);
           dvacc[dt] = MFMA_BF16(pa, db, dvacc[dt]);
         }
       }
@@ -633,7 +647,7 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dkv_kernel(
         f4 acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
         for (int kc = 0; kc < D / 32; ++kc) {
-          bf8 db = read_b_frag(do_lds, RS, ct * 16, kc * 32, lane);
+          bf8 db = read_b_frag(db_lds, RS, ct * 16, kc * 32, lane);
           acc = MFMA_BF16(vfrag[kc], db, acc);
         }
         dpt[ct] = acc;
@@ -642,7 +656,7 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dkv_kernel(
       // ---- dS^T -> LDS; dK += dS^T Q ----
 #pragma unroll
       for (int ct = 0; ct < 4; ++ct) {
-        float dlt = lsed_lds[TILE + ct * 16 + ccol];
+        float dlt = lsed_b[TILE + ct * 16 + ccol];
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           float ds = pt[ct][r] * (dpt[ct][r] - dlt) * scale;
@@ -654,17 +668,15 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dkv_kernel(
         bf8 dsa = read_a_frag_lds(myp, TS, kc * 32, lane);
 #pragma unroll
         for (int dt = 0; dt < NDT; ++dt) {
-          bf8 qb = read_b_frag(qt_lds, TRS, dt * 16, kc * 32, lane);
+          bf8 qb = read_b_frag(qtb_lds, TRS, dt * 16, kc * 32, lane);
           dkacc[dt] = MFMA_BF16(dsa, qb, dkacc[dt]);
         }
       }
     }
 
-    if (have_next) {
-      __syncthreads();   // all waves done reading this tile's LDS
-      write_tiles();
-      __syncthreads();   // next tile visible
-    }
+    // write the next tile into the spare buffer; one barrier per tile
+    if (have_next) write_tiles(cur ^ 1);
+    __syncthreads();
   }
 
   __hip_bfloat16* dkp = dk.at(b, hh);
