@@ -177,9 +177,12 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
   unsigned short* vt_lds = smem + 2 * KSZ;       // [2][VSZ]
   unsigned short* p_lds = smem + 2 * KSZ + 2 * VSZ;
 
-  const int qt = blockIdx.x;
-  const int b = blockIdx.y / H, hh = blockIdx.y % H;
-  const long bh = blockIdx.y;
+  // bh on blockIdx.x: the linear dispatch then gives every CU one block
+  // of each qt, balancing the causal tile-count imbalance (B*H >= 256
+  // keeps all CUs fed within one qt wave)
+  const int qt = blockIdx.y;
+  const int b = blockIdx.x / H, hh = blockIdx.x % H;
+  const long bh = blockIdx.x;
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
 
@@ -333,23 +336,43 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
 #pragma unroll
         for (int r = 0; r < 4; ++r) pmax[r] = group16_reduce_max(pmax[r]);
 
-        float alpha[4];
+        // rescale only when some row's max actually grew (alpha==1 is
+        // exact otherwise) — wave-uniform skip saves the O(NDT*4) VALU
+        // rescale on most tiles
+        bool grew = false;
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          float nm = fmaxf(m_r[rb][r], pmax[r]);
-          if (nm == -INFINITY) nm = 0.f;
-          alpha[r] = (m_r[rb][r] == -INFINITY) ? 0.f : __expf(m_r[rb][r] - nm);
-          m_r[rb][r] = (m_r[rb][r] == -INFINITY && pmax[r] == -INFINITY)
-              ? -INFINITY : nm;
+        for (int r = 0; r < 4; ++r) grew |= pmax[r] > m_r[rb][r];
+        if (__builtin_amdgcn_ballot_w64(grew)) {
+          float alpha[4];
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            float nm = fmaxf(m_r[rb][r], pmax[r]);
+            alpha[r] = (m_r[rb][r] == -INFINITY) ? 0.f
+                : __expf(m_r[rb][r] - nm);
+            m_r[rb][r] = (m_r[rb][r] == -INFINITY && pmax[r] == -INFINITY)
+                ? -INFINITY : nm;
+            l_r[rb][r] *= alpha[r];
+          }
+#pragma unroll
+          for (int i = 0; i < NDT; ++i) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) oacc[rb][i][r] *= alpha[r];
+          }
         }
+
+        // exp base: -inf rows exponentiate against 0 so exp(-inf-0)=0
+        // without a per-element select
+        float me[4];
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          me[r] = (m_r[rb][r] == -INFINITY) ? 0.f : m_r[rb][r];
 
         float psum[4] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
         for (int ct = 0; ct < 4; ++ct) {
 #pragma unroll
           for (int r = 0; r < 4; ++r) {
-            float p = (s[rb][ct][r] == -INFINITY) ? 0.f
-                : __expf(s[rb][ct][r] - m_r[rb][r]);
+            float p = __expf(s[rb][ct][r] - me[r]);
             s[rb][ct][r] = p;
             psum[r] += p;
           }
@@ -357,21 +380,20 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           psum[r] = group16_reduce_sum(psum[r]);
-          l_r[rb][r] = l_r[rb][r] * alpha[r] + psum[r];
-        }
-#pragma unroll
-        for (int i = 0; i < NDT; ++i) {
-#pragma unroll
-          for (int r = 0; r < 4; ++r) oacc[rb][i][r] *= alpha[r];
+          l_r[rb][r] += psum[r];
         }
 
-        // ---- P -> per-wave LDS for A-fragments ----
+        // ---- P -> per-wave LDS for A-fragments (cvt_pk packs row
+        // pairs: 1 VALU conversion per 2 values) ----
 #pragma unroll
         for (int ct = 0; ct < 4; ++ct) {
 #pragma unroll
-          for (int r = 0; r < 4; ++r) {
-            myp[(rb * 16 + crow4 + r) * PRS + ct * 16 + ccol] =
-                f2bf_raw(s[rb][ct][r]);
+          for (int r = 0; r < 4; r += 2) {
+            unsigned int u = cvt_pk_bf16(s[rb][ct][r], s[rb][ct][r + 1]);
+            unsigned short* base =
+                myp + (rb * 16 + crow4 + r) * PRS + ct * 16 + ccol;
+            base[0] = (unsigned short)u;
+            base[PRS] = (unsigned short)(u >> 16);
           }
         }
       }
@@ -420,8 +442,11 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
       float inv = (l_r[rb][r] > 0.f) ? 1.f / l_r[rb][r] : 0.f;
       unsigned short* orow = (unsigned short*)op + (long)qrow * o.rs;
 #pragma unroll
-      for (int dt = 0; dt < NDT; ++dt) {
-        orow[dt * 16 + ccol] = f2bf_raw(oacc[rb][dt][r] * inv);
+      for (int dt = 0; dt < NDT; dt += 2) {
+        unsigned int u = cvt_pk_bf16(oacc[rb][dt][r] * inv,
+                                     oacc[rb][dt + 1][r] * inv);
+        orow[dt * 16 + ccol] = (unsigned short)u;
+        orow[(dt + 1) * 16 + ccol] = (unsigned short)(u >> 16);
       }
       if (ccol == 0 && lse_out) {
         float lv = (l_r[rb][r] > 0.f) ? m_r[rb][r] + logf(l_r[rb][r])
@@ -482,9 +507,9 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dkv_kernel(
   __shared__ unsigned short p_lds[FW_WAVES * 16 * TS];
   __shared__ float lsed_lds[2][2 * TILE];
 
-  const int kt = blockIdx.x;            // 128-row kv block
-  const int b = blockIdx.y / H, hh = blockIdx.y % H;
-  const long bh = blockIdx.y;
+  const int kt = blockIdx.y;            // 128-row kv block
+  const int b = blockIdx.x / H, hh = blockIdx.x % H;
+  const long bh = blockIdx.x;
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
 
@@ -629,8 +654,12 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dkv_kernel(
 #pragma unroll
       for (int ct = 0; ct < 4; ++ct)
 #pragma unroll
-        for (int r = 0; r < 4; ++r)
-          myp[(crow4 + r) * TS + ct * 16 + ccol] = f2bf_raw(pt[ct][r]);
+        for (int r = 0; r < 4; r += 2) {
+          unsigned int u = cvt_pk_bf16(pt[ct][r], pt[ct][r + 1]);
+          unsigned short* base = myp + (crow4 + r) * TS + ct * 16 + ccol;
+          base[0] = (unsigned short)u;
+          base[TS] = (unsigned short)(u >> 16);
+        }
 #pragma unroll
       for (int kc = 0; kc < 2; ++kc) {
         bf8 pa = read_a_frag_lds(myp, TS, kc * 32, lane);
@@ -657,10 +686,16 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dkv_kernel(
 #pragma unroll
       for (int ct = 0; ct < 4; ++ct) {
         float dlt = lsed_b[TILE + ct * 16 + ccol];
+        float ds[4];
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          float ds = pt[ct][r] * (dpt[ct][r] - dlt) * scale;
-          myp[(crow4 + r) * TS + ct * 16 + ccol] = f2bf_raw(ds);
+        for (int r = 0; r < 4; ++r)
+          ds[r] = pt[ct][r] * (dpt[ct][r] - dlt) * scale;
+#pragma unroll
+        for (int r = 0; r < 4; r += 2) {
+          unsigned int u = cvt_pk_bf16(ds[r], ds[r + 1]);
+          unsigned short* base = myp + (crow4 + r) * TS + ct * 16 + ccol;
+          base[0] = (unsigned short)u;
+          base[TS] = (unsigned short)(u >> 16);
         }
       }
 #pragma unroll
@@ -688,9 +723,13 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dkv_kernel(
     unsigned short* krow = (unsigned short*)dkp + (long)kvr * dk.rs;
     unsigned short* vrow = (unsigned short*)dvp + (long)kvr * dv.rs;
 #pragma unroll
-    for (int dt = 0; dt < NDT; ++dt) {
-      krow[dt * 16 + ccol] = f2bf_raw(dkacc[dt][r]);
-      vrow[dt * 16 + ccol] = f2bf_raw(dvacc[dt][r]);
+    for (int dt = 0; dt < NDT; dt += 2) {
+      unsigned int uk = cvt_pk_bf16(dkacc[dt][r], dkacc[dt + 1][r]);
+      unsigned int uv = cvt_pk_bf16(dvacc[dt][r], dvacc[dt + 1][r]);
+      krow[dt * 16 + ccol] = (unsigned short)uk;
+      krow[(dt + 1) * 16 + ccol] = (unsigned short)(uk >> 16);
+      vrow[dt * 16 + ccol] = (unsigned short)uv;
+      vrow[(dt + 1) * 16 + ccol] = (unsigned short)(uv >> 16);
     }
   }
 }
@@ -721,9 +760,9 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dq_kernel(
   unsigned short* v_lds = smem + 2 * (KSZ + TSZ);
   unsigned short* p_lds = smem + 2 * (KSZ + TSZ + VSZ);
 
-  const int qt = blockIdx.x;   // 256-row q block
-  const int b = blockIdx.y / H, hh = blockIdx.y % H;
-  const long bh = blockIdx.y;
+  const int qt = blockIdx.y;   // 256-row q block
+  const int b = blockIdx.x / H, hh = blockIdx.x % H;
+  const long bh = blockIdx.x;
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
 
@@ -865,14 +904,22 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dq_kernel(
 #pragma unroll
         for (int rb = 0; rb < RB; ++rb) {
           const int row_base = qrow0 + rb * 16 + crow4;
+          float ds[4];
 #pragma unroll
           for (int r = 0; r < 4; ++r) {
             float pv = 0.f;
             if ((!CAUSAL || kcol <= row_base + r) && kcol < S &&
                 row_base + r < S)
               pv = __expf(sacc[rb][r] * scale - lse_r[rb][r]);
-            float ds = pv * (dpacc[rb][r] - dlt_r[rb][r]) * scale;
-            myp[(rb * 16 + crow4 + r) * TS + ct * 16 + ccol] = f2bf_raw(ds);
+            ds[r] = pv * (dpacc[rb][r] - dlt_r[rb][r]) * scale;
+          }
+#pragma unroll
+          for (int r = 0; r < 4; r += 2) {
+            unsigned int u = cvt_pk_bf16(ds[r], ds[r + 1]);
+            unsigned short* base =
+                myp + (rb * 16 + crow4 + r) * TS + ct * 16 + ccol;
+            base[0] = (unsigned short)u;
+            base[TS] = (unsigned short)(u >> 16);
           }
         }
       }
@@ -910,8 +957,11 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dq_kernel(
       if (qr >= S) continue;
       unsigned short* qrow_p = (unsigned short*)dqp + (long)qr * dq.rs;
 #pragma unroll
-      for (int dt = 0; dt < NDT; ++dt) {
-        qrow_p[dt * 16 + ccol] = f2bf_raw(dqacc[rb][dt][r]);
+      for (int dt = 0; dt < NDT; dt += 2) {
+        unsigned int u = cvt_pk_bf16(dqacc[rb][dt][r],
+                                     dqacc[rb][dt + 1][r]);
+        qrow_p[dt * 16 + ccol] = (unsigned short)u;
+        qrow_p[(dt + 1) * 16 + ccol] = (unsigned short)(u >> 16);
       }
     }
 }
@@ -943,7 +993,7 @@ void launch_fwd(Strided q, Strided k, Strided v, StridedMut o, float* lse,
                 int B, int H, int S, int D, float scale, bool causal) {
   int q_blocks = (S + QTILE * 2 - 1) / (QTILE * 2);  // RB=2 home tiles
   int kv_total = (S + TILE - 1) / TILE;
-  dim3 grid(q_blocks, B * H);
+  dim3 grid(B * H, q_blocks);
   auto stream = at::hip::getCurrentHIPStream();
 #define LAUNCH_FWD(DD, CC)                                                  \
   hipLaunchKernelGGL((attn_fwd_kernel<DD, CC>), grid, dim3(FW_BLOCKT), 0,   \
@@ -965,8 +1015,8 @@ void launch_bwd(Strided q, Strided k, Strided v, Strided dout, Strided o,
   int tiles64 = (S + TILE - 1) / TILE;       // inner streamed tiles
   int blocks128 = (S + QTILE - 1) / QTILE;   // dkv per-block home tile
   int blocks256 = (S + QTILE * 2 - 1) / (QTILE * 2);  // dq RB=2 home tiles
-  dim3 grid(blocks128, B * H);
-  dim3 grid_dq(blocks256, B * H);
+  dim3 grid(B * H, blocks128);
+  dim3 grid_dq(B * H, blocks256);
 #define LAUNCH_BWD(DD, CC)                                                   \
   do {                                                                       \
     hipLaunchKernelGGL((attn_bwd_dkv_kernel<DD, CC>), grid,                  \

@@ -37,6 +37,21 @@ DEV_INLINE unsigned short f2bf_raw(float f) {
   return (unsigned short)(v.i >> 16);
 }
 
+#ifdef __HIP_DEVICE_COMPILE__
+// Pack two f32 into two bf16 (RTNE) in ONE VALU op — the hand-rolled
+// f2bf_raw costs ~4 integer ops per value (guide T12: no builtin on
+// gfx950, use asm). lo -> bits [15:0], hi -> bits [31:16].
+DEV_INLINE unsigned int cvt_pk_bf16(float lo, float hi) {
+  unsigned int r;
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(lo), "v"(hi));
+  return r;
+}
+#else
+DEV_INLINE unsigned int cvt_pk_bf16(float lo, float hi) {
+  return (unsigned int)f2bf_raw(lo) | ((unsigned int)f2bf_raw(hi) << 16);
+}
+#endif
+
 // ---- wave reductions (64-lane) ----
 DEV_INLINE float wave_reduce_sum(float x) {
 #pragma unroll

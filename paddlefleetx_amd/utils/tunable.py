@@ -45,4 +45,6 @@ def enable_tuned_gemms(csv_path: str | None = None) -> bool:
     tunable.tuning_enable(False)  # lookup only; never tune in production
     tunable.read_file(path)
     _loaded = True
+    from paddlefleetx_amd.utils.log import logger
+    logger.info(f"TunableOp: loaded tuned GEMM table {path}")
     return True
