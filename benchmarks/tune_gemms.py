@@ -27,7 +27,7 @@ SEQ = 1024
 
 def shapes():
     out = set()
-    for tp, micro in ((1, 8), (2, 4), (2, 8)):
+    for tp, micro in ((1, 8), (1, 16), (2, 4), (2, 8)):
         m = micro * SEQ
         for i, o in (
                 (H, 3 * H // tp),      # col QKV
