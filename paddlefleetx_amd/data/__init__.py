@@ -36,6 +36,9 @@ def _register_lazy():
     _DATASETS.setdefault("Lambada_Eval_Dataset", Lambada_Eval_Dataset)
     _DATASETS.setdefault("GLUEDataset", GLUEDataset)
     _DATASETS.setdefault("SyntheticGLUEDataset", SyntheticGLUEDataset)
+    from paddlefleetx_amd.data.multimodal_dataset import \
+        SyntheticImagenDataset
+    _DATASETS.setdefault("SyntheticImagenDataset", SyntheticImagenDataset)
 
 
 _register_lazy()

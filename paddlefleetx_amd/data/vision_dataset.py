@@ -23,10 +23,13 @@ class SyntheticImageNetDataset(Dataset):
     collate_fn = None  # default collate
 
     def __init__(self, num_samples: int = 10000, image_size: int = 224,
-                 num_classes: int = 1000, mode: str = "Train", **unused):
+                 num_classes: int = 1000, mode: str = "Train",
+                 moco_two_crops: bool = False, **unused):
         self.num_samples = int(num_samples)
         self.image_size = int(image_size)
         self.num_classes = int(num_classes)
+        # MoCo contract: sample = ((crop_q, crop_k), label)
+        self.moco_two_crops = bool(moco_two_crops)
 
     def __len__(self):
         return self.num_samples
@@ -35,6 +38,9 @@ class SyntheticImageNetDataset(Dataset):
         g = torch.Generator().manual_seed(int(idx))
         img = torch.randn(3, self.image_size, self.image_size, generator=g)
         label = int(torch.randint(0, self.num_classes, (1,), generator=g))
+        if self.moco_two_crops:
+            crop_k = img + 0.1 * torch.randn(img.shape, generator=g)
+            return (img, crop_k), label
         return img, label
 
 

@@ -11,7 +11,17 @@ from paddlefleetx_amd.optims.optimizer import AdamW, FusedAdamW
 __all__ = ["build_optimizer", "build_lr_scheduler", "FusedAdamW", "AdamW",
            "CosineAnnealingWithWarmupDecay", "LinearDecayWithWarmup", "ConstantLR"]
 
-_OPTIMIZERS = {"FusedAdamW": FusedAdamW, "AdamW": AdamW}
+def _momentum(named_params, lr=0.1, weight_decay=0.0, momentum=0.9,
+              **unused):
+    """SGD+momentum for the vision configs (reference optims Momentum)."""
+    import torch
+    params = [p for _, p in named_params]
+    return torch.optim.SGD(params, lr=lr, momentum=momentum,
+                           weight_decay=weight_decay)
+
+
+_OPTIMIZERS = {"FusedAdamW": FusedAdamW, "AdamW": AdamW,
+               "Momentum": _momentum}
 
 
 def build_optimizer(cfg, model, lr_value: float = None, **extra):

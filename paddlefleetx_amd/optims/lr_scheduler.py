@@ -99,6 +99,22 @@ class ViTLRScheduler(_Scheduler):
         return self.lr * (1.0 - ratio)
 
 
+class CosineAnnealingDecay(_Scheduler):
+    """Plain cosine decay (reference MoCo configs: CosineAnnealingDecay)."""
+
+    def __init__(self, learning_rate: float, T_max: int = 100,
+                 eta_min: float = 0.0, **unused):
+        super().__init__()
+        self.lr = learning_rate
+        self.T_max = max(1, int(T_max))
+        self.eta_min = eta_min
+
+    def get_lr(self) -> float:
+        ratio = min(1.0, self.num / self.T_max)
+        return self.eta_min + 0.5 * (self.lr - self.eta_min) * (
+            1.0 + math.cos(math.pi * ratio))
+
+
 class MultiStepDecay(_Scheduler):
     def __init__(self, learning_rate: float, milestones=(30, 60, 90),
                  gamma: float = 0.1, **unused):
@@ -121,6 +137,7 @@ def build_lr_scheduler(cfg) -> _Scheduler:
         "ConstantLR": ConstantLR,
         "ViTLRScheduler": ViTLRScheduler,
         "MultiStepDecay": MultiStepDecay,
+        "CosineAnnealingDecay": CosineAnnealingDecay,
     }
     if name not in table:
         raise ValueError(f"unknown lr scheduler {name}")

@@ -1,0 +1,4 @@
+#!/usr/bin/env bash
+# reference projects/gpt/finetune_gpt_345M_single_card.sh
+cd "$(dirname "$0")/../.."
+python tools/train.py -c paddlefleetx_amd/configs/nlp/gpt/finetune_gpt_345M_single_card_glue.yaml "$@"
