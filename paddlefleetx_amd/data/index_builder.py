@@ -78,3 +78,65 @@ def build_blending_indices(weights: np.ndarray, num_samples: int):
         dataset_sample_index[i] = current[d]
         current[d] += 1
     return dataset_index, dataset_sample_index
+
+
+def _xs64(x: int) -> int:
+    x ^= (x << 13) & 0xFFFFFFFFFFFFFFFF
+    x ^= x >> 7
+    x ^= (x << 17) & 0xFFFFFFFFFFFFFFFF
+    return x & 0xFFFFFFFFFFFFFFFF
+
+
+def build_pair_mapping(docs: np.ndarray, sizes: np.ndarray,
+                       num_epochs: int, max_num_samples: int,
+                       max_seq_length: int, short_seq_prob: float,
+                       seed: int, min_num_sent: int = 2) -> np.ndarray:
+    """BERT/ERNIE sentence-pair sample mapping
+    (fast_index_map_helpers.cpp:195-430 semantics): greedy sentence
+    packing per doc up to a per-sample target length; short targets with
+    prob `short_seq_prob`. Bit-identical to the C++ path (same xorshift64
+    stream)."""
+    if _cpp is not None and hasattr(_cpp, "build_pair_mapping"):
+        return _cpp.build_pair_mapping(
+            np.ascontiguousarray(docs, dtype=np.int64),
+            np.ascontiguousarray(sizes, dtype=np.int32), int(num_epochs),
+            int(max_num_samples), int(max_seq_length), float(short_seq_prob),
+            int(seed), int(min_num_sent))
+    rng = seed if seed else 1
+    short_cut = int(short_seq_prob * 18446744073709551615.0)
+    out = []
+    n_docs = len(docs) - 1
+
+    def draw_target(r):
+        target = max_seq_length
+        if short_seq_prob > 0:
+            r = _xs64(r)
+            if r < short_cut:
+                r2 = _xs64(r)
+                target = 4 + (r2 % (max_seq_length - 3))
+                return target, r2
+        return target, r
+
+    for _ in range(num_epochs):
+        for doc in range(n_docs):
+            if len(out) >= max_num_samples:
+                break
+            s0, s1 = int(docs[doc]), int(docs[doc + 1])
+            if s1 - s0 < min_num_sent:
+                continue
+            start, tok, nsent = s0, 0, 0
+            target, rng = draw_target(rng)
+            for s in range(s0, s1):
+                tok += int(sizes[s])
+                nsent += 1
+                last = s + 1 == s1
+                if (tok >= target and nsent >= min_num_sent) or last:
+                    if nsent >= min_num_sent:
+                        out.append((start, s + 1, target))
+                    start, tok, nsent = s + 1, 0, 0
+                    target, rng = draw_target(rng)
+                    if len(out) >= max_num_samples:
+                        break
+        if len(out) >= max_num_samples:
+            break
+    return np.asarray(out, dtype=np.int64).reshape(-1, 3)

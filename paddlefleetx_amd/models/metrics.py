@@ -13,7 +13,8 @@ from typing import List, Tuple
 
 import torch
 
-__all__ = ["Accuracy", "AccuracyAndF1", "Mcc", "PearsonAndSpearman"]
+__all__ = ["Accuracy", "AccuracyAndF1", "Mcc", "PearsonAndSpearman",
+           "MultiLabelsMetric"]
 
 
 class Metric:
@@ -159,3 +160,70 @@ class PearsonAndSpearman(Metric):
 
     def name(self):
         return "pearson", "spearman", "pearson and spearman"
+
+
+class MultiLabelsMetric(Metric):
+    """Per-label precision/recall/F1 with None/binary/micro/macro/weighted
+    averaging over streaming per-label confusion matrices
+    (metrics.py:445-620 semantics)."""
+
+    def __init__(self, num_labels: int):
+        assert num_labels > 1, "num_labels must be > 1"
+        self.num_labels = num_labels
+        self.reset()
+
+    def reset(self):
+        # [label][truth][pred] 2x2 one-vs-rest confusion counts
+        self.cm = torch.zeros(self.num_labels, 2, 2, dtype=torch.long)
+
+    def update(self, preds: torch.Tensor, labels: torch.Tensor):
+        if preds.ndim > 1:
+            preds = preds.argmax(dim=-1)
+        preds = preds.reshape(-1)
+        labels = labels.reshape(-1)
+        for c in range(self.num_labels):
+            p = preds == c
+            t = labels == c
+            self.cm[c, 1, 1] += int((p & t).sum())
+            self.cm[c, 1, 0] += int((~p & t).sum())
+            self.cm[c, 0, 1] += int((p & ~t).sum())
+            self.cm[c, 0, 0] += int((~p & ~t).sum())
+
+    @staticmethod
+    def _prf(tp, fp, fn):
+        precision = tp / (tp + fp) if tp + fp > 0 else 0.0
+        recall = tp / (tp + fn) if tp + fn > 0 else 0.0
+        f1 = 2 * precision * recall / (precision + recall) \
+            if precision + recall > 0 else 0.0
+        return precision, recall, f1
+
+    def accumulate(self, average=None, pos_label: int = 1):
+        tp = self.cm[:, 1, 1].float()
+        fp = self.cm[:, 0, 1].float()
+        fn = self.cm[:, 1, 0].float()
+        if average == "binary":
+            return self._prf(float(tp[pos_label]), float(fp[pos_label]),
+                             float(fn[pos_label]))
+        if average == "micro":
+            return self._prf(float(tp.sum()), float(fp.sum()),
+                             float(fn.sum()))
+        per = [self._prf(float(tp[c]), float(fp[c]), float(fn[c]))
+               for c in range(self.num_labels)]
+        if average == "macro":
+            n = self.num_labels
+            return (sum(p for p, _, _ in per) / n,
+                    sum(r for _, r, _ in per) / n,
+                    sum(f for _, _, f in per) / n)
+        if average == "weighted":
+            support = (tp + fn)
+            tot = float(support.sum()) or 1.0
+            w = [float(s) / tot for s in support]
+            return (sum(wi * p for wi, (p, _, _) in zip(w, per)),
+                    sum(wi * r for wi, (_, r, _) in zip(w, per)),
+                    sum(wi * f for wi, (_, _, f) in zip(w, per)))
+        # average=None: per-label arrays
+        return ([p for p, _, _ in per], [r for _, r, _ in per],
+                [f for _, _, f in per])
+
+    def name(self):
+        return "precision", "recall", "f1"

@@ -183,3 +183,20 @@ def test_finetune_module_cls_and_regression():
                    for j in range(3))
     loss2 = mod2.training_step(batch2)
     assert loss2.ndim == 0
+
+
+def test_multilabels_metric_reference_values():
+    """Values from the reference docstring (metrics.py:463-484)."""
+    import torch
+    from paddlefleetx_amd.models.metrics import MultiLabelsMetric
+    x = torch.tensor([[0.1, 0.2, 0.9], [0.5, 0.8, 0.5],
+                      [0.6, 1.5, 0.4], [2.8, 0.7, 0.3]])
+    y = torch.tensor([2, 1, 2, 1])
+    m = MultiLabelsMetric(num_labels=3)
+    m.update(x, y)
+    assert m.accumulate(average="micro") == (0.5, 0.5, 0.5)
+    p, r, f = m.accumulate(average="macro")
+    assert abs(p - 0.5) < 1e-9 and abs(r - 1 / 3) < 1e-9
+    p, r, f = m.accumulate(average="weighted")
+    assert abs(p - 0.75) < 1e-9 and abs(f - 0.5833333333333333) < 1e-9
+    assert m.accumulate(average="binary", pos_label=2)[0] == 1.0

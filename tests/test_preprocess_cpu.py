@@ -59,3 +59,29 @@ def test_full_preprocess_pipeline(tmp_path):
     tokens, pos, labels, mask = ds[0]
     assert tokens.shape == (32,)
     assert (labels[:-1] == tokens[1:]).all()  # packed shift-by-one
+
+
+def test_build_pair_mapping_cpp_matches_python():
+    """ERNIE/BERT sentence-pair mapping (reference
+    fast_index_map_helpers.cpp:195-430): C++ and Python paths are
+    bit-identical; samples respect min_num_sent and doc boundaries."""
+    import numpy as np
+    from paddlefleetx_amd.data import index_builder as ib
+    docs = np.array([0, 3, 5, 10, 11], dtype=np.int64)
+    sizes = np.array([30, 40, 50, 60, 20, 10, 10, 10, 10, 80, 5],
+                     dtype=np.int32)
+    args = (docs, sizes, 2, 100, 128, 0.3, 1234, 2)
+    got = ib.build_pair_mapping(*args)
+    saved = ib._cpp
+    try:
+        ib._cpp = None
+        py = ib.build_pair_mapping(*args)
+    finally:
+        ib._cpp = saved
+    assert np.array_equal(got, py)
+    assert got.shape[1] == 3 and len(got) > 0
+    # every sample stays inside one document and has >= 2 sentences
+    bounds = list(zip(docs[:-1], docs[1:]))
+    for s, e, t in got:
+        assert e - s >= 2 and 4 <= t <= 128
+        assert any(s >= lo and e <= hi for lo, hi in bounds)
