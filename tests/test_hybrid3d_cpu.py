@@ -11,7 +11,7 @@ import torch.multiprocessing as mp
 REPO = os.path.join(os.path.dirname(__file__), "..")
 
 
-def _worker(rank, world, port):
+def _worker(rank, world, port, vpp=1):
     import sys
     sys.path.insert(0, REPO)
     os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
@@ -37,7 +37,8 @@ def _worker(rank, world, port):
                   "attention_probs_dropout_prob": 0.0, "fused_attn": False},
         "Optimizer": {"name": "FusedAdamW", "weight_decay": 0.0,
                       "lr": {"name": "ConstantLR", "learning_rate": 1e-3}},
-        "Distributed": {"dp_degree": 2, "mp_degree": 2, "pp_degree": 2},
+        "Distributed": {"dp_degree": 2, "mp_degree": 2, "pp_degree": 2,
+                        "pipeline": {"virtual_pp_degree": vpp}},
     }
     module = build_module(cfg)
     engine = EagerEngine(cfg, module)
@@ -63,6 +64,21 @@ def test_dp2_tp2_pp2_world8():
     port = int(torch.randint(20000, 40000, (1,)))
     ctx = mp.get_context("spawn")
     procs = [ctx.Process(target=_worker, args=(r, 8, port))
+             for r in range(8)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(500)
+        assert p.exitcode == 0, f"worker failed with {p.exitcode}"
+
+
+@pytest.mark.timeout(600)
+def test_dp2_tp2_pp2_vpp2_world8():
+    """The driver topology with interleaved virtual stages on top
+    (4 layers -> 4 model chunks, acc=2 == pp_degree)."""
+    port = int(torch.randint(20000, 40000, (1,)))
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_worker, args=(r, 8, port, 2))
              for r in range(8)]
     for p in procs:
         p.start()
