@@ -237,6 +237,14 @@ class TransformerDecoderLayer(nn.Module):
 
     def _attn_block(self, x, cache=None, use_cache=False):
         h = self.ln1(x)
+        if self.use_recompute and self.recompute_granularity == "core_attn" \
+                and self.training and cache is None and not use_cache \
+                and torch.is_grad_enabled():
+            # recompute only the attention core (QK^T/softmax/PV + projs);
+            # reference granularity "core_attn" (hybrid_model.py:303-346)
+            a = checkpoint(lambda t: self.attn(t)[0], h,
+                           use_reentrant=False)
+            return x + self._dropout(a), None
         a, new_cache = self.attn(h, cache=cache, use_cache=use_cache)
         return x + self._dropout(a), new_cache
 

@@ -82,3 +82,33 @@ def test_checkpoint_roundtrip(tmp_path):
         assert n1 == n2
         assert torch.equal(p1.detach(), p2.detach()), n1
     assert engine2._load_recovery["step"] == 5
+
+
+def test_recompute_granularities_equivalent():
+    """full / full_attn / core_attn recompute all reproduce the
+    no-recompute grads (reference granularity surface,
+    hybrid_model.py:377/456/637)."""
+    import torch
+    from paddlefleetx_amd.models.gpt.model import (GPTForPretraining,
+                                                   GPTModel,
+                                                   GPTPretrainingCriterion)
+    grads = {}
+    for gran in (None, "full", "full_attn", "core_attn"):
+        torch.manual_seed(3)
+        m = GPTForPretraining(GPTModel(
+            vocab_size=128, hidden_size=32, num_layers=2,
+            num_attention_heads=4, max_position_embeddings=32,
+            fused_attn=False, hidden_dropout_prob=0.0,
+            attention_probs_dropout_prob=0.0,
+            use_recompute=gran is not None,
+            recompute_granularity=gran or "full"))
+        m.train()
+        torch.manual_seed(7)
+        tokens = torch.randint(0, 128, (2, 16))
+        labels = torch.randint(0, 128, (2, 16))
+        loss = GPTPretrainingCriterion()(m(tokens), labels,
+                                         torch.ones(2, 16))
+        loss.backward()
+        grads[gran] = m.gpt.layers[0].attn.qkv.weight.grad.clone()
+    for gran in ("full", "full_attn", "core_attn"):
+        assert torch.allclose(grads[None], grads[gran], atol=1e-6), gran
