@@ -69,6 +69,7 @@ class GPTForPretrainingPipe(PipelineModule):
                  sequence_parallel: bool = False,
                  initializer_range: float = 0.02,
                  virtual_pp_degree: int = 1,
+                 partial_send_recv: bool = False,
                  dtype: Optional[torch.dtype] = None, **unused: Any):
         ffn_hidden_size = ffn_hidden_size or 4 * hidden_size
         from paddlefleetx_amd.parallel.env import get_hcg
@@ -105,7 +106,9 @@ class GPTForPretrainingPipe(PipelineModule):
                                      init_std=initializer_range))
         super().__init__(descs, seg_method="layer:TransformerDecoderLayer",
                          act_dtype=dtype or torch.float32,
-                         num_virtual_stages=virtual_pp_degree)
+                         num_virtual_stages=virtual_pp_degree,
+                         partial_send_recv=partial_send_recv
+                         and not sequence_parallel)
         self.hidden_size = hidden_size
         self.sequence_parallel = sequence_parallel
         self._sp_degree = mp

@@ -97,7 +97,8 @@ class PipelineModule(nn.Module):
 
     def __init__(self, descs: List[LayerDesc], loss_shape_hint=None,
                  seg_method: str = "uniform", act_dtype=torch.bfloat16,
-                 num_virtual_stages: int = 1):
+                 num_virtual_stages: int = 1,
+                 partial_send_recv: bool = False):
         super().__init__()
         hcg = get_hcg()
         self.hcg = hcg
@@ -105,6 +106,14 @@ class PipelineModule(nn.Module):
         self.pp_size = hcg.get_pipe_parallel_world_size()
         self.act_dtype = act_dtype
         self.descs = descs
+        # partial send/recv (reference enable_partial_send_recv,
+        # env.py:143): with mp>1 the boundary activation is replicated
+        # across mp ranks, so each rank p2p-sends only its 1/mp flat
+        # chunk and the receiver allgathers over the mp group. Off by
+        # default on MI355X: full-tensor p2p rides a dedicated xGMI
+        # link, so the split mostly adds latency; the knob exists for
+        # parity and for bandwidth-starved interconnects.
+        self.partial_send_recv = bool(partial_send_recv)
         self.num_virtual = int(num_virtual_stages or 1)
         if self.num_virtual > 1 and self.pp_size == 1:
             # virtual stages only make sense with a real pipeline
@@ -277,45 +286,77 @@ class PipelineModule(nn.Module):
         is_first = self.pp_rank == 0
         is_last = self.pp_rank == self.pp_size - 1
 
+        # ---- partial send/recv: p2p only this mp rank's 1/mp flat
+        # chunk; the receiver allgathers over the mp group (boundary
+        # activations AND their grads are mp-replicated) ----
+        mp_g = hcg.get_model_parallel_group()
+        use_psr = (self.partial_send_recv and mp_g.world_size > 1)
+        numel = 1
+        for d in shape:
+            numel *= d
+        assert not use_psr or numel % mp_g.world_size == 0
+
+        def pack(t):
+            t = t.contiguous()
+            if not use_psr:
+                return t
+            return t.view(-1).chunk(mp_g.world_size)[mp_g.rank].contiguous()
+
+        def wire_shape():
+            return (numel // mp_g.world_size,) if use_psr else shape
+
+        def unwire(buf):
+            if not use_psr:
+                return buf
+            parts = [torch.empty_like(buf) for _ in range(mp_g.world_size)]
+            dist.all_gather(parts, buf, group=mp_g.group)
+            return torch.cat(parts).view(shape)
+
         def recv_forward():
             if is_first:
                 return None
-            buf = torch.empty(shape, dtype=self.act_dtype, device=device)
+            buf = torch.empty(wire_shape(), dtype=self.act_dtype,
+                              device=device)
             self._p2p([dist.P2POp(dist.irecv, buf, prev)])
-            buf.requires_grad_(True)
-            return buf
+            full = unwire(buf)
+            full.requires_grad_(True)
+            return full
 
         def send_forward(out):
             if not is_last:
-                self._p2p([dist.P2POp(dist.isend, out.detach().contiguous(), nxt)])
+                self._p2p([dist.P2POp(dist.isend, pack(out.detach()), nxt)])
 
         def recv_backward():
             if is_last:
                 return None
-            buf = torch.empty(shape, dtype=self.act_dtype, device=device)
+            buf = torch.empty(wire_shape(), dtype=self.act_dtype,
+                              device=device)
             self._p2p([dist.P2POp(dist.irecv, buf, nxt)])
-            return buf
+            return unwire(buf)
 
         def send_backward(in_grad):
             if not is_first and in_grad is not None:
-                self._p2p([dist.P2POp(dist.isend, in_grad.contiguous(), prev)])
+                self._p2p([dist.P2POp(dist.isend, pack(in_grad), prev)])
 
         def send_forward_recv_backward(out):
             if is_last:
                 return None
-            buf = torch.empty(shape, dtype=self.act_dtype, device=device)
-            self._p2p([dist.P2POp(dist.isend, out.detach().contiguous(), nxt),
+            buf = torch.empty(wire_shape(), dtype=self.act_dtype,
+                              device=device)
+            self._p2p([dist.P2POp(dist.isend, pack(out.detach()), nxt),
                        dist.P2POp(dist.irecv, buf, nxt)])
-            return buf
+            return unwire(buf)
 
         def send_backward_recv_forward(in_grad):
             if is_first:
                 return None
-            buf = torch.empty(shape, dtype=self.act_dtype, device=device)
-            self._p2p([dist.P2POp(dist.isend, in_grad.contiguous(), prev),
+            buf = torch.empty(wire_shape(), dtype=self.act_dtype,
+                              device=device)
+            self._p2p([dist.P2POp(dist.isend, pack(in_grad), prev),
                        dist.P2POp(dist.irecv, buf, prev)])
-            buf.requires_grad_(True)
-            return buf
+            full = unwire(buf)
+            full.requires_grad_(True)
+            return full
 
         def forward_step(inp, i):
             if is_first:

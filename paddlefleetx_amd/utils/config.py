@@ -166,7 +166,11 @@ def process_dist_config(cfg: AttrDict) -> None:
     pipeline = _setdef(dist, "pipeline", AttrDict())
     _setdef(pipeline, "schedule", "1F1B")
     _setdef(pipeline, "virtual_pp_degree", 1)
-    _setdef(pipeline, "enable_partial_send_recv", True)
+    # default OFF on MI355X: each pp p2p channel has a dedicated xGMI
+    # link, so splitting the boundary tensor across mp ranks (reference
+    # default True for PCIe-bound clusters) mostly adds an allgather of
+    # latency; the knob is implemented (parallel/pp.py) for parity
+    _setdef(pipeline, "enable_partial_send_recv", False)
     if cfg.get("Model", {}).get("sequence_parallel", False):
         # SP shards activations along seq; partial send recv is incompatible
         # (reference config.py:112-119)
