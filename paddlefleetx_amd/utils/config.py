@@ -215,7 +215,13 @@ def process_engine_config(cfg: AttrDict) -> None:
     e = _setdef(cfg, "Engine", AttrDict())
     g = cfg["Global"]
     e["accumulate_steps"] = g["local_batch_size"] // g["micro_batch_size"]
-    _setdef(e, "max_steps", g.get("max_steps"))
+    if g.get("max_steps") is not None:
+        # an explicit Global.max_steps (CLI override or yaml) wins over
+        # the Engine default so `-o Global.max_steps=N` always bounds
+        # the run
+        e["max_steps"] = g["max_steps"]
+    else:
+        _setdef(e, "max_steps", None)
     _setdef(e, "logging_freq", g.get("logging_freq", 10))
     _setdef(e, "eval_freq", g.get("eval_freq"))
     _setdef(e, "eval_iters", 10)
