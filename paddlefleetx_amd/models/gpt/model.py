@@ -55,10 +55,11 @@ class MultiHeadAttention(nn.Module):
                  dtype: Optional[torch.dtype] = None,
                  sequence_parallel: bool = False,
                  init_std: float = 0.02, output_layer_init_std: float = 0.02,
-                 cp_backend: str = "ulysses"):
+                 cp_backend: str = "ulysses", cp_zigzag: bool = False):
         super().__init__()
         assert cp_backend in ("ulysses", "ring")
         self.cp_backend = cp_backend
+        self.cp_zigzag = cp_zigzag
         mp = get_hcg().get_model_parallel_world_size()
         assert num_heads % mp == 0, f"heads {num_heads} not divisible by mp {mp}"
         assert hidden_size % num_heads == 0
@@ -105,7 +106,8 @@ class MultiHeadAttention(nn.Module):
             q, k, v = qkv.split(self.head_dim, dim=-1)  # [B, S/cp, h, D]
             if self.cp_backend == "ring":
                 from paddlefleetx_amd.parallel.ring import RingAttention
-                ca = RingAttention(scale=self.scale, causal=True)
+                ca = RingAttention(scale=self.scale, causal=True,
+                                   zigzag=self.cp_zigzag)
             else:
                 from paddlefleetx_amd.parallel.cp import UlyssesAttention
                 ca = UlyssesAttention(scale=self.scale, causal=True)
@@ -202,7 +204,7 @@ class TransformerDecoderLayer(nn.Module):
                  sequence_parallel: bool = False, init_std: float = 0.02,
                  num_layers_for_scale: int = 1, expert_module: Optional[nn.Module] = None,
                  recompute_granularity: str = "full", use_recompute: bool = False,
-                 cp_backend: str = "ulysses"):
+                 cp_backend: str = "ulysses", cp_zigzag: bool = False):
         super().__init__()
         out_std = init_std / math.sqrt(2.0 * num_layers_for_scale)
         self.ln1 = FusedLayerNorm(hidden_size, dtype=dtype)
@@ -222,7 +224,8 @@ class TransformerDecoderLayer(nn.Module):
                                        sequence_parallel=sequence_parallel,
                                        init_std=init_std,
                                        output_layer_init_std=out_std,
-                                       cp_backend=cp_backend)
+                                       cp_backend=cp_backend,
+                                       cp_zigzag=cp_zigzag)
         if expert_module is not None:
             self.ffn = expert_module
         else:
@@ -310,7 +313,7 @@ class GPTModel(nn.Module):
                  sequence_parallel: bool = False,
                  initializer_range: float = 0.02,
                  moe_configs: Optional[Dict[str, Any]] = None,
-                 cp_backend: str = "ulysses",
+                 cp_backend: str = "ulysses", cp_zigzag: bool = False,
                  dtype: Optional[torch.dtype] = None, **unused: Any):
         super().__init__()
         ffn_hidden_size = ffn_hidden_size or 4 * hidden_size
@@ -347,7 +350,8 @@ class GPTModel(nn.Module):
                                     use_recompute=use_recompute,
                                     recompute_granularity=recompute_granularity,
                                     expert_module=_make_expert_module(),
-                                    cp_backend=cp_backend)
+                                    cp_backend=cp_backend,
+                                    cp_zigzag=cp_zigzag)
             for _ in range(num_layers)])
         self.final_ln = FusedLayerNorm(hidden_size, dtype=dtype)
         if sequence_parallel:

@@ -108,12 +108,24 @@ class GPTModule(LanguageModule):
         hcg = get_hcg()
         cp = hcg.get_context_parallel_world_size()
         if cp > 1:
-            # each cp rank takes its sequence chunk (Ulysses sharding)
             r = hcg.get_context_parallel_rank()
-            batch = tuple(
-                torch.chunk(t, cp, dim=1)[r].contiguous()
-                if torch.is_tensor(t) and t.ndim >= 2 else t
-                for t in batch)
+            mcfg = self.configs.get("Model", {})
+            if mcfg.get("cp_backend") == "ring" and \
+                    mcfg.get("cp_zigzag", False):
+                # zigzag sharding: half-chunks (r, 2cp-1-r) level the
+                # causal work across the ring (parallel/ring.py)
+                from paddlefleetx_amd.parallel.ring import zigzag_slice
+                batch = tuple(
+                    zigzag_slice(t, cp, r, dim=1)
+                    if torch.is_tensor(t) and t.ndim >= 2 else t
+                    for t in batch)
+            else:
+                # each cp rank takes its sequence chunk (Ulysses/ring
+                # sequential sharding)
+                batch = tuple(
+                    torch.chunk(t, cp, dim=1)[r].contiguous()
+                    if torch.is_tensor(t) and t.ndim >= 2 else t
+                    for t in batch)
         return batch
 
 

@@ -72,35 +72,78 @@ def _rotate(tensors, g):
     return outs
 
 
+def _halves(rank: int, cp: int):
+    """Zigzag layout: rank r's local sequence = global half-chunks
+    (r, 2cp-1-r) concatenated — balances the causal triangle so every
+    rank attends the same number of key blocks."""
+    return (rank, 2 * cp - 1 - rank)
+
+
+def _pair_mode(gq: int, gk: int):
+    """full / causal / skip relation between two global half-chunks."""
+    if gq > gk:
+        return "full"
+    if gq == gk:
+        return "causal"
+    return None
+
+
 class _RingAttnFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, q, k, v, scale):
+    def forward(ctx, q, k, v, scale, zigzag):
         g = get_hcg().get_context_parallel_group()
         cp = g.world_size
         r = g.rank if cp > 1 else 0
+        Sl = q.shape[2]
+        assert not zigzag or Sl % 2 == 0
 
-        o_acc = None
-        lse_acc = None
+        def merge(acc, o_j, lse_j):
+            if acc[0] is None:
+                return [o_j.float(), lse_j]
+            o_acc, lse_acc = acc
+            lse_new = torch.logaddexp(lse_acc, lse_j)
+            w_old = torch.exp(lse_acc - lse_new)[..., None]
+            w_new = torch.exp(lse_j - lse_new)[..., None]
+            return [o_acc * w_old + o_j.float() * w_new, lse_new]
+
         cur_k, cur_v = k, v
-        for j in range(cp):
-            src = (r - j) % cp
-            if j < cp - 1:
-                nk, nv = _rotate([cur_k, cur_v], g)
-            if src <= r:
-                o_j, lse_j = _block_fwd(q, cur_k, cur_v, src == r, scale)
-                if o_acc is None:
-                    o_acc, lse_acc = o_j.float(), lse_j
-                else:
-                    lse_new = torch.logaddexp(lse_acc, lse_j)
-                    w_old = torch.exp(lse_acc - lse_new)[..., None]
-                    w_new = torch.exp(lse_j - lse_new)[..., None]
-                    o_acc = o_acc * w_old + o_j.float() * w_new
-                    lse_acc = lse_new
-            if j < cp - 1:
-                cur_k, cur_v = nk, nv
-        out = o_acc.to(q.dtype)
-        ctx.save_for_backward(q, k, v, out, lse_acc)
+        if not zigzag:
+            acc = [None, None]
+            for j in range(cp):
+                src = (r - j) % cp
+                if j < cp - 1:
+                    nk, nv = _rotate([cur_k, cur_v], g)
+                if src <= r:
+                    acc = merge(acc, *_block_fwd(q, cur_k, cur_v, src == r,
+                                                 scale))
+                if j < cp - 1:
+                    cur_k, cur_v = nk, nv
+            out, lse = acc[0].to(q.dtype), acc[1]
+        else:
+            h = Sl // 2
+            qh = (q[:, :, :h], q[:, :, h:])
+            accs = [[None, None], [None, None]]
+            for j in range(cp):
+                src = (r - j) % cp
+                if j < cp - 1:
+                    nk, nv = _rotate([cur_k, cur_v], g)
+                for qi, gq in enumerate(_halves(r, cp)):
+                    for ki, gk in enumerate(_halves(src, cp)):
+                        mode = _pair_mode(gq, gk)
+                        if mode is None:
+                            continue
+                        kk = cur_k[:, :, ki * h:(ki + 1) * h]
+                        vv = cur_v[:, :, ki * h:(ki + 1) * h]
+                        accs[qi] = merge(accs[qi], *_block_fwd(
+                            qh[qi].contiguous(), kk.contiguous(),
+                            vv.contiguous(), mode == "causal", scale))
+                if j < cp - 1:
+                    cur_k, cur_v = nk, nv
+            out = torch.cat([accs[0][0], accs[1][0]], dim=2).to(q.dtype)
+            lse = torch.cat([accs[0][1], accs[1][1]], dim=2)
+        ctx.save_for_backward(q, k, v, out, lse)
         ctx.scale = scale
+        ctx.zigzag = zigzag
         return out
 
     @staticmethod
@@ -115,43 +158,80 @@ class _RingAttnFn(torch.autograd.Function):
         cur_k, cur_v = k, v
         cur_dk = torch.zeros_like(k, dtype=torch.float32)
         cur_dv = torch.zeros_like(v, dtype=torch.float32)
+        h = q.shape[2] // 2
         for j in range(cp):
             src = (r - j) % cp
-            if src <= r:
-                dq_j, dk_j, dv_j = _block_bwd(do, q, cur_k, cur_v, o, lse,
-                                              src == r, ctx.scale)
-                dq_acc += dq_j.float()
-                cur_dk += dk_j.float()
-                cur_dv += dv_j.float()
+            if not ctx.zigzag:
+                if src <= r:
+                    dq_j, dk_j, dv_j = _block_bwd(do, q, cur_k, cur_v, o,
+                                                  lse, src == r, ctx.scale)
+                    dq_acc += dq_j.float()
+                    cur_dk += dk_j.float()
+                    cur_dv += dv_j.float()
+            else:
+                for qi, gq in enumerate(_halves(r, cp)):
+                    qs = slice(qi * h, (qi + 1) * h)
+                    for ki, gk in enumerate(_halves(src, cp)):
+                        mode = _pair_mode(gq, gk)
+                        if mode is None:
+                            continue
+                        ks = slice(ki * h, (ki + 1) * h)
+                        dq_j, dk_j, dv_j = _block_bwd(
+                            do[:, :, qs].contiguous(),
+                            q[:, :, qs].contiguous(),
+                            cur_k[:, :, ks].contiguous(),
+                            cur_v[:, :, ks].contiguous(),
+                            o[:, :, qs].contiguous(),
+                            lse[:, :, qs].contiguous(),
+                            mode == "causal", ctx.scale)
+                        dq_acc[:, :, qs] += dq_j.float()
+                        cur_dk[:, :, ks] += dk_j.float()
+                        cur_dv[:, :, ks] += dv_j.float()
             if cp > 1:
                 # rotate every step: after cp hops each (k, v, dk, dv)
                 # quartet is back at its owning rank
                 cur_k, cur_v, cur_dk, cur_dv = _rotate(
                     [cur_k, cur_v, cur_dk, cur_dv], g)
         return (dq_acc.to(q.dtype), cur_dk.to(k.dtype), cur_dv.to(v.dtype),
-                None)
+                None, None)
 
 
-def ring_attention(q, k, v, scale: Optional[float] = None):
-    """Causal ring attention; q,k,v [B, H, S/cp, D] sequence shards
-    (sequential sharding, same slicing as `pretreating_batch`)."""
+def ring_attention(q, k, v, scale: Optional[float] = None,
+                   zigzag: bool = False):
+    """Causal ring attention; q,k,v [B, H, S/cp, D] sequence shards.
+    zigzag=False: sequential sharding (rank r owns chunk r).
+    zigzag=True: rank r owns half-chunks (r, 2cp-1-r) — every rank then
+    attends the same number of key blocks, levelling the causal-triangle
+    load imbalance of sequential sharding (slice batches with
+    `zigzag_slice`)."""
     scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
     return _RingAttnFn.apply(q.contiguous(), k.contiguous(), v.contiguous(),
-                             scale)
+                             scale, zigzag)
+
+
+def zigzag_slice(t: torch.Tensor, cp: int, rank: int,
+                 dim: int = 1) -> torch.Tensor:
+    """Take rank's zigzag shard (half-chunks rank and 2cp-1-rank) of a
+    full-sequence tensor along `dim`."""
+    chunks = t.chunk(2 * cp, dim=dim)
+    return torch.cat([chunks[rank], chunks[2 * cp - 1 - rank]],
+                     dim=dim).contiguous()
 
 
 class RingAttention(torch.nn.Module):
     """Drop-in alternative to UlyssesAttention: q,k,v [B, S/cp, h, D] ->
     o [B, S/cp, h, D] (same interface/layout as parallel/cp.py)."""
 
-    def __init__(self, scale: Optional[float] = None, causal: bool = True):
+    def __init__(self, scale: Optional[float] = None, causal: bool = True,
+                 zigzag: bool = False):
         super().__init__()
         assert causal, "ring attention: causal only (GPT pretraining path)"
         self.scale = scale
+        self.zigzag = zigzag
 
     def forward(self, q, k, v):
         q = q.permute(0, 2, 1, 3)  # [B, h, Sl, D]
         k = k.permute(0, 2, 1, 3)
         v = v.permute(0, 2, 1, 3)
-        o = ring_attention(q, k, v, self.scale)
+        o = ring_attention(q, k, v, self.scale, zigzag=self.zigzag)
         return o.permute(0, 2, 1, 3)
