@@ -346,3 +346,42 @@ def test_zero_sharding_matches_unsharded():
     with tempfile.TemporaryDirectory() as tmpdir:
         _run(_shard_worker, 2, (tmpdir,))
         _run(_unsharded_worker, 1, (tmpdir,))
+
+
+def _seed_worker(rank, world, port):
+    """Seed discipline (reference env.py:34-98): replicated params agree
+    across mp ranks, TP shards differ (distinct local streams), and the
+    same seed reproduces the same init."""
+    hcg = _init(rank, world, port, mp_deg=2)
+    import torch.distributed as dist
+    from paddlefleetx_amd.models.gpt.model import GPTModel
+    from paddlefleetx_amd.parallel.env import set_seed
+
+    def build():
+        set_seed(1234)
+        torch.manual_seed(7)
+        return GPTModel(vocab_size=128, hidden_size=32, num_layers=2,
+                        num_attention_heads=4, max_position_embeddings=32,
+                        fused_attn=False)
+
+    m1, m2 = build(), build()
+    # determinism: identical init for identical seeds
+    for p1, p2 in zip(m1.parameters(), m2.parameters()):
+        assert torch.equal(p1, p2)
+    # replicated position table identical across mp ranks
+    pos = m1.embeddings.position_embeddings.weight
+    ref = pos.clone()
+    dist.broadcast(ref, src=0)
+    assert torch.equal(ref, pos)
+    # TP shard (qkv weight) differs across mp ranks (local_seed stream)
+    w = m1.layers[0].attn.qkv.weight
+    other = w.clone()
+    dist.broadcast(other, src=0)
+    if hcg.get_model_parallel_rank() == 1:
+        assert not torch.equal(other, w)
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_seed_discipline_mp2():
+    _run(_seed_worker, 2)
