@@ -186,3 +186,48 @@ def test_ernie_pipe2_matches_single():
     with tempfile.TemporaryDirectory() as tmpdir:
         _run(_pipe_ref_worker, 1, (tmpdir,))
         _run(_pipe2_worker, 2, (tmpdir,))
+
+
+def _pipe_vpp_worker(rank, world, port, tmpdir):
+    """ERNIE pipeline with interleaved virtual stages (pp2 x V2 over 4
+    encoder layers) matches the pp1 reference."""
+    _init(rank, world, port, pp=2)
+    from paddlefleetx_amd.models.ernie.pipeline_model import (
+        ErnieForPretrainingPipe, ErniePipeCriterion)
+    cfg4 = dict(CFG, num_hidden_layers=4)
+    m = ErnieForPretrainingPipe(virtual_pp_degree=2, **cfg4)
+    assert m.num_virtual == 2
+    for i, layer in enumerate(m.layers):
+        sd = torch.load(
+            os.path.join(tmpdir, f"vl_{m._layer_desc_idx[i]}.pt"),
+            weights_only=False)
+        layer.load_state_dict(sd)
+    loss = m.forward_backward_pipeline(_batch(), ErniePipeCriterion(),
+                                       accumulate_steps=2)
+    ref = torch.load(os.path.join(tmpdir, "vref.pt"), weights_only=False)
+    assert torch.allclose(loss, ref["loss"], atol=1e-5), (loss, ref["loss"])
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def _pipe_vpp_ref_worker(rank, world, port, tmpdir):
+    _init(rank, world, port, pp=1)
+    from paddlefleetx_amd.models.ernie.pipeline_model import (
+        ErnieForPretrainingPipe, ErniePipeCriterion)
+    torch.manual_seed(7)
+    cfg4 = dict(CFG, num_hidden_layers=4)
+    m = ErnieForPretrainingPipe(**cfg4)
+    for i, layer in enumerate(m.layers):
+        torch.save(layer.state_dict(),
+                   os.path.join(tmpdir, f"vl_{m._layer_desc_idx[i]}.pt"))
+    loss = m.forward_backward_pipeline(_batch(), ErniePipeCriterion(),
+                                       accumulate_steps=2)
+    torch.save({"loss": loss}, os.path.join(tmpdir, "vref.pt"))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_ernie_pipe_vpp2_matches_single():
+    with tempfile.TemporaryDirectory() as tmpdir:
+        _run(_pipe_vpp_ref_worker, 1, (tmpdir,))
+        _run(_pipe_vpp_worker, 2, (tmpdir,))
