@@ -55,20 +55,34 @@ def _block_bwd(do, q, k, v, o, lse, causal, scale):
     return ref.attention_bwd(do, q, k, v, o, lse, causal, scale)
 
 
-def _rotate(tensors, g):
-    """Send each tensor to the next ring rank, receive from the previous.
-    Returns the received tensors (blocking, one fused p2p batch)."""
+def _rotate_start(tensors, g):
+    """Post the ring exchange (send to next, recv from previous) and
+    return (requests, out_tensors, sent_refs) WITHOUT waiting — the
+    caller overlaps block compute with the transfer and calls
+    `_rotate_wait` when it needs the data. `sent_refs` keeps the
+    contiguous send copies alive until completion."""
     ranks = g.ranks
     me = g.rank
     nxt = ranks[(me + 1) % g.world_size]
     prv = ranks[(me - 1) % g.world_size]
     outs = [torch.empty_like(t) for t in tensors]
+    sends = [t.contiguous() for t in tensors]
     ops = []
-    for t, o in zip(tensors, outs):
-        ops.append(dist.P2POp(dist.isend, t.contiguous(), nxt))
+    for t, o in zip(sends, outs):
+        ops.append(dist.P2POp(dist.isend, t, nxt))
         ops.append(dist.P2POp(dist.irecv, o, prv))
-    for r in dist.batch_isend_irecv(ops):
+    return dist.batch_isend_irecv(ops), outs, sends
+
+
+def _rotate_wait(reqs):
+    for r in reqs:
         r.wait()
+
+
+def _rotate(tensors, g):
+    """Blocking ring exchange (used where no compute can overlap)."""
+    reqs, outs, _sends = _rotate_start(tensors, g)
+    _rotate_wait(reqs)
     return outs
 
 
@@ -111,13 +125,16 @@ class _RingAttnFn(torch.autograd.Function):
             acc = [None, None]
             for j in range(cp):
                 src = (r - j) % cp
-                if j < cp - 1:
-                    nk, nv = _rotate([cur_k, cur_v], g)
+                # post block j+1's transfer BEFORE computing block j so
+                # the hop overlaps the flash kernel
+                pend = _rotate_start([cur_k, cur_v], g) if j < cp - 1 \
+                    else None
                 if src <= r:
                     acc = merge(acc, *_block_fwd(q, cur_k, cur_v, src == r,
                                                  scale))
-                if j < cp - 1:
-                    cur_k, cur_v = nk, nv
+                if pend is not None:
+                    _rotate_wait(pend[0])
+                    cur_k, cur_v = pend[1]
             out, lse = acc[0].to(q.dtype), acc[1]
         else:
             h = Sl // 2
@@ -125,8 +142,8 @@ class _RingAttnFn(torch.autograd.Function):
             accs = [[None, None], [None, None]]
             for j in range(cp):
                 src = (r - j) % cp
-                if j < cp - 1:
-                    nk, nv = _rotate([cur_k, cur_v], g)
+                pend = _rotate_start([cur_k, cur_v], g) if j < cp - 1 \
+                    else None
                 for qi, gq in enumerate(_halves(r, cp)):
                     for ki, gk in enumerate(_halves(src, cp)):
                         mode = _pair_mode(gq, gk)
@@ -137,8 +154,9 @@ class _RingAttnFn(torch.autograd.Function):
                         accs[qi] = merge(accs[qi], *_block_fwd(
                             qh[qi].contiguous(), kk.contiguous(),
                             vv.contiguous(), mode == "causal", scale))
-                if j < cp - 1:
-                    cur_k, cur_v = nk, nv
+                if pend is not None:
+                    _rotate_wait(pend[0])
+                    cur_k, cur_v = pend[1]
             out = torch.cat([accs[0][0], accs[1][0]], dim=2).to(q.dtype)
             lse = torch.cat([accs[0][1], accs[1][1]], dim=2)
         ctx.save_for_backward(q, k, v, out, lse)
@@ -161,6 +179,9 @@ class _RingAttnFn(torch.autograd.Function):
         h = q.shape[2] // 2
         for j in range(cp):
             src = (r - j) % cp
+            # k/v for the NEXT step can travel during this step's block
+            # backward; the dk/dv accumulators move after the update
+            kv_pend = _rotate_start([cur_k, cur_v], g) if cp > 1 else None
             if not ctx.zigzag:
                 if src <= r:
                     dq_j, dk_j, dv_j = _block_bwd(do, q, cur_k, cur_v, o,
@@ -188,10 +209,11 @@ class _RingAttnFn(torch.autograd.Function):
                         cur_dk[:, :, ks] += dk_j.float()
                         cur_dv[:, :, ks] += dv_j.float()
             if cp > 1:
-                # rotate every step: after cp hops each (k, v, dk, dv)
-                # quartet is back at its owning rank
-                cur_k, cur_v, cur_dk, cur_dv = _rotate(
-                    [cur_k, cur_v, cur_dk, cur_dv], g)
+                # after cp hops each (k, v, dk, dv) quartet is back at
+                # its owning rank
+                _rotate_wait(kv_pend[0])
+                cur_k, cur_v = kv_pend[1]
+                cur_dk, cur_dv = _rotate([cur_dk, cur_dv], g)
         return (dq_acc.to(q.dtype), cur_dk.to(k.dtype), cur_dv.to(v.dtype),
                 None, None)
 
