@@ -61,7 +61,7 @@ def _worker(rank, world, port, vpp=1):
 
 @pytest.mark.timeout(600)
 def test_dp2_tp2_pp2_world8():
-    port = int(torch.randint(20000, 40000, (1,)))
+    from port_util import free_port; port = free_port()
     ctx = mp.get_context("spawn")
     procs = [ctx.Process(target=_worker, args=(r, 8, port))
              for r in range(8)]
@@ -76,7 +76,7 @@ def test_dp2_tp2_pp2_world8():
 def test_dp2_tp2_pp2_vpp2_world8():
     """The driver topology with interleaved virtual stages on top
     (4 layers -> 4 model chunks, acc=2 == pp_degree)."""
-    port = int(torch.randint(20000, 40000, (1,)))
+    from port_util import free_port; port = free_port()
     ctx = mp.get_context("spawn")
     procs = [ctx.Process(target=_worker, args=(r, 8, port, 2))
              for r in range(8)]

@@ -72,7 +72,7 @@ def _worker(rank, world, port, tmpdir):
 
 @pytest.mark.timeout(600)
 def test_gpt_sp_pipeline_matches_plain():
-    port = int(torch.randint(20000, 40000, (1,)))
+    from port_util import free_port; port = free_port()
     ctx = mp.get_context("spawn")
     with tempfile.TemporaryDirectory() as tmpdir:
         procs = [ctx.Process(target=_worker, args=(r, 4, port, tmpdir))
@@ -136,7 +136,7 @@ def _psr_worker(rank, world, port, tmpdir):
 def test_partial_send_recv_matches_full():
     """enable_partial_send_recv: 1/mp chunked pp p2p + mp allgather must
     be exactly equivalent to full-tensor p2p (reference env.py:143)."""
-    port = int(torch.randint(20000, 40000, (1,)))
+    from port_util import free_port; port = free_port()
     ctx = mp.get_context("spawn")
     with tempfile.TemporaryDirectory() as tmpdir:
         procs = [ctx.Process(target=_psr_worker, args=(r, 4, port, tmpdir))

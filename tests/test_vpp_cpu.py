@@ -32,7 +32,7 @@ def _init(rank, world, port, pp=1):
 
 
 def _run(fn, world, args=()):
-    port = int(torch.randint(20000, 40000, (1,)))
+    from port_util import free_port; port = free_port()
     ctx = mp.get_context("spawn")
     procs = [ctx.Process(target=fn, args=(r, world, port) + args)
              for r in range(world)]
