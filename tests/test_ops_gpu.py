@@ -146,6 +146,8 @@ def test_flash_attention_fwd(dev, cfg):
 @pytest.mark.parametrize("cfg", [
     dict(B=2, H=4, S=256, D=128),
     dict(B=1, H=2, S=512, D=64),
+    dict(B=2, H=2, S=192, D=64),   # ragged S (tail masking in dkv/dq)
+    dict(B=1, H=2, S=320, D=128),  # ragged S, D=128
 ])
 def test_flash_attention_bwd(dev, cfg):
     B, H, S, D = cfg["B"], cfg["H"], cfg["S"], cfg["D"]

@@ -143,3 +143,41 @@ def test_vpp_requires_divisible_microbatches():
     bounds = PipelineModule._partition_chunks(
         None, descs, "uniform", 4)
     assert bounds == [(0, 2), (2, 4), (4, 6), (6, 8)]
+
+
+@pytest.mark.timeout(600)
+def test_interleaved_deep_steady_state():
+    """M=8 micro-batches (deep 1F1B steady state, multiple schedule
+    groups per chunk)."""
+    with tempfile.TemporaryDirectory() as tmpdir:
+        _run(_ref_worker, 1, (tmpdir, 8))
+        _run(_vpp_worker, 2, (tmpdir, 8))
+
+
+def test_partition_layer_seg_with_remainder():
+    """`layer:` segmentation over P*V chunks with a non-divisible layer
+    count: leading/trailing aux descs pin to the edge chunks, matched
+    layers balance with the remainder up front."""
+    import sys
+    sys.path.insert(0, REPO)
+    import torch.nn as nn
+    from paddlefleetx_amd.parallel.pp import LayerDesc, PipelineModule
+
+    class Emb(nn.Module):
+        pass
+
+    class Dec(nn.Module):
+        pass
+
+    class Head(nn.Module):
+        pass
+
+    descs = [LayerDesc(Emb)] + [LayerDesc(Dec) for _ in range(5)] + \
+        [LayerDesc(Head)]
+    bounds = PipelineModule._partition_chunks(None, descs, "layer:Dec", 4)
+    # 5 matched layers -> counts [2,1,1,1]; chunk0 absorbs the embedding,
+    # the last chunk absorbs the head
+    assert bounds == [(0, 3), (3, 4), (4, 5), (5, 7)]
+    # every desc assigned exactly once
+    covered = sorted(i for lo, hi in bounds for i in range(lo, hi))
+    assert covered == list(range(len(descs)))
