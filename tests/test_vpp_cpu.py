@@ -181,3 +181,29 @@ def test_partition_layer_seg_with_remainder():
     # every desc assigned exactly once
     covered = sorted(i for lo, hi in bounds for i in range(lo, hi))
     assert covered == list(range(len(descs)))
+
+
+def _vpp_eval_worker(rank, world, port, tmpdir):
+    _init(rank, world, port, pp=2)
+    from paddlefleetx_amd.models.gpt.model import GPTPretrainingCriterion
+    m = _make_model(vpp=2)
+    for i, layer in enumerate(m.layers):
+        gi = m._layer_desc_idx[i]
+        sd = torch.load(os.path.join(tmpdir, f"layer_{gi}.pt"),
+                        weights_only=False)
+        layer.load_state_dict(sd)
+    loss = m.eval_pipeline(_batch(), GPTPretrainingCriterion(),
+                           accumulate_steps=4)
+    ref = torch.load(os.path.join(tmpdir, "ref.pt"), weights_only=False)
+    assert torch.allclose(loss, ref["loss"], atol=1e-5), (loss, ref["loss"])
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_interleaved_eval_pipeline():
+    """Forward-only interleaved walk returns the same loss as the
+    training-schedule forward."""
+    with tempfile.TemporaryDirectory() as tmpdir:
+        _run(_ref_worker, 1, (tmpdir, 4))
+        _run(_vpp_eval_worker, 2, (tmpdir,))
