@@ -204,11 +204,10 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
   for (int rb = 0; rb < RB; ++rb)
 #pragma unroll
     for (int i = 0; i < NDT; ++i) oacc[rb][i] = f4{0.f, 0.f, 0.f, 0.f};
-  float m_r[RB][4], l_r[RB][4];
+  // per-lane ONE-row softmax state (swapped-QK^T layout: row = ccol)
+  float m_r[RB][1], l_r[RB][1];
 #pragma unroll
-  for (int rb = 0; rb < RB; ++rb)
-#pragma unroll
-    for (int r = 0; r < 4; ++r) { m_r[rb][r] = -INFINITY; l_r[rb][r] = 0.f; }
+  for (int rb = 0; rb < RB; ++rb) { m_r[rb][0] = -INFINITY; l_r[rb][0] = 0.f; }
 
   const int ccol = lane & 15;
   const int crow4 = (lane >> 4) * 4;
@@ -294,9 +293,12 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
     const bool active = !CAUSAL ||
         (kv0 <= min(qrow0 + RB * 16 - 1, S - 1));
     if (active) {
-      // ---- QK^T: both 16-row groups share each K B-fragment read; the
-      // (ct,kc) stream is software-pipelined 2 deep so ds_read latency
-      // hides under the previous fragment's 2 MFMAs ----
+      // ---- swapped QK^T (guide T12 direction): mfma(K, Q) gives
+      // C[kv][q] so each lane's 16 values are 16 kv scores of ONE q row
+      // (row = ccol) — softmax state collapses to scalars and the
+      // cross-lane reduction is 2 shuffles instead of a 4-step chain
+      // per row. A- and B-fragments share a register layout, so the
+      // operand swap is just argument order. ----
       f4 s[RB][4];
       bf8 kbuf[3];
       kbuf[0] = read_b_frag(kb_lds, KRS, 0, 0, lane);
@@ -313,88 +315,76 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
 #pragma unroll
         for (int rb = 0; rb < RB; ++rb) {
           f4 acc = (kc == 0) ? f4{0.f, 0.f, 0.f, 0.f} : s[rb][ct];
-          s[rb][ct] = MFMA_BF16(qfrag[rb][kc], kb, acc);
+          s[rb][ct] = MFMA_BF16(kb, qfrag[rb][kc], acc);
         }
       }
 
-      // ---- mask + scale + online softmax (both row groups: 2x VALU ILP) --
+      // ---- mask + scale + per-lane online softmax ----
 #pragma unroll
       for (int rb = 0; rb < RB; ++rb) {
-        const int my_qrow = qrow0 + rb * 16 + crow4;
-        float pmax[4] = {-INFINITY, -INFINITY, -INFINITY, -INFINITY};
+        const int my_qrow = qrow0 + rb * 16 + ccol;  // this lane's q row
+        float pmax = -INFINITY;
 #pragma unroll
         for (int ct = 0; ct < 4; ++ct) {
-          int kcol = kv0 + ct * 16 + ccol;
 #pragma unroll
           for (int r = 0; r < 4; ++r) {
+            int kcol = kv0 + ct * 16 + crow4 + r;
             float val = s[rb][ct][r] * scale;
-            if ((CAUSAL && kcol > my_qrow + r) || kcol >= S) val = -INFINITY;
+            if ((CAUSAL && kcol > my_qrow) || kcol >= S) val = -INFINITY;
             s[rb][ct][r] = val;
-            pmax[r] = fmaxf(pmax[r], val);
+            pmax = fmaxf(pmax, val);
           }
         }
-#pragma unroll
-        for (int r = 0; r < 4; ++r) pmax[r] = group16_reduce_max(pmax[r]);
+        // row max across the 4 lane groups holding this q row
+        pmax = fmaxf(pmax, __shfl_xor(pmax, 16, WAVE));
+        pmax = fmaxf(pmax, __shfl_xor(pmax, 32, WAVE));
 
-        // rescale only when some row's max actually grew (alpha==1 is
-        // exact otherwise) — wave-uniform skip saves the O(NDT*4) VALU
-        // rescale on most tiles
-        bool grew = false;
-#pragma unroll
-        for (int r = 0; r < 4; ++r) grew |= pmax[r] > m_r[rb][r];
+        bool grew = pmax > m_r[rb][0];
         if (__builtin_amdgcn_ballot_w64(grew)) {
-          float alpha[4];
+          float nm = fmaxf(m_r[rb][0], pmax);
+          float alpha = (m_r[rb][0] == -INFINITY) ? 0.f
+              : __expf(m_r[rb][0] - nm);
+          m_r[rb][0] = (m_r[rb][0] == -INFINITY && pmax == -INFINITY)
+              ? -INFINITY : nm;
+          l_r[rb][0] *= alpha;
+          // redistribute alpha from softmax rows (row=ccol) to the O
+          // accumulator rows (row=crow4+r): row x's alpha lives in lane x
+          float a4[4];
 #pragma unroll
-          for (int r = 0; r < 4; ++r) {
-            float nm = fmaxf(m_r[rb][r], pmax[r]);
-            alpha[r] = (m_r[rb][r] == -INFINITY) ? 0.f
-                : __expf(m_r[rb][r] - nm);
-            m_r[rb][r] = (m_r[rb][r] == -INFINITY && pmax[r] == -INFINITY)
-                ? -INFINITY : nm;
-            l_r[rb][r] *= alpha[r];
-          }
+          for (int r = 0; r < 4; ++r)
+            a4[r] = __shfl(alpha, crow4 + r, WAVE);
 #pragma unroll
           for (int i = 0; i < NDT; ++i) {
 #pragma unroll
-            for (int r = 0; r < 4; ++r) oacc[rb][i][r] *= alpha[r];
+            for (int r = 0; r < 4; ++r) oacc[rb][i][r] *= a4[r];
           }
         }
 
-        // exp base: -inf rows exponentiate against 0 so exp(-inf-0)=0
-        // without a per-element select
-        float me[4];
-#pragma unroll
-        for (int r = 0; r < 4; ++r)
-          me[r] = (m_r[rb][r] == -INFINITY) ? 0.f : m_r[rb][r];
-
-        float psum[4] = {0.f, 0.f, 0.f, 0.f};
+        float me = (m_r[rb][0] == -INFINITY) ? 0.f : m_r[rb][0];
+        float psum = 0.f;
 #pragma unroll
         for (int ct = 0; ct < 4; ++ct) {
 #pragma unroll
           for (int r = 0; r < 4; ++r) {
-            float p = __expf(s[rb][ct][r] - me[r]);
-            s[rb][ct][r] = p;
-            psum[r] += p;
+            float pv = __expf(s[rb][ct][r] - me);
+            s[rb][ct][r] = pv;
+            psum += pv;
           }
         }
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          psum[r] = group16_reduce_sum(psum[r]);
-          l_r[rb][r] += psum[r];
-        }
+        psum += __shfl_xor(psum, 16, WAVE);
+        psum += __shfl_xor(psum, 32, WAVE);
+        l_r[rb][0] += psum;
 
-        // ---- P -> per-wave LDS for A-fragments (cvt_pk packs row
-        // pairs: 1 VALU conversion per 2 values) ----
+        // ---- P^T -> per-wave LDS [q][kv] image: each lane writes its
+        // own q row; the 4 r-values are CONSECUTIVE kv columns, so two
+        // cvt_pk + two b32 stores cover a ct tile ----
+        unsigned short* prow = myp + (rb * 16 + ccol) * PRS + crow4;
 #pragma unroll
         for (int ct = 0; ct < 4; ++ct) {
-#pragma unroll
-          for (int r = 0; r < 4; r += 2) {
-            unsigned int u = cvt_pk_bf16(s[rb][ct][r], s[rb][ct][r + 1]);
-            unsigned short* base =
-                myp + (rb * 16 + crow4 + r) * PRS + ct * 16 + ccol;
-            base[0] = (unsigned short)u;
-            base[PRS] = (unsigned short)(u >> 16);
-          }
+          unsigned int lo = cvt_pk_bf16(s[rb][ct][0], s[rb][ct][1]);
+          unsigned int hi = cvt_pk_bf16(s[rb][ct][2], s[rb][ct][3]);
+          *reinterpret_cast<unsigned int*>(prow + ct * 16) = lo;
+          *reinterpret_cast<unsigned int*>(prow + ct * 16 + 2) = hi;
         }
       }
 
@@ -431,29 +421,35 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
     __syncthreads();
   }
 
-  // ---- epilogue (strided o) ----
+  // ---- epilogue (strided o): softmax state lives on row=ccol lanes;
+  // redistribute 1/l to the O-accumulator rows (crow4+r) via shfl ----
   __hip_bfloat16* op = o.at(b, hh);
 #pragma unroll
-  for (int rb = 0; rb < RB; ++rb)
+  for (int rb = 0; rb < RB; ++rb) {
+    float inv_own = (l_r[rb][0] > 0.f) ? 1.f / l_r[rb][0] : 0.f;
+    float inv4[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) inv4[r] = __shfl(inv_own, crow4 + r, WAVE);
+    if ((lane >> 4) == 0 && lse_out) {
+      int qrow = qrow0 + rb * 16 + ccol;
+      if (qrow < S)
+        lse_out[bh * (long)S + qrow] =
+            (l_r[rb][0] > 0.f) ? m_r[rb][0] + logf(l_r[rb][0]) : -INFINITY;
+    }
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       int qrow = qrow0 + rb * 16 + crow4 + r;
       if (qrow >= S) continue;
-      float inv = (l_r[rb][r] > 0.f) ? 1.f / l_r[rb][r] : 0.f;
       unsigned short* orow = (unsigned short*)op + (long)qrow * o.rs;
 #pragma unroll
       for (int dt = 0; dt < NDT; dt += 2) {
-        unsigned int u = cvt_pk_bf16(oacc[rb][dt][r] * inv,
-                                     oacc[rb][dt + 1][r] * inv);
+        unsigned int u = cvt_pk_bf16(oacc[rb][dt][r] * inv4[r],
+                                     oacc[rb][dt + 1][r] * inv4[r]);
         orow[dt * 16 + ccol] = (unsigned short)u;
         orow[(dt + 1) * 16 + ccol] = (unsigned short)(u >> 16);
       }
-      if (ccol == 0 && lse_out) {
-        float lv = (l_r[rb][r] > 0.f) ? m_r[rb][r] + logf(l_r[rb][r])
-                                      : -INFINITY;
-        lse_out[bh * (long)S + qrow] = lv;
-      }
     }
+  }
 }
 
 // ===========================================================================
