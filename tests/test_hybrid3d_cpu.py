@@ -155,3 +155,66 @@ def test_pp2_checkpoint_roundtrip():
         for p in procs:
             p.join(300)
             assert p.exitcode == 0, f"worker failed with {p.exitcode}"
+
+
+def _fp16_pp_worker(rank, world, port):
+    """fp16 dynamic loss scaling composed with the 1F1B pipeline: the
+    scale rides `forward_backward_pipeline(scale=...)`, grads unscale in
+    the optimizer step, and the loss/params stay finite."""
+    import sys
+    sys.path.insert(0, REPO)
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank))
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from paddlefleetx_amd.core.engine import EagerEngine
+    from paddlefleetx_amd.models import build_module
+    from paddlefleetx_amd.parallel.env import set_hcg, set_seed
+    from paddlefleetx_amd.parallel.topology import HybridTopology
+    hcg = HybridTopology(pp=2)
+    set_hcg(hcg)
+    set_seed(1234)
+    cfg = {
+        "Global": {"global_batch_size": 4},
+        "Engine": {"mix_precision": {"enable": True, "dtype": "float16",
+                                     "scale_loss": 1024.0},
+                   "accumulate_steps": 2},
+        "Model": {"name": "GPTModule", "vocab_size": 128, "hidden_size": 32,
+                  "num_layers": 4, "num_attention_heads": 2,
+                  "max_position_embeddings": 16,
+                  "hidden_dropout_prob": 0.0,
+                  "attention_probs_dropout_prob": 0.0, "fused_attn": False},
+        "Optimizer": {"name": "FusedAdamW", "weight_decay": 0.0,
+                      "lr": {"name": "ConstantLR", "learning_rate": 1e-3}},
+        "Distributed": {"pp_degree": 2},
+    }
+    module = build_module(cfg)
+    engine = EagerEngine(cfg, module)
+    assert engine.loss_scale == 1024.0
+    torch.manual_seed(31)
+    batch = (torch.randint(0, 128, (4, 16)),
+             torch.arange(16).repeat(4, 1),
+             torch.randint(0, 128, (4, 16)), torch.ones(4, 16))
+    for _ in range(2):
+        loss = engine._fit_impl(batch)
+    if hcg.is_last_stage():
+        assert torch.isfinite(loss), loss
+    for p in module.model.parameters():
+        assert torch.isfinite(p).all()
+    assert engine._found_inf == 0.0
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_fp16_scaler_with_pipeline():
+    from port_util import free_port
+    port = free_port()
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_fp16_pp_worker, args=(r, 2, port))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(300)
+        assert p.exitcode == 0, f"worker failed with {p.exitcode}"
