@@ -3,13 +3,18 @@
 // Replaces the reference's paddle `flash_attention` consumption
 // (ppfleetx hybrid_model.py:284-301) with hand-written MFMA kernels.
 //
-// Forward (v2): 8-wave workgroup, 128-row Q tile (16 rows/wave), 64-row
-// KV tiles double-buffered in LDS with split staging (issue global->reg
-// loads BEFORE the tile's compute, ds_write after — guide T14): HBM
-// latency for tile t+1 hides under tile t's MFMAs, one __syncthreads per
-// tile. Online softmax in registers per 16-row fragment; hardware
-// v_exp (__expf). mfma_f32_16x16x32_bf16 tiles; K row-major +16B-padded
-// (conflict-free b128 reads), V transposed at stage time.
+// Forward: 8-wave workgroup, 256-row Q tile (RB=2 16-row groups/wave),
+// 64-row KV tiles double-buffered in LDS with split staging (issue
+// global->reg loads BEFORE the tile's compute, ds_write after — guide
+// T14): HBM latency for tile t+1 hides under tile t's MFMAs, one
+// __syncthreads per tile. QK^T runs with SWAPPED operands (mfma(K, Q) ->
+// C[kv][q]) so each lane's 16 scores belong to ONE q row: the online
+// softmax state is a per-lane scalar, row reductions are 2 shuffles, and
+// P lands in the per-wave LDS image with two b32 stores per 16-col tile.
+// bh rides blockIdx.x so the linear dispatch deals every CU one block of
+// each causal tile position (load balance). hardware v_exp (__expf);
+// mfma_f32_16x16x32_bf16 tiles; K row-major +16B-padded (conflict-free
+// b128 reads), V transposed at stage time.
 //
 // Backward: split into dKV kernel (parallel over kv tiles) and dQ kernel
 // (parallel over q tiles) so neither needs atomics; both recompute
