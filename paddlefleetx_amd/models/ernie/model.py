@@ -281,9 +281,16 @@ class ErnieLMPredictionHead(nn.Module):
         if embedding_weights is not None:
             self.decoder_weight = embedding_weights
         else:
+            # untied: shard the decoder weight on vocab like the tied
+            # VocabParallelEmbedding so the parallel_matmul output and the
+            # sharded bias agree at [.., V/mp]
+            vocab_n = vocab_size // self.mp if self.mp > 1 else vocab_size
             self.decoder_weight = nn.Parameter(
-                torch.empty(vocab_size, hidden_size, dtype=dtype))
+                torch.empty(vocab_n, hidden_size, dtype=dtype))
             nn.init.normal_(self.decoder_weight, std=0.02)
+            if self.mp > 1:
+                self.decoder_weight.is_mp = True
+                self.decoder_weight.partition_dim = 0
         bias_n = vocab_size // self.mp if self.mp > 1 else vocab_size
         self.decoder_bias = nn.Parameter(torch.zeros(bias_n, dtype=dtype))
         if self.mp > 1:

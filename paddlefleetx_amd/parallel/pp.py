@@ -115,6 +115,14 @@ class PipelineModule(nn.Module):
         # parity and for bandwidth-starved interconnects.
         self.partial_send_recv = bool(partial_send_recv)
         self.num_virtual = int(num_virtual_stages or 1)
+        if self.partial_send_recv and self.num_virtual > 1:
+            # the interleaved schedule's fused comm slots always move
+            # full-shape activations; honour the request loudly instead
+            # of silently ignoring it
+            logger.warning(
+                "enable_partial_send_recv is not supported with "
+                "virtual_pp_degree > 1; sending full activations")
+            self.partial_send_recv = False
         if self.num_virtual > 1 and self.pp_size == 1:
             # virtual stages only make sense with a real pipeline
             # (reference utils.py:96-99)

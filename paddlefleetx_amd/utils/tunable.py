@@ -35,7 +35,8 @@ def enable_tuned_gemms(csv_path: str | None = None) -> bool:
         return True
     if not torch.cuda.is_available():
         return False
-    if os.environ.get("PYTORCH_TUNABLEOP_ENABLED"):
+    env = os.environ.get("PYTORCH_TUNABLEOP_ENABLED", "").strip()
+    if env and env != "0":
         return False  # user-driven session (tuning or custom file)
     path = csv_path or _DEFAULT
     if not os.path.exists(path):
