@@ -6,8 +6,8 @@ Single node, one rank per GPU over RCCL:
   python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
       --master-addr 127.0.0.1 bench.py --gpus N ...
 
-Parallelism by N (scaled-down versions of the 8-GPU DP2xTP2xPP2 target):
-  1 -> single | 2 -> TP2 | 4 -> TP2xPP2 | 8 -> DP2xTP2xPP2
+Parallelism by N (weak scaling toward the 8-GPU DP2xTP2xPP2 target):
+  1 -> single | 2 -> DP2 | 4 -> DP4 | 8 -> DP2xTP2xPP2
 
 Rank 0 prints ONE JSON line with the whole-job tokens/s.
 """
@@ -30,7 +30,7 @@ def parse_args():
     p.add_argument("--steps", type=int, default=10)
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--model", type=str, default="GPT-6.7B",
-                   choices=["GPT-345M", "GPT-1.3B", "GPT-6.7B"])
+                   choices=["GPT-TEST", "GPT-345M", "GPT-1.3B", "GPT-6.7B"])
     p.add_argument("--micro-batch", type=int, default=None)
     p.add_argument("--acc-steps", type=int, default=None)
     p.add_argument("--seq-len", type=int, default=1024)
@@ -38,6 +38,9 @@ def parse_args():
 
 
 MODELS = {
+    # tiny config for CPU rehearsal of the multi-rank code paths
+    # (tests/test_bench_rehearsal_cpu.py); never a reportable number
+    "GPT-TEST": dict(hidden_size=64, num_layers=4, num_attention_heads=4),
     "GPT-345M": dict(hidden_size=1024, num_layers=24, num_attention_heads=16),
     "GPT-1.3B": dict(hidden_size=2048, num_layers=24, num_attention_heads=16),
     "GPT-6.7B": dict(hidden_size=4096, num_layers=32, num_attention_heads=32),
