@@ -13,6 +13,16 @@ std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w,
                                        double eps);
 std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
                                        torch::Tensor w, torch::Tensor rstd);
+std::vector<torch::Tensor> layernorm_fwd_residual(torch::Tensor x,
+                                                  torch::Tensor res,
+                                                  torch::Tensor w,
+                                                  torch::Tensor b, double eps);
+std::vector<torch::Tensor> layernorm_bwd_residual(torch::Tensor dy,
+                                                  torch::Tensor x,
+                                                  torch::Tensor w,
+                                                  torch::Tensor mean,
+                                                  torch::Tensor rstd,
+                                                  torch::Tensor dsum);
 // elementwise.hip
 torch::Tensor bias_gelu_fwd(torch::Tensor x, torch::Tensor bias);
 std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
@@ -21,6 +31,7 @@ void adamw_flat(torch::Tensor master, torch::Tensor grad, torch::Tensor m,
                 torch::Tensor v, torch::Tensor model, double lr, double beta1,
                 double beta2, double eps, double wd, long step);
 torch::Tensor rope_fwd(torch::Tensor x, torch::Tensor cos, torch::Tensor sin);
+torch::Tensor colsum(torch::Tensor x);
 // softmax.hip
 torch::Tensor softmax_causal_fwd(torch::Tensor s, double scale);
 torch::Tensor softmax_causal_bwd(torch::Tensor dy, torch::Tensor y,
@@ -63,6 +74,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_bwd", &layernorm_bwd, "fused LayerNorm bwd");
   m.def("rmsnorm_fwd", &rmsnorm_fwd, "fused RMSNorm fwd");
   m.def("rmsnorm_bwd", &rmsnorm_bwd, "fused RMSNorm bwd");
+  m.def("layernorm_fwd_residual", &layernorm_fwd_residual,
+        "LN(a+b) fwd, also emits a+b");
+  m.def("layernorm_bwd_residual", &layernorm_bwd_residual,
+        "LN bwd with fused residual-grad addend");
+  m.def("colsum", &colsum, "column sum [N,H] -> fp32 [H] (dbias)");
   m.def("bias_gelu_fwd", &bias_gelu_fwd, "fused bias+gelu fwd");
   m.def("bias_gelu_bwd", &bias_gelu_bwd, "fused bias+gelu bwd");
   m.def("adamw_flat", &adamw_flat, "fused AdamW on flat buffers");

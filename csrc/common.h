@@ -114,6 +114,62 @@ DEV_INLINE float block_reduce_max(float x, float* lds_scratch) {
   return r;
 }
 
+// ---- width-generic vector load/store (bf16 raw-short or fp32), VEC 4/8 ----
+template <typename T, int VEC>
+DEV_INLINE void vload(const T* p, float* v) {
+  if constexpr (sizeof(T) == 2) {
+    if constexpr (VEC == 8) {
+      short8v pk = *reinterpret_cast<const short8v*>(p);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v[j] = bf_raw2f(((unsigned short*)&pk)[j]);
+    } else {
+      short4v pk = *reinterpret_cast<const short4v*>(p);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) v[j] = bf_raw2f(((unsigned short*)&pk)[j]);
+    }
+  } else {
+#pragma unroll
+    for (int j = 0; j < VEC; j += 4) {
+      float4v pk = *reinterpret_cast<const float4v*>((const float*)p + j);
+#pragma unroll
+      for (int k = 0; k < 4; ++k) v[j + k] = ((float*)&pk)[k];
+    }
+  }
+}
+
+template <typename T, int VEC>
+DEV_INLINE void vstore(T* p, const float* v) {
+  if constexpr (sizeof(T) == 2) {
+    if constexpr (VEC == 8) {
+      short8v out;
+#pragma unroll
+      for (int j = 0; j < 8; j += 2) {
+        unsigned int u = cvt_pk_bf16(v[j], v[j + 1]);
+        ((unsigned short*)&out)[j] = (unsigned short)u;
+        ((unsigned short*)&out)[j + 1] = (unsigned short)(u >> 16);
+      }
+      *reinterpret_cast<short8v*>(p) = out;
+    } else {
+      short4v out;
+#pragma unroll
+      for (int j = 0; j < 4; j += 2) {
+        unsigned int u = cvt_pk_bf16(v[j], v[j + 1]);
+        ((unsigned short*)&out)[j] = (unsigned short)u;
+        ((unsigned short*)&out)[j + 1] = (unsigned short)(u >> 16);
+      }
+      *reinterpret_cast<short4v*>(p) = out;
+    }
+  } else {
+#pragma unroll
+    for (int j = 0; j < VEC; j += 4) {
+      float4v out;
+#pragma unroll
+      for (int k = 0; k < 4; ++k) ((float*)&out)[k] = v[j + k];
+      *reinterpret_cast<float4v*>((float*)p + j) = out;
+    }
+  }
+}
+
 #define HIP_CHECK(x)                                                       \
   do {                                                                     \
     hipError_t _e = (x);                                                   \

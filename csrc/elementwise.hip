@@ -10,108 +10,85 @@ namespace {
 constexpr int BLOCK = 256;
 constexpr float GELU_K = 0.7978845608028654f;  // sqrt(2/pi)
 
+// tanh via the hardware exp unit: tanh(y) = 1 - 2/(exp(2y)+1).
+// libm tanhf lowers to a branchy polynomial (~20 VALU ops); this is 4,
+// exact to ~1 ulp fp32 for the gelu argument range.
+DEV_INLINE float fast_tanh(float y) {
+  return 1.f - 2.f / (__expf(2.f * y) + 1.f);
+}
 DEV_INLINE float gelu_tanh(float x) {
-  float t = tanhf(GELU_K * (x + 0.044715f * x * x * x));
+  float t = fast_tanh(GELU_K * (x + 0.044715f * x * x * x));
   return 0.5f * x * (1.f + t);
 }
 DEV_INLINE float gelu_tanh_grad(float x) {
-  float t = tanhf(GELU_K * (x + 0.044715f * x * x * x));
+  float t = fast_tanh(GELU_K * (x + 0.044715f * x * x * x));
   return 0.5f * (1.f + t) +
          0.5f * x * (1.f - t * t) * GELU_K * (1.f + 3.f * 0.044715f * x * x);
 }
 
 // ---- bias + gelu: x [N, H] + bias[H] ----
-template <typename T, bool HAS_BIAS>
+template <typename T, bool HAS_BIAS, int VEC>
 __global__ void bias_gelu_fwd_kernel(const T* __restrict__ x,
                                      const T* __restrict__ bias,
                                      T* __restrict__ y, long total, int H) {
-  const int VEC = 4;
   for (long i = ((long)blockIdx.x * BLOCK + threadIdx.x) * VEC; i < total;
        i += (long)gridDim.x * BLOCK * VEC) {
     float v[VEC];
-    if constexpr (sizeof(T) == 2) {
-      short4v pk = *reinterpret_cast<const short4v*>((const unsigned short*)x + i);
-#pragma unroll
-      for (int j = 0; j < VEC; ++j) v[j] = bf_raw2f(((unsigned short*)&pk)[j]);
-    } else {
-      float4v pk = *reinterpret_cast<const float4v*>((const float*)x + i);
-#pragma unroll
-      for (int j = 0; j < VEC; ++j) v[j] = ((float*)&pk)[j];
-    }
+    vload<T, VEC>(x + i, v);
     if constexpr (HAS_BIAS) {
-      int col = (int)(i % H);  // VEC divides H, so col..col+3 stay in row
-      if constexpr (sizeof(T) == 2) {
-        short4v bk = *reinterpret_cast<const short4v*>((const unsigned short*)bias + col);
+      int col = (int)(i % H);  // VEC divides H, so the span stays in-row
+      float bv[VEC];
+      vload<T, VEC>(bias + col, bv);
 #pragma unroll
-        for (int j = 0; j < VEC; ++j) v[j] += bf_raw2f(((unsigned short*)&bk)[j]);
-      } else {
-        float4v bk = *reinterpret_cast<const float4v*>((const float*)bias + col);
-#pragma unroll
-        for (int j = 0; j < VEC; ++j) v[j] += ((float*)&bk)[j];
-      }
+      for (int j = 0; j < VEC; ++j) v[j] += bv[j];
     }
-    if constexpr (sizeof(T) == 2) {
-      short4v out;
+    float o[VEC];
 #pragma unroll
-      for (int j = 0; j < VEC; ++j) ((unsigned short*)&out)[j] = f2bf_raw(gelu_tanh(v[j]));
-      *reinterpret_cast<short4v*>((unsigned short*)y + i) = out;
-    } else {
-      float4v out;
-#pragma unroll
-      for (int j = 0; j < VEC; ++j) ((float*)&out)[j] = gelu_tanh(v[j]);
-      *reinterpret_cast<float4v*>((float*)y + i) = out;
-    }
+    for (int j = 0; j < VEC; ++j) o[j] = gelu_tanh(v[j]);
+    vstore<T, VEC>(y + i, o);
   }
 }
 
-template <typename T, bool HAS_BIAS>
+template <typename T, bool HAS_BIAS, int VEC>
 __global__ void bias_gelu_bwd_kernel(const T* __restrict__ dy,
                                      const T* __restrict__ x,
                                      const T* __restrict__ bias,
                                      T* __restrict__ dx, long total, int H) {
-  const int VEC = 4;
   for (long i = ((long)blockIdx.x * BLOCK + threadIdx.x) * VEC; i < total;
        i += (long)gridDim.x * BLOCK * VEC) {
     float v[VEC], d[VEC];
-    if constexpr (sizeof(T) == 2) {
-      short4v px = *reinterpret_cast<const short4v*>((const unsigned short*)x + i);
-      short4v pd = *reinterpret_cast<const short4v*>((const unsigned short*)dy + i);
-#pragma unroll
-      for (int j = 0; j < VEC; ++j) {
-        v[j] = bf_raw2f(((unsigned short*)&px)[j]);
-        d[j] = bf_raw2f(((unsigned short*)&pd)[j]);
-      }
-    } else {
-      float4v px = *reinterpret_cast<const float4v*>((const float*)x + i);
-      float4v pd = *reinterpret_cast<const float4v*>((const float*)dy + i);
-#pragma unroll
-      for (int j = 0; j < VEC; ++j) { v[j] = ((float*)&px)[j]; d[j] = ((float*)&pd)[j]; }
-    }
+    vload<T, VEC>(x + i, v);
+    vload<T, VEC>(dy + i, d);
     if constexpr (HAS_BIAS) {
       int col = (int)(i % H);
-      if constexpr (sizeof(T) == 2) {
-        short4v bk = *reinterpret_cast<const short4v*>((const unsigned short*)bias + col);
+      float bv[VEC];
+      vload<T, VEC>(bias + col, bv);
 #pragma unroll
-        for (int j = 0; j < VEC; ++j) v[j] += bf_raw2f(((unsigned short*)&bk)[j]);
-      } else {
-        float4v bk = *reinterpret_cast<const float4v*>((const float*)bias + col);
-#pragma unroll
-        for (int j = 0; j < VEC; ++j) v[j] += ((float*)&bk)[j];
-      }
+      for (int j = 0; j < VEC; ++j) v[j] += bv[j];
     }
-    if constexpr (sizeof(T) == 2) {
-      short4v out;
+    float o[VEC];
 #pragma unroll
-      for (int j = 0; j < VEC; ++j)
-        ((unsigned short*)&out)[j] = f2bf_raw(d[j] * gelu_tanh_grad(v[j]));
-      *reinterpret_cast<short4v*>((unsigned short*)dx + i) = out;
-    } else {
-      float4v out;
-#pragma unroll
-      for (int j = 0; j < VEC; ++j) ((float*)&out)[j] = d[j] * gelu_tanh_grad(v[j]);
-      *reinterpret_cast<float4v*>((float*)dx + i) = out;
-    }
+    for (int j = 0; j < VEC; ++j) o[j] = d[j] * gelu_tanh_grad(v[j]);
+    vstore<T, VEC>(dx + i, o);
   }
+}
+
+// ---- column sum: x [N, H] -> fp32 [H] (dbias; chunked atomics) ----
+template <typename T>
+__global__ void colsum_kernel(const T* __restrict__ x, float* __restrict__ out,
+                              long N, int H, int rows_per_chunk) {
+  int col = blockIdx.x * BLOCK + threadIdx.x;
+  if (col >= H) return;
+  long r0 = (long)blockIdx.y * rows_per_chunk;
+  long r1 = r0 + rows_per_chunk < N ? r0 + rows_per_chunk : N;
+  float s = 0.f;
+  for (long r = r0; r < r1; ++r) {
+    if constexpr (sizeof(T) == 2)
+      s += bf_raw2f(((const unsigned short*)x)[r * H + col]);
+    else
+      s += ((const float*)x)[r * H + col];
+  }
+  atomicAdd(&out[col], s);
 }
 
 // ---- AdamW over flat fp32 state, low-precision model copy + grads ----
@@ -218,19 +195,22 @@ torch::Tensor bias_gelu_fwd(torch::Tensor x, torch::Tensor bias) {
   if (has_bias) TORCH_CHECK(H % 4 == 0);
   auto y = torch::empty_like(x);
   auto stream = at::hip::getCurrentHIPStream();
-  dim3 grid(grid_for(total, BLOCK * 4));
-#define LAUNCH_BG(T, HB)                                                     \
-  hipLaunchKernelGGL((bias_gelu_fwd_kernel<T, HB>), grid, dim3(BLOCK), 0,    \
+  bool v8 = (total % 8 == 0) && (!has_bias || H % 8 == 0);
+  dim3 grid(grid_for(total, BLOCK * (v8 ? 8 : 4)));
+#define LAUNCH_BG(T, HB, V)                                                  \
+  hipLaunchKernelGGL((bias_gelu_fwd_kernel<T, HB, V>), grid, dim3(BLOCK), 0, \
                      stream, (const T*)x.data_ptr(),                         \
                      has_bias ? (const T*)bias.data_ptr() : nullptr,         \
                      (T*)y.data_ptr(), total, H)
+#define PICK_BG(T, HB) do { if (v8) LAUNCH_BG(T, HB, 8); else LAUNCH_BG(T, HB, 4); } while (0)
   if (x.scalar_type() == torch::kBFloat16) {
-    if (has_bias) LAUNCH_BG(__hip_bfloat16, true);
-    else LAUNCH_BG(__hip_bfloat16, false);
+    if (has_bias) PICK_BG(__hip_bfloat16, true);
+    else PICK_BG(__hip_bfloat16, false);
   } else {
-    if (has_bias) LAUNCH_BG(float, true);
-    else LAUNCH_BG(float, false);
+    if (has_bias) PICK_BG(float, true);
+    else PICK_BG(float, false);
   }
+#undef PICK_BG
 #undef LAUNCH_BG
   return y;
 }
@@ -243,23 +223,63 @@ std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
   bool has_bias = bias.defined() && bias.numel() > 0;
   auto dx = torch::empty_like(x);
   auto stream = at::hip::getCurrentHIPStream();
-  dim3 grid(grid_for(total, BLOCK * 4));
-#define LAUNCH_BG(T, HB)                                                     \
-  hipLaunchKernelGGL((bias_gelu_bwd_kernel<T, HB>), grid, dim3(BLOCK), 0,    \
+  bool v8 = (total % 8 == 0) && (!has_bias || H % 8 == 0);
+  dim3 grid(grid_for(total, BLOCK * (v8 ? 8 : 4)));
+#define LAUNCH_BG(T, HB, V)                                                  \
+  hipLaunchKernelGGL((bias_gelu_bwd_kernel<T, HB, V>), grid, dim3(BLOCK), 0, \
                      stream, (const T*)dy.data_ptr(), (const T*)x.data_ptr(), \
                      has_bias ? (const T*)bias.data_ptr() : nullptr,         \
                      (T*)dx.data_ptr(), total, H)
+#define PICK_BG(T, HB) do { if (v8) LAUNCH_BG(T, HB, 8); else LAUNCH_BG(T, HB, 4); } while (0)
   if (x.scalar_type() == torch::kBFloat16) {
-    if (has_bias) LAUNCH_BG(__hip_bfloat16, true);
-    else LAUNCH_BG(__hip_bfloat16, false);
+    if (has_bias) PICK_BG(__hip_bfloat16, true);
+    else PICK_BG(__hip_bfloat16, false);
   } else {
-    if (has_bias) LAUNCH_BG(float, true);
-    else LAUNCH_BG(float, false);
+    if (has_bias) PICK_BG(float, true);
+    else PICK_BG(float, false);
   }
+#undef PICK_BG
 #undef LAUNCH_BG
   torch::Tensor db;
-  if (has_bias) db = dx.view({-1, H}).to(torch::kFloat).sum(0);
+  if (has_bias) {
+    // fused column sum (fp32 accumulate) — replaces the materialized
+    // .to(float).sum(0), which cost two extra full passes over dx
+    long N = total / H;
+    db = torch::zeros({H}, x.options().dtype(torch::kFloat));
+    int rows_per_chunk = std::max<long>(16, N / 256);
+    int nchunks = (int)((N + rows_per_chunk - 1) / rows_per_chunk);
+    dim3 grid2((H + BLOCK - 1) / BLOCK, nchunks);
+    if (x.scalar_type() == torch::kBFloat16)
+      hipLaunchKernelGGL((colsum_kernel<__hip_bfloat16>), grid2, dim3(BLOCK),
+                         0, stream, (const __hip_bfloat16*)dx.data_ptr(),
+                         db.data_ptr<float>(), N, H, rows_per_chunk);
+    else
+      hipLaunchKernelGGL((colsum_kernel<float>), grid2, dim3(BLOCK), 0,
+                         stream, (const float*)dx.data_ptr(),
+                         db.data_ptr<float>(), N, H, rows_per_chunk);
+  }
   return {dx, db};
+}
+
+// standalone column sum for linear dbias: x [N, H] -> fp32 [H]
+torch::Tensor colsum(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous());
+  long N = x.size(0);
+  int H = x.size(1);
+  auto out = torch::zeros({H}, x.options().dtype(torch::kFloat));
+  auto stream = at::hip::getCurrentHIPStream();
+  int rows_per_chunk = std::max<long>(16, N / 256);
+  int nchunks = (int)((N + rows_per_chunk - 1) / rows_per_chunk);
+  dim3 grid2((H + BLOCK - 1) / BLOCK, nchunks);
+  if (x.scalar_type() == torch::kBFloat16)
+    hipLaunchKernelGGL((colsum_kernel<__hip_bfloat16>), grid2, dim3(BLOCK), 0,
+                       stream, (const __hip_bfloat16*)x.data_ptr(),
+                       out.data_ptr<float>(), N, H, rows_per_chunk);
+  else
+    hipLaunchKernelGGL((colsum_kernel<float>), grid2, dim3(BLOCK), 0, stream,
+                       (const float*)x.data_ptr(), out.data_ptr<float>(), N, H,
+                       rows_per_chunk);
+  return out;
 }
 
 void adamw_flat(torch::Tensor master, torch::Tensor grad, torch::Tensor m,

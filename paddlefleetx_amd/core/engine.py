@@ -133,6 +133,13 @@ class EagerEngine(BasicEngine):
                     and not self.is_pipeline:
                 self._overlap_reduce = self.optimizer.enable_overlap(
                     self.hcg.get_data_parallel_group())
+            # arm in-GEMM weight-grad accumulation (ops/linear.py): grads
+            # land in the bucket views inside the wgrad GEMM (beta=1),
+            # skipping autograd's per-micro dW materialize + add
+            if isinstance(self.optimizer, FusedAdamW) \
+                    and not self.optimizer._fp32_main_grad:
+                from paddlefleetx_amd.ops.linear import set_wgrad_fusion
+                set_wgrad_fusion(True)
         else:
             self.optimizer = None
             self._overlap_reduce = False

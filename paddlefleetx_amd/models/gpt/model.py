@@ -238,7 +238,9 @@ class TransformerDecoderLayer(nn.Module):
             return F.dropout(x, p=self.hidden_dropout_p)
         return x
 
-    def _attn_block(self, x, cache=None, use_cache=False):
+    def _attn_branch(self, x, cache=None, use_cache=False):
+        """dropout(attn(ln1(x))) — the branch WITHOUT the residual add
+        (the add is fused into ln2, see forward)."""
         h = self.ln1(x)
         if self.use_recompute and self.recompute_granularity == "core_attn" \
                 and self.training and cache is None and not use_cache \
@@ -247,21 +249,26 @@ class TransformerDecoderLayer(nn.Module):
             # reference granularity "core_attn" (hybrid_model.py:303-346)
             a = checkpoint(lambda t: self.attn(t)[0], h,
                            use_reentrant=False)
-            return x + self._dropout(a), None
+            return self._dropout(a), None
         a, new_cache = self.attn(h, cache=cache, use_cache=use_cache)
-        return x + self._dropout(a), new_cache
+        return self._dropout(a), new_cache
 
-    def _attn_block_nocache(self, x):
-        return self._attn_block(x)[0]
+    def _attn_branch_nocache(self, x):
+        return self._attn_branch(x)[0]
 
     def forward(self, x, cache: Optional[KVCache] = None, use_cache: bool = False):
         if self.use_recompute and self.recompute_granularity == "full_attn" \
                 and self.training and not use_cache and torch.is_grad_enabled():
-            y = checkpoint(self._attn_block_nocache, x, use_reentrant=False)
+            a = checkpoint(self._attn_branch_nocache, x, use_reentrant=False)
             new_cache = None
         else:
-            y, new_cache = self._attn_block(x, cache, use_cache)
-        z = y + self._dropout(self.ffn(self.ln2(y)))
+            a, new_cache = self._attn_branch(x, cache, use_cache)
+        # fused: h2 = LN(x + a), y = x + a (one kernel; the backward joins
+        # the residual gradient in the same pass)
+        from paddlefleetx_amd.ops import layernorm_residual
+        h2, y = layernorm_residual(a, x, self.ln2.weight, self.ln2.bias,
+                                   self.ln2.eps)
+        z = y + self._dropout(self.ffn(h2))
         if use_cache:
             return z, new_cache
         return z

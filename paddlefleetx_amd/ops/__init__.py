@@ -60,6 +60,7 @@ from paddlefleetx_amd.ops.functional import (  # noqa: E402,F401
     fused_adamw_flat,
     fused_softmax_causal,
     layernorm,
+    layernorm_residual,
     rmsnorm,
     rope,
     topp_sampling,

@@ -41,6 +41,25 @@ def layernorm_bwd(dy: torch.Tensor, x: torch.Tensor, weight: torch.Tensor,
     return dx.to(x.dtype), dw, db
 
 
+def layernorm_fwd_residual(x: torch.Tensor, res: torch.Tensor,
+                           weight: torch.Tensor, bias: torch.Tensor,
+                           eps: float):
+    """LN(x + res); returns (y, mean, rstd, sum) with sum = x + res
+    in x.dtype (the fused residual stream)."""
+    s = (x.float() + res.float()).to(x.dtype)
+    y, mean, rstd = layernorm_fwd(s, weight, bias, eps)
+    return y, mean, rstd, s
+
+
+def layernorm_bwd_residual(dy: torch.Tensor, s: torch.Tensor,
+                           weight: torch.Tensor, mean: torch.Tensor,
+                           rstd: torch.Tensor, dsum: torch.Tensor):
+    """LN bwd with the residual-stream gradient fused into dx."""
+    dx, dw, db = layernorm_bwd(dy, s, weight, mean, rstd)
+    dx = (dx.float() + dsum.float()).to(dx.dtype)
+    return dx, dw, db
+
+
 def rmsnorm_fwd(x: torch.Tensor, weight: torch.Tensor, eps: float
                 ) -> Tuple[torch.Tensor, torch.Tensor]:
     xf = x.float()

@@ -21,7 +21,8 @@ from paddlefleetx_amd.ops import _reference as ref
 from paddlefleetx_amd.ops import hip_ext, use_hip
 
 __all__ = [
-    "layernorm", "rmsnorm", "FusedLayerNorm", "FusedRMSNorm", "bias_gelu",
+    "layernorm", "layernorm_residual", "rmsnorm", "FusedLayerNorm",
+    "FusedRMSNorm", "bias_gelu",
     "flash_attention", "flash_attention_packed", "fused_softmax_causal", "cross_entropy",
     "fused_adamw_flat", "rope", "topp_sampling",
 ]
@@ -56,6 +57,54 @@ class _LayerNormFn(torch.autograd.Function):
 
 def layernorm(x, weight, bias, eps: float = 1e-5):
     return _LayerNormFn.apply(x, weight, bias, eps)
+
+
+class _LayerNormResidualFn(torch.autograd.Function):
+    """y = LN(x + res); also returns s = x + res for the residual stream.
+    One kernel does the add + stats + normalize (fwd) and the residual
+    gradient join (bwd) — removes two full-tensor elementwise passes per
+    call vs separate add + LN."""
+
+    @staticmethod
+    def forward(ctx, x, res, weight, bias, eps):
+        shape = x.shape
+        x2 = x.contiguous().view(-1, shape[-1])
+        r2 = res.contiguous().view(-1, shape[-1])
+        if use_hip(x2):
+            y, mean, rstd, s = hip_ext().layernorm_fwd_residual(
+                x2, r2, weight, bias, eps)
+        else:
+            y, mean, rstd, s = ref.layernorm_fwd_residual(
+                x2, r2, weight, bias, eps)
+        ctx.save_for_backward(s, weight, mean, rstd)
+        return y.view(shape), s.view(shape)
+
+    @staticmethod
+    def backward(ctx, dy, ds):
+        s2, weight, mean, rstd = ctx.saved_tensors
+        dy2 = dy.contiguous().view(-1, dy.shape[-1])
+        if ds is not None:
+            ds2 = ds.contiguous().view(-1, dy.shape[-1])
+            if use_hip(s2):
+                dx, dw, db = hip_ext().layernorm_bwd_residual(
+                    dy2, s2, weight, mean, rstd, ds2)
+            else:
+                dx, dw, db = ref.layernorm_bwd_residual(
+                    dy2, s2, weight, mean, rstd, ds2)
+        else:
+            if use_hip(s2):
+                dx, dw, db = hip_ext().layernorm_bwd(dy2, s2, weight, mean,
+                                                     rstd)
+            else:
+                dx, dw, db = ref.layernorm_bwd(dy2, s2, weight, mean, rstd)
+        dxv = dx.view(dy.shape)
+        # d(x) = d(res) = dx (s = x + res is linear in both)
+        return dxv, dxv, dw.to(weight.dtype), db.to(weight.dtype), None
+
+
+def layernorm_residual(x, res, weight, bias, eps: float = 1e-5):
+    """Fused (LN(x+res), x+res)."""
+    return _LayerNormResidualFn.apply(x, res, weight, bias, eps)
 
 
 class _RMSNormFn(torch.autograd.Function):

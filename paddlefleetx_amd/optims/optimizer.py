@@ -192,23 +192,28 @@ class FusedAdamW(torch.optim.Optimizer):
             for p in b.params:
                 self._param_bucket[id(p)] = bi
 
-        def make_hook():
-            def hook(p):
-                if not self._ov_active:
-                    return
-                bi = self._param_bucket[id(p)]
-                self._bucket_pending[bi] -= 1
-                if self._bucket_pending[bi] == 0:
-                    h = dist.all_reduce(self.buckets[bi].grad_flat,
-                                        group=self._ov_group, async_op=True)
-                    self._ov_handles.append(h)
-            return hook
-
         for b in self.buckets:
             for p in b.params:
                 self._hooks.append(
-                    p.register_post_accumulate_grad_hook(make_hook()))
+                    p.register_post_accumulate_grad_hook(
+                        self._param_grad_ready))
         return True
+
+    def _param_grad_ready(self, p):
+        """One param's grad for the armed backward is complete — fired by
+        the post-accumulate hook (plain autograd params) OR by the fused
+        wgrad linear's notify (ops/linear.py, where AccumulateGrad never
+        runs). Issues the bucket's async allreduce when all arrive."""
+        if not getattr(self, "_ov_active", False):
+            return
+        bi = self._param_bucket.get(id(p))
+        if bi is None:
+            return
+        self._bucket_pending[bi] -= 1
+        if self._bucket_pending[bi] == 0:
+            h = dist.all_reduce(self.buckets[bi].grad_flat,
+                                group=self._ov_group, async_op=True)
+            self._ov_handles.append(h)
 
     def begin_overlap_reduce(self):
         """Arm the hooks for the LAST micro-batch's backward."""

@@ -18,6 +18,8 @@ import torch.distributed as dist
 import torch.nn as nn
 import torch.nn.functional as F
 
+from paddlefleetx_amd.ops.linear import fused_bias_add, fused_linear
+
 from paddlefleetx_amd.parallel.env import get_hcg
 from paddlefleetx_amd.parallel.rng import model_parallel_rng
 
@@ -163,7 +165,7 @@ class ColumnSequenceParallelLinear(nn.Module):
     def forward(self, x):
         # x: [s/mp, b, in] -> [s, b, out/mp]
         x = AllGatherOp.apply(x)
-        return F.linear(x, self.weight, self.bias)
+        return fused_linear(x, self.weight, self.bias)
 
 
 class RowSequenceParallelLinear(nn.Module):
@@ -189,10 +191,10 @@ class RowSequenceParallelLinear(nn.Module):
 
     def forward(self, x):
         # x: [s, b, in/mp] -> [s/mp, b, out]
-        y = F.linear(x, self.weight, None)
+        y = fused_linear(x, self.weight, None)
         y = ReduceScatterOp.apply(y)
         if self.bias is not None:
-            y = y + self.bias
+            y = fused_bias_add(y, self.bias)
         return y
 
 

@@ -18,6 +18,8 @@ import torch.distributed as dist
 import torch.nn as nn
 import torch.nn.functional as F
 
+from paddlefleetx_amd.ops.linear import fused_bias_add, fused_linear
+
 from paddlefleetx_amd.parallel.env import get_hcg
 from paddlefleetx_amd.parallel.rng import model_parallel_rng
 
@@ -132,7 +134,7 @@ class ColumnParallelLinear(nn.Module):
 
     def forward(self, x):
         x = copy_to_mp_region(x)
-        y = F.linear(x, self.weight, self.bias)
+        y = fused_linear(x, self.weight, self.bias)
         if self.gather_output:
             y = gather_from_mp_region(y)
         return y
@@ -170,10 +172,10 @@ class RowParallelLinear(nn.Module):
             g = _mp_group()
             chunk = x.shape[-1] // g.world_size
             x = x[..., g.rank * chunk:(g.rank + 1) * chunk]
-        y = F.linear(x, self.weight, None)
+        y = fused_linear(x, self.weight, None)
         y = reduce_from_mp_region(y)
         if self.bias is not None:
-            y = y + self.bias
+            y = fused_bias_add(y, self.bias)
         return y
 
 

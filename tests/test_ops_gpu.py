@@ -51,6 +51,51 @@ def test_layernorm_fwd_bwd(dev, shape):
     assert torch.allclose(db, db2, atol=0.5, rtol=1e-2)
 
 
+@pytest.mark.parametrize("shape", [(512, 4096), (128, 1024), (64, 3072)])
+def test_layernorm_residual_fused(dev, shape):
+    """LN(a+b) fused kernel (also emits a+b) vs reference; bwd with the
+    fused residual-grad addend."""
+    N, H = shape
+    a = torch.randn(N, H, device=dev, dtype=torch.bfloat16)
+    r = torch.randn(N, H, device=dev, dtype=torch.bfloat16)
+    w = torch.randn(H, device=dev, dtype=torch.bfloat16)
+    b = torch.randn(H, device=dev, dtype=torch.bfloat16)
+    y, mean, rstd, s = hip_ext().layernorm_fwd_residual(a, r, w, b, 1e-5)
+    y2, mean2, rstd2, s2 = ref.layernorm_fwd_residual(a, r, w, b, 1e-5)
+    assert torch.allclose(s.float(), s2.float(), atol=2e-2, rtol=2e-2)
+    assert torch.allclose(mean, mean2, atol=1e-3)
+    assert torch.allclose(y.float(), y2.float(), atol=3e-2, rtol=3e-2)
+
+    dy = torch.randn_like(a)
+    ds = torch.randn_like(a)
+    dx, dw, db = hip_ext().layernorm_bwd_residual(dy, s, w, mean, rstd, ds)
+    dx2, dw2, db2 = ref.layernorm_bwd_residual(dy, s2, w, mean2, rstd2, ds)
+    assert torch.allclose(dx.float(), dx2.float(), atol=5e-2, rtol=5e-2)
+    assert torch.allclose(dw, dw2, atol=0.5, rtol=1e-2)
+    assert torch.allclose(db, db2, atol=0.5, rtol=1e-2)
+
+
+def test_colsum(dev):
+    x = torch.randn(8192, 4096, device=dev, dtype=torch.bfloat16)
+    out = hip_ext().colsum(x)
+    want = x.float().sum(0)
+    assert torch.allclose(out, want, atol=0.5, rtol=1e-2)
+
+
+def test_bias_gelu_vec8_wide(dev):
+    """FFN-shaped (H%8==0) input exercises the VEC=8 path."""
+    x = torch.randn(1024, 2048, device=dev, dtype=torch.bfloat16)
+    b = torch.randn(2048, device=dev, dtype=torch.bfloat16)
+    y = hip_ext().bias_gelu_fwd(x, b)
+    y2 = ref.bias_gelu_fwd(x, b)
+    assert torch.allclose(y.float(), y2.float(), atol=2e-2, rtol=2e-2)
+    dy = torch.randn_like(x)
+    dx, db = hip_ext().bias_gelu_bwd(dy, x, b)
+    dx2, db2 = ref.bias_gelu_bwd(dy, x, b)
+    assert torch.allclose(dx.float(), dx2.float(), atol=3e-2, rtol=3e-2)
+    assert torch.allclose(db.float(), db2.float(), atol=0.5, rtol=1e-2)
+
+
 def test_rmsnorm(dev):
     x = torch.randn(256, 2048, device=dev, dtype=torch.bfloat16)
     w = torch.randn(2048, device=dev, dtype=torch.bfloat16)
