@@ -461,24 +461,32 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
 // Backward: delta = rowsum(dO * O), strided inputs
 // ===========================================================================
 __global__ void attn_bwd_delta_kernel(Strided dout, Strided o,
-                                      float* __restrict__ delta, int H, int S,
-                                      int D) {
-  long row = (long)blockIdx.x * (BLOCKT / WAVE) + threadIdx.x / WAVE;
-  long bhs = row;
-  int s = (int)(bhs % S);
-  long bh = bhs / S;
+                                      float* __restrict__ delta, long rows,
+                                      int H, int S, int D) {
+  // one row per 16-lane group (4 b64 loads per row at D=128): wider
+  // loads + 4 rows in flight per wave instead of one latency-bound row
+  constexpr int GROUPS = BLOCKT / 16;
+  long row = (long)blockIdx.x * GROUPS + threadIdx.x / 16;
+  if (row >= rows) return;
+  int s = (int)(row % S);
+  long bh = row / S;
   int hh = (int)(bh % H);
   int b = (int)(bh / H);
-  if (s >= S) return;
-  int lane = threadIdx.x % WAVE;
+  int lane16 = threadIdx.x % 16;
   const unsigned short* dp =
       (const unsigned short*)dout.at(b, hh) + (long)s * dout.rs;
   const unsigned short* op = (const unsigned short*)o.at(b, hh) + (long)s * o.rs;
   float acc = 0.f;
-  for (int i = lane; i < D; i += WAVE)
-    acc += bf_raw2f(dp[i]) * bf_raw2f(op[i]);
-  acc = wave_reduce_sum(acc);
-  if (lane == 0) delta[bh * (long)S + s] = acc;
+  for (int i = lane16 * 4; i < D; i += 16 * 4) {
+    short4v dv = *reinterpret_cast<const short4v*>(dp + i);
+    short4v ov = *reinterpret_cast<const short4v*>(op + i);
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      acc += bf_raw2f(((unsigned short*)&dv)[j]) *
+             bf_raw2f(((unsigned short*)&ov)[j]);
+  }
+  acc = group16_reduce_sum(acc);
+  if (lane16 == 0) delta[bh * (long)S + s] = acc;
 }
 
 // ===========================================================================
@@ -1011,8 +1019,8 @@ void launch_bwd(Strided q, Strided k, Strided v, Strided dout, Strided o,
   auto stream = at::hip::getCurrentHIPStream();
   long rows = (long)B * H * S;
   hipLaunchKernelGGL(attn_bwd_delta_kernel,
-                     dim3((rows + (BLOCKT / WAVE) - 1) / (BLOCKT / WAVE)),
-                     dim3(BLOCKT), 0, stream, dout, o, delta, H, S, D);
+                     dim3((rows + (BLOCKT / 16) - 1) / (BLOCKT / 16)),
+                     dim3(BLOCKT), 0, stream, dout, o, delta, rows, H, S, D);
   int tiles64 = (S + TILE - 1) / TILE;       // inner streamed tiles
   int blocks128 = (S + QTILE - 1) / QTILE;   // dkv per-block home tile
   int blocks256 = (S + QTILE * 2 - 1) / (QTILE * 2);  // dq RB=2 home tiles
