@@ -170,17 +170,17 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
   // while avoiding the 0-mod-128B d-stride that made the transpose writes
   // 16-way bank conflicted (measured SQ_LDS_BANK_CONFLICT 15% of cycles)
   constexpr int VRS = TILE + 4;
-  constexpr int PRS = TILE + PAD;       // per-wave P row stride
   constexpr int NDT = D / 16;
   constexpr int KSZ = TILE * KRS;       // one K buffer (shorts)
   constexpr int VSZ = D * VRS;          // one V^T buffer
 
-  // single __shared__ object (guide §5 trap 4a); P buffer holds BOTH
-  // 16-row groups of the wave (RB=2) so PV can share each V read
-  __shared__ unsigned short smem[2 * KSZ + 2 * VSZ + FW_WAVES * 32 * PRS];
+  // single __shared__ object (guide §5 trap 4a). P never touches LDS:
+  // the swapped-QK^T score registers are redistributed into PV A-fragments
+  // with permlane16/32 swaps (guide T12) — smem is 69.6 KB at D=128, so
+  // TWO blocks fit per CU (4 waves/SIMD) instead of one
+  __shared__ unsigned short smem[2 * KSZ + 2 * VSZ];
   unsigned short* k_lds = smem;                  // [2][KSZ]
   unsigned short* vt_lds = smem + 2 * KSZ;       // [2][VSZ]
-  unsigned short* p_lds = smem + 2 * KSZ + 2 * VSZ;
 
   // bh on blockIdx.x: the linear dispatch then gives every CU one block
   // of each qt, balancing the causal tile-count imbalance (B*H >= 256
@@ -220,7 +220,6 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
   const int kv_tiles = CAUSAL
       ? min(kv_total, (qt * QTILE * RB + QTILE * RB - 1) / TILE + 1)
       : kv_total;
-  unsigned short* myp = p_lds + wid * 32 * PRS;
 
   // ---- staging thread maps ----
   // K (row-major image): D/16 threads per row, 16 shorts each
@@ -305,6 +304,7 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
       // per row. A- and B-fragments share a register layout, so the
       // operand swap is just argument order. ----
       f4 s[RB][4];
+      bf8 pfrag[RB][2];  // PV A-fragments, assembled in registers (T12)
       bf8 kbuf[3];
       kbuf[0] = read_b_frag(kb_lds, KRS, 0, 0, lane);
       kbuf[1] = read_b_frag(kb_lds, KRS, 0, 32, lane);
@@ -380,28 +380,63 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
         psum += __shfl_xor(psum, 32, WAVE);
         l_r[rb][0] += psum;
 
-        // ---- P^T -> per-wave LDS [q][kv] image: each lane writes its
-        // own q row; the 4 r-values are CONSECUTIVE kv columns, so two
-        // cvt_pk + two b32 stores cover a ct tile ----
-        unsigned short* prow = myp + (rb * 16 + ccol) * PRS + crow4;
+        // ---- in-register P hand-off (guide T12): this lane holds ONE
+        // q row's 16 P values as 8 packed u32 chunks C[ct][h]
+        // (kv = 4g + 16·ct + {2h, 2h+1}, g = lane>>4). The PV A-fragment
+        // needs kv = 8g+e (+32·kc) on this lane instead — a fixed
+        // redistribution among the 4 lanes sharing this q row (l, l±16,
+        // l±32). Phase A (permlane16_swap on a duplicated reg) leaves
+        // e16 = the even-16-group's chunk and o16 = the odd group's on
+        // every lane of each 16-pair; phase B (permlane16/32_swap of
+        // e16[c_even] against e16[c_odd]) lands exactly the fragment's
+        // low/high chunks split by even/odd 16-group, so ONE cndmask per
+        // u32 finishes the job. Replaces 8 ds_write_b32 + 4 ds_read_b64
+        // per row group AND frees the 36.9 KB P image -> 2 blocks/CU.
+        unsigned int own[4][2];
 #pragma unroll
         for (int ct = 0; ct < 4; ++ct) {
-          unsigned int lo = cvt_pk_bf16(s[rb][ct][0], s[rb][ct][1]);
-          unsigned int hi = cvt_pk_bf16(s[rb][ct][2], s[rb][ct][3]);
-          *reinterpret_cast<unsigned int*>(prow + ct * 16) = lo;
-          *reinterpret_cast<unsigned int*>(prow + ct * 16 + 2) = hi;
+          own[ct][0] = cvt_pk_bf16(s[rb][ct][0], s[rb][ct][1]);
+          own[ct][1] = cvt_pk_bf16(s[rb][ct][2], s[rb][ct][3]);
+        }
+        unsigned int e16[4][2], o16[4][2];
+#pragma unroll
+        for (int ct = 0; ct < 4; ++ct)
+#pragma unroll
+          for (int h = 0; h < 2; ++h) {
+            auto pr = __builtin_amdgcn_permlane16_swap(own[ct][h],
+                                                       own[ct][h],
+                                                       false, false);
+            e16[ct][h] = pr[0];  // even 16-group's chunk (both lanes)
+            o16[ct][h] = pr[1];  // odd 16-group's chunk
+          }
+        const bool even16 = ((lane >> 4) & 1) == 0;
+#pragma unroll
+        for (int kc = 0; kc < 2; ++kc) {
+          const int ce = 2 * kc, co = 2 * kc + 1;
+          unsigned int fr[4];
+#pragma unroll
+          for (int h = 0; h < 2; ++h) {
+            auto pe = __builtin_amdgcn_permlane32_swap(e16[ce][h],
+                                                       e16[co][h],
+                                                       false, false);
+            auto po = __builtin_amdgcn_permlane32_swap(o16[ce][h],
+                                                       o16[co][h],
+                                                       false, false);
+            // pe[0]: lower lanes (0,ce) | upper (0,co); pe[1]: lower
+            // (2,ce) | upper (2,co) -> even 16-group takes pe[0]
+            fr[h] = even16 ? pe[0] : pe[1];
+            fr[2 + h] = even16 ? po[0] : po[1];
+          }
+          union { unsigned int u[4]; bf8 b; } pu;
+#pragma unroll
+          for (int j = 0; j < 4; ++j) pu.u[j] = fr[j];
+          pfrag[rb][kc] = pu.b;
         }
       }
 
       // ---- PV: both row groups share each V B-fragment read, 2-deep
       // prefetch over the (kc,dt) stream ----
-      bf8 pa[RB][2];
-#pragma unroll
-      for (int rb = 0; rb < RB; ++rb)
-#pragma unroll
-        for (int kc = 0; kc < 2; ++kc)
-          pa[rb][kc] = read_a_frag_lds(myp + rb * 16 * PRS, PRS,
-                                       kc * 32, lane);
+      const bf8(&pa)[RB][2] = pfrag;
       bf8 vbuf[3];
       vbuf[0] = read_b_frag(vb_lds, VRS, 0, 0, lane);
       vbuf[1] = read_b_frag(vb_lds, VRS, 16, 0, lane);

@@ -27,68 +27,88 @@ DEV_INLINE float gelu_tanh_grad(float x) {
 }
 
 // ---- bias + gelu: x [N, H] + bias[H] ----
+// 2D grid: blockIdx.x covers a column span (the bias chunk loads ONCE per
+// block), blockIdx.y strides rows — no 64-bit div/mod per element.
 template <typename T, bool HAS_BIAS, int VEC>
-__global__ void bias_gelu_fwd_kernel(const T* __restrict__ x,
-                                     const T* __restrict__ bias,
-                                     T* __restrict__ y, long total, int H) {
-  for (long i = ((long)blockIdx.x * BLOCK + threadIdx.x) * VEC; i < total;
-       i += (long)gridDim.x * BLOCK * VEC) {
+__global__ __launch_bounds__(BLOCK) void bias_gelu_fwd_kernel(
+    const T* __restrict__ x, const T* __restrict__ bias, T* __restrict__ y,
+    long N, int H) {
+  const int col = (blockIdx.x * BLOCK + threadIdx.x) * VEC;
+  if (col >= H) return;
+  float bv[VEC];
+  if constexpr (HAS_BIAS) vload<T, VEC>(bias + col, bv);
+  for (long r = blockIdx.y; r < N; r += gridDim.y) {
+    const long i = r * H + col;
     float v[VEC];
     vload<T, VEC>(x + i, v);
-    if constexpr (HAS_BIAS) {
-      int col = (int)(i % H);  // VEC divides H, so the span stays in-row
-      float bv[VEC];
-      vload<T, VEC>(bias + col, bv);
-#pragma unroll
-      for (int j = 0; j < VEC; ++j) v[j] += bv[j];
-    }
     float o[VEC];
 #pragma unroll
-    for (int j = 0; j < VEC; ++j) o[j] = gelu_tanh(v[j]);
+    for (int j = 0; j < VEC; ++j) {
+      if constexpr (HAS_BIAS) v[j] += bv[j];
+      o[j] = gelu_tanh(v[j]);
+    }
     vstore<T, VEC>(y + i, o);
   }
 }
 
 template <typename T, bool HAS_BIAS, int VEC>
-__global__ void bias_gelu_bwd_kernel(const T* __restrict__ dy,
-                                     const T* __restrict__ x,
-                                     const T* __restrict__ bias,
-                                     T* __restrict__ dx, long total, int H) {
-  for (long i = ((long)blockIdx.x * BLOCK + threadIdx.x) * VEC; i < total;
-       i += (long)gridDim.x * BLOCK * VEC) {
+__global__ __launch_bounds__(BLOCK) void bias_gelu_bwd_kernel(
+    const T* __restrict__ dy, const T* __restrict__ x,
+    const T* __restrict__ bias, T* __restrict__ dx, long N, int H) {
+  const int col = (blockIdx.x * BLOCK + threadIdx.x) * VEC;
+  if (col >= H) return;
+  float bv[VEC];
+  if constexpr (HAS_BIAS) vload<T, VEC>(bias + col, bv);
+  for (long r = blockIdx.y; r < N; r += gridDim.y) {
+    const long i = r * H + col;
     float v[VEC], d[VEC];
     vload<T, VEC>(x + i, v);
     vload<T, VEC>(dy + i, d);
-    if constexpr (HAS_BIAS) {
-      int col = (int)(i % H);
-      float bv[VEC];
-      vload<T, VEC>(bias + col, bv);
-#pragma unroll
-      for (int j = 0; j < VEC; ++j) v[j] += bv[j];
-    }
     float o[VEC];
 #pragma unroll
-    for (int j = 0; j < VEC; ++j) o[j] = d[j] * gelu_tanh_grad(v[j]);
+    for (int j = 0; j < VEC; ++j) {
+      if constexpr (HAS_BIAS) v[j] += bv[j];
+      o[j] = d[j] * gelu_tanh_grad(v[j]);
+    }
     vstore<T, VEC>(dx + i, o);
   }
 }
 
 // ---- column sum: x [N, H] -> fp32 [H] (dbias; chunked atomics) ----
-template <typename T>
-__global__ void colsum_kernel(const T* __restrict__ x, float* __restrict__ out,
-                              long N, int H, int rows_per_chunk) {
-  int col = blockIdx.x * BLOCK + threadIdx.x;
+template <typename T, int CV>
+__global__ __launch_bounds__(BLOCK) void colsum_kernel(
+    const T* __restrict__ x, float* __restrict__ out, long N, int H,
+    int rows_per_chunk) {
+  const int col = (blockIdx.x * BLOCK + threadIdx.x) * CV;
   if (col >= H) return;
   long r0 = (long)blockIdx.y * rows_per_chunk;
   long r1 = r0 + rows_per_chunk < N ? r0 + rows_per_chunk : N;
-  float s = 0.f;
+  float s[CV];
+#pragma unroll
+  for (int j = 0; j < CV; ++j) s[j] = 0.f;
   for (long r = r0; r < r1; ++r) {
-    if constexpr (sizeof(T) == 2)
-      s += bf_raw2f(((const unsigned short*)x)[r * H + col]);
-    else
-      s += ((const float*)x)[r * H + col];
+    float v[CV];
+    vload<T, CV>(x + r * H + col, v);
+#pragma unroll
+    for (int j = 0; j < CV; ++j) s[j] += v[j];
   }
-  atomicAdd(&out[col], s);
+#pragma unroll
+  for (int j = 0; j < CV; ++j) atomicAdd(&out[col + j], s[j]);
+}
+
+void launch_colsum(const void* x, float* out, long N, int H, bool bf16,
+                   hipStream_t stream) {
+  constexpr int CV = 4;
+  int rows_per_chunk = (int)std::max<long>(32, N / 128);
+  int nchunks = (int)((N + rows_per_chunk - 1) / rows_per_chunk);
+  dim3 grid((H + BLOCK * CV - 1) / (BLOCK * CV), nchunks);
+  if (bf16)
+    hipLaunchKernelGGL((colsum_kernel<__hip_bfloat16, CV>), grid, dim3(BLOCK),
+                       0, stream, (const __hip_bfloat16*)x, out, N, H,
+                       rows_per_chunk);
+  else
+    hipLaunchKernelGGL((colsum_kernel<float, CV>), grid, dim3(BLOCK), 0,
+                       stream, (const float*)x, out, N, H, rows_per_chunk);
 }
 
 // ---- AdamW over flat fp32 state, low-precision model copy + grads ----
@@ -190,18 +210,20 @@ torch::Tensor bias_gelu_fwd(torch::Tensor x, torch::Tensor bias) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous());
   long total = x.numel();
   int H = x.size(-1);
-  TORCH_CHECK(total % 4 == 0, "numel must be divisible by 4");
+  long N = total / H;
+  TORCH_CHECK(H % 4 == 0, "last dim must be divisible by 4");
   bool has_bias = bias.defined() && bias.numel() > 0;
-  if (has_bias) TORCH_CHECK(H % 4 == 0);
   auto y = torch::empty_like(x);
   auto stream = at::hip::getCurrentHIPStream();
-  bool v8 = (total % 8 == 0) && (!has_bias || H % 8 == 0);
-  dim3 grid(grid_for(total, BLOCK * (v8 ? 8 : 4)));
+  bool v8 = H % 8 == 0;
+  int vec = v8 ? 8 : 4;
+  dim3 grid((H + BLOCK * vec - 1) / (BLOCK * vec),
+            (unsigned)std::min<long>(N, 16384));
 #define LAUNCH_BG(T, HB, V)                                                  \
   hipLaunchKernelGGL((bias_gelu_fwd_kernel<T, HB, V>), grid, dim3(BLOCK), 0, \
                      stream, (const T*)x.data_ptr(),                         \
                      has_bias ? (const T*)bias.data_ptr() : nullptr,         \
-                     (T*)y.data_ptr(), total, H)
+                     (T*)y.data_ptr(), N, H)
 #define PICK_BG(T, HB) do { if (v8) LAUNCH_BG(T, HB, 8); else LAUNCH_BG(T, HB, 4); } while (0)
   if (x.scalar_type() == torch::kBFloat16) {
     if (has_bias) PICK_BG(__hip_bfloat16, true);
@@ -220,16 +242,20 @@ std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() && dy.is_contiguous());
   long total = x.numel();
   int H = x.size(-1);
+  long N = total / H;
+  TORCH_CHECK(H % 4 == 0, "last dim must be divisible by 4");
   bool has_bias = bias.defined() && bias.numel() > 0;
   auto dx = torch::empty_like(x);
   auto stream = at::hip::getCurrentHIPStream();
-  bool v8 = (total % 8 == 0) && (!has_bias || H % 8 == 0);
-  dim3 grid(grid_for(total, BLOCK * (v8 ? 8 : 4)));
+  bool v8 = H % 8 == 0;
+  int vec = v8 ? 8 : 4;
+  dim3 grid((H + BLOCK * vec - 1) / (BLOCK * vec),
+            (unsigned)std::min<long>(N, 16384));
 #define LAUNCH_BG(T, HB, V)                                                  \
   hipLaunchKernelGGL((bias_gelu_bwd_kernel<T, HB, V>), grid, dim3(BLOCK), 0, \
                      stream, (const T*)dy.data_ptr(), (const T*)x.data_ptr(), \
                      has_bias ? (const T*)bias.data_ptr() : nullptr,         \
-                     (T*)dx.data_ptr(), total, H)
+                     (T*)dx.data_ptr(), N, H)
 #define PICK_BG(T, HB) do { if (v8) LAUNCH_BG(T, HB, 8); else LAUNCH_BG(T, HB, 4); } while (0)
   if (x.scalar_type() == torch::kBFloat16) {
     if (has_bias) PICK_BG(__hip_bfloat16, true);
@@ -244,19 +270,9 @@ std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
   if (has_bias) {
     // fused column sum (fp32 accumulate) — replaces the materialized
     // .to(float).sum(0), which cost two extra full passes over dx
-    long N = total / H;
     db = torch::zeros({H}, x.options().dtype(torch::kFloat));
-    int rows_per_chunk = std::max<long>(16, N / 256);
-    int nchunks = (int)((N + rows_per_chunk - 1) / rows_per_chunk);
-    dim3 grid2((H + BLOCK - 1) / BLOCK, nchunks);
-    if (x.scalar_type() == torch::kBFloat16)
-      hipLaunchKernelGGL((colsum_kernel<__hip_bfloat16>), grid2, dim3(BLOCK),
-                         0, stream, (const __hip_bfloat16*)dx.data_ptr(),
-                         db.data_ptr<float>(), N, H, rows_per_chunk);
-    else
-      hipLaunchKernelGGL((colsum_kernel<float>), grid2, dim3(BLOCK), 0,
-                         stream, (const float*)dx.data_ptr(),
-                         db.data_ptr<float>(), N, H, rows_per_chunk);
+    launch_colsum(dx.data_ptr(), db.data_ptr<float>(), N, H,
+                  x.scalar_type() == torch::kBFloat16, stream);
   }
   return {dx, db};
 }
@@ -266,19 +282,11 @@ torch::Tensor colsum(torch::Tensor x) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous());
   long N = x.size(0);
   int H = x.size(1);
+  TORCH_CHECK(H % 4 == 0, "H must be divisible by 4");
   auto out = torch::zeros({H}, x.options().dtype(torch::kFloat));
   auto stream = at::hip::getCurrentHIPStream();
-  int rows_per_chunk = std::max<long>(16, N / 256);
-  int nchunks = (int)((N + rows_per_chunk - 1) / rows_per_chunk);
-  dim3 grid2((H + BLOCK - 1) / BLOCK, nchunks);
-  if (x.scalar_type() == torch::kBFloat16)
-    hipLaunchKernelGGL((colsum_kernel<__hip_bfloat16>), grid2, dim3(BLOCK), 0,
-                       stream, (const __hip_bfloat16*)x.data_ptr(),
-                       out.data_ptr<float>(), N, H, rows_per_chunk);
-  else
-    hipLaunchKernelGGL((colsum_kernel<float>), grid2, dim3(BLOCK), 0, stream,
-                       (const float*)x.data_ptr(), out.data_ptr<float>(), N, H,
-                       rows_per_chunk);
+  launch_colsum(x.data_ptr(), out.data_ptr<float>(), N, H,
+                x.scalar_type() == torch::kBFloat16, stream);
   return out;
 }
 
