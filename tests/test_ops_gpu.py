@@ -234,6 +234,26 @@ def test_flash_attention_noncausal(dev, cfg):
             f"{name}: max err {err}"
 
 
+def test_flash_attention_defer_max_spike(dev):
+    """T13 defer-max correctness (guide §5.4 rule 26): a spiked K row at
+    a late kv tile forces the rescale branch; output must still match the
+    fp32 reference."""
+    B, H, S, D = 1, 2, 512, 128
+    torch.manual_seed(9)
+    q = torch.randn(B, H, S, D, device=dev, dtype=torch.bfloat16)
+    k = torch.randn_like(q)
+    v = torch.randn_like(q)
+    # spike: one K row at kv=300 (tile 4) correlates strongly with all Q
+    k[:, :, 300, :] = q.mean(dim=2) * 40.0
+    scale = D ** -0.5
+    o, lse = hip_ext().attn_fwd(q, k, v, True, scale)
+    o2, lse2 = ref.attention_fwd(q, k, v, True, scale)
+    assert torch.allclose(o.float(), o2.float(), atol=5e-2, rtol=5e-2), \
+        (o.float() - o2.float()).abs().max()
+    m = torch.isfinite(lse2)
+    assert torch.allclose(lse[m], lse2[m], atol=1e-2, rtol=1e-3)
+
+
 def test_topp_sampling(dev):
     B, V = 8, 50304
     logits = torch.randn(B, V, device=dev)
