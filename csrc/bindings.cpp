@@ -36,6 +36,8 @@ torch::Tensor colsum(torch::Tensor x);
 torch::Tensor softmax_causal_fwd(torch::Tensor s, double scale);
 torch::Tensor softmax_causal_bwd(torch::Tensor dy, torch::Tensor y,
                                  double scale);
+torch::Tensor softmax_bias_fwd(torch::Tensor s, torch::Tensor bias,
+                               long inner, long outer, double scale);
 std::vector<torch::Tensor> cross_entropy_fwd(torch::Tensor logits,
                                              torch::Tensor labels,
                                              long ignore_index);
@@ -86,6 +88,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_causal_fwd", &softmax_causal_fwd,
         "fused scale+causal-mask+softmax fwd");
   m.def("softmax_causal_bwd", &softmax_causal_bwd, "causal softmax bwd");
+  m.def("softmax_bias_fwd", &softmax_bias_fwd,
+        "fused scale+bias+softmax (gated attention core)");
   m.def("cross_entropy_fwd", &cross_entropy_fwd, "softmax CE fwd");
   m.def("cross_entropy_bwd", &cross_entropy_bwd, "softmax CE bwd");
   m.def("row_max", &row_max, "rowwise max (vocab-parallel CE)");

@@ -50,6 +50,56 @@ __global__ void softmax_causal_fwd_kernel(const T* __restrict__ s,
   }
 }
 
+// ---- fused scale + bias-add + softmax (non-causal) on [R, K] ----
+// bias broadcasts with period `inner` over an outer stride `outer`:
+// bias_row = (row / outer) * inner + row % inner — covers the folding
+// model's [B, 1, h, Q, K] pair-bias against [B, S, h, Q, K] logits
+// (reference fused_gate_attention, protein_folding/attentions.py:126).
+template <typename T>
+__global__ void softmax_bias_fwd_kernel(const T* __restrict__ s,
+                                        const T* __restrict__ bias,
+                                        T* __restrict__ y, int K,
+                                        long inner, long outer,
+                                        float scale) {
+  __shared__ float sred[BLOCK / WAVE];
+  long row = blockIdx.x;
+  long brow = (row / outer) * inner + row % inner;
+  const T* sr = s + row * K;
+  const T* br = bias + brow * K;
+  T* yr = y + row * K;
+  float mx = -INFINITY;
+  for (int i = threadIdx.x; i < K; i += BLOCK) {
+    float v = (sizeof(T) == 2) ? bf_raw2f(((const unsigned short*)sr)[i])
+                               : ((const float*)sr)[i];
+    float b = (sizeof(T) == 2) ? bf_raw2f(((const unsigned short*)br)[i])
+                               : ((const float*)br)[i];
+    mx = fmaxf(mx, v * scale + b);
+  }
+  mx = block_reduce_max<BLOCK>(mx, sred);
+  __syncthreads();
+  float sum = 0.f;
+  for (int i = threadIdx.x; i < K; i += BLOCK) {
+    float v = (sizeof(T) == 2) ? bf_raw2f(((const unsigned short*)sr)[i])
+                               : ((const float*)sr)[i];
+    float b = (sizeof(T) == 2) ? bf_raw2f(((const unsigned short*)br)[i])
+                               : ((const float*)br)[i];
+    sum += __expf(v * scale + b - mx);
+  }
+  sum = block_reduce_sum<BLOCK>(sum, sred);
+  float inv = 1.f / sum;
+  for (int i = threadIdx.x; i < K; i += BLOCK) {
+    float v = (sizeof(T) == 2) ? bf_raw2f(((const unsigned short*)sr)[i])
+                               : ((const float*)sr)[i];
+    float b = (sizeof(T) == 2) ? bf_raw2f(((const unsigned short*)br)[i])
+                               : ((const float*)br)[i];
+    float o = __expf(v * scale + b - mx) * inv;
+    if constexpr (sizeof(T) == 2)
+      ((unsigned short*)yr)[i] = f2bf_raw(o);
+    else
+      ((float*)yr)[i] = o;
+  }
+}
+
 // ds = y * (dy - sum(dy*y)) * scale
 template <typename T>
 __global__ void softmax_causal_bwd_kernel(const T* __restrict__ dy,
@@ -251,6 +301,25 @@ torch::Tensor softmax_causal_fwd(torch::Tensor s, double scale) {
                                    dim3(BLOCK), 0, stream,
                                    (const T*)s.data_ptr(), (T*)y.data_ptr(),
                                    Sq, Sk, (float)scale));
+  return y;
+}
+
+// softmax(s*scale + bias) rowwise over the last dim; bias broadcast
+// with period `inner` / outer stride `outer` (see kernel comment)
+torch::Tensor softmax_bias_fwd(torch::Tensor s, torch::Tensor bias,
+                               long inner, long outer, double scale) {
+  TORCH_CHECK(s.is_cuda() && s.is_contiguous() && bias.is_contiguous());
+  int K = s.size(-1);
+  TORCH_CHECK(bias.size(-1) == K);
+  long rows = s.numel() / K;
+  auto y = torch::empty_like(s);
+  auto stream = at::hip::getCurrentHIPStream();
+  DISPATCH_T(s, hipLaunchKernelGGL((softmax_bias_fwd_kernel<T>), dim3(rows),
+                                   dim3(BLOCK), 0, stream,
+                                   (const T*)s.data_ptr(),
+                                   (const T*)bias.data_ptr(),
+                                   (T*)y.data_ptr(), K, inner, outer,
+                                   (float)scale));
   return y;
 }
 

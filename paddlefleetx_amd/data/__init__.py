@@ -39,6 +39,8 @@ def _register_lazy():
     from paddlefleetx_amd.data.multimodal_dataset import \
         SyntheticImagenDataset
     _DATASETS.setdefault("SyntheticImagenDataset", SyntheticImagenDataset)
+    from paddlefleetx_amd.data.folding_dataset import SyntheticFoldingDataset
+    _DATASETS.setdefault("SyntheticFoldingDataset", SyntheticFoldingDataset)
 
 
 _register_lazy()

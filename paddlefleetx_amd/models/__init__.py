@@ -34,6 +34,9 @@ def build_module(config):
     if name == "MOCOModule":
         from paddlefleetx_amd.models.moco import MOCOModule
         table["MOCOModule"] = MOCOModule
+    if name == "FoldingModule":
+        from paddlefleetx_amd.models.folding_module import FoldingModule
+        table["FoldingModule"] = FoldingModule
     if name in ("ErnieModule", "ErnieSeqClsModule"):
         from paddlefleetx_amd.models.ernie_module import (ErnieModule,
                                                           ErnieSeqClsModule)

@@ -45,7 +45,13 @@ class GatedAttention(nn.Module):
 
     def forward(self, q_in, kv_in=None, bias=None):
         """q_in [*, Q, Cq]; kv_in [*, K, Ck]; bias broadcastable to
-        [*, heads, Q, K]."""
+        [*, heads, Q, K].
+
+        Fused MI355X paths (the reference's fused_gate_attention,
+        attentions.py:126): with a pair bias the scale+bias+softmax runs
+        in ONE kernel (ops.fused_softmax_bias); without a bias and at
+        head_dim 64/128 the whole core runs through the flash-attention
+        MFMA kernel (non-causal)."""
         kv_in = q_in if kv_in is None else kv_in
         *lead, Q, _ = q_in.shape
         K = kv_in.shape[-2]
@@ -53,11 +59,24 @@ class GatedAttention(nn.Module):
         q = self.q(q_in).view(*lead, Q, h, d).transpose(-2, -3) * self.scale
         k = self.k(kv_in).view(*lead, K, h, d).transpose(-2, -3)
         v = self.v(kv_in).view(*lead, K, h, d).transpose(-2, -3)
-        logits = torch.matmul(q, k.transpose(-1, -2))
-        if bias is not None:
-            logits = logits + bias
-        weights = logits.float().softmax(dim=-1).to(q_in.dtype)
-        o = torch.matmul(weights, v).transpose(-2, -3)  # [*, Q, h, d]
+        if (bias is None and q.is_cuda and q.dtype == torch.bfloat16
+                and d in (64, 128) and Q == K):
+            from paddlefleetx_amd.ops import flash_attention
+            q4 = q.reshape(-1, h, Q, d)
+            k4 = k.reshape(-1, h, K, d)
+            v4 = v.reshape(-1, h, K, d)
+            o = flash_attention(q4, k4, v4, causal=False, scale=1.0)
+            o = o.view(*lead, h, Q, d).transpose(-2, -3)
+        else:
+            logits = torch.matmul(q, k.transpose(-1, -2))
+            if bias is not None:
+                from paddlefleetx_amd.ops import fused_softmax_bias
+                weights = fused_softmax_bias(
+                    logits, bias.expand(*([-1] * (bias.dim() - 3)), h, Q, K),
+                    scale=1.0)
+            else:
+                weights = logits.float().softmax(dim=-1).to(q_in.dtype)
+            o = torch.matmul(weights, v).transpose(-2, -3)  # [*, Q, h, d]
         o = o.reshape(*lead, Q, h * d)
         if self.gating:
             o = o * torch.sigmoid(self.gate(q_in))

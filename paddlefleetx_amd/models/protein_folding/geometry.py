@@ -192,7 +192,13 @@ class Rigid:
                           self.trans], dim=-1)
 
     def __getitem__(self, idx) -> "Rigid":
-        return Rigid(self.rot[idx], self.trans[idx])
+        """Index the BATCH dims (the trailing 3x3 / 3 stay intact), so
+        `frames[..., None]` unsqueezes a broadcast dim like on a [*, 3]
+        tensor of translations."""
+        if not isinstance(idx, tuple):
+            idx = (idx,)
+        return Rigid(self.rot[idx + (slice(None), slice(None))],
+                     self.trans[idx + (slice(None),)])
 
     @property
     def shape(self):

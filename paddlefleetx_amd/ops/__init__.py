@@ -58,6 +58,7 @@ from paddlefleetx_amd.ops.functional import (  # noqa: E402,F401
     flash_attention,
     flash_attention_packed,
     fused_adamw_flat,
+    fused_softmax_bias,
     fused_softmax_causal,
     layernorm,
     layernorm_residual,

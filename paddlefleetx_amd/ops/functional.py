@@ -23,7 +23,8 @@ from paddlefleetx_amd.ops import hip_ext, use_hip
 __all__ = [
     "layernorm", "layernorm_residual", "rmsnorm", "FusedLayerNorm",
     "FusedRMSNorm", "bias_gelu",
-    "flash_attention", "flash_attention_packed", "fused_softmax_causal", "cross_entropy",
+    "flash_attention", "flash_attention_packed", "fused_softmax_causal",
+    "fused_softmax_bias", "cross_entropy",
     "fused_adamw_flat", "rope", "topp_sampling",
 ]
 
@@ -311,6 +312,70 @@ class _SoftmaxCausalFn(torch.autograd.Function):
 def fused_softmax_causal(scores, scale: float = 1.0):
     """[B,H,Sq,Sk] scores -> causal softmax(scale*scores). Fused mask, no mask tensor."""
     return _SoftmaxCausalFn.apply(scores, scale)
+
+
+def _bias_period(s_shape, b_shape):
+    """For bias broadcasting over one contiguous run of leading dims
+    (e.g. logits [B, S, h, Q, K] vs bias [B, 1, h, Q, K]) return
+    (inner, outer) row periods for the fused kernel, or None."""
+    if len(s_shape) != len(b_shape) or s_shape[-1] != b_shape[-1]:
+        return None
+    lead_s, lead_b = s_shape[:-1], b_shape[:-1]
+    bc = [i for i, (a, b) in enumerate(zip(lead_s, lead_b))
+          if b == 1 and a != 1]
+    eq = all(a == b for i, (a, b) in enumerate(zip(lead_s, lead_b))
+             if i not in bc)
+    if not eq:
+        return None
+    if not bc:
+        return 1, 1  # no broadcast: brow == row
+    if bc != list(range(bc[0], bc[-1] + 1)):
+        return None  # non-contiguous broadcast run
+    inner = 1
+    for d in lead_s[bc[-1] + 1:]:
+        inner *= d
+    outer = inner
+    for d in lead_s[bc[0]:bc[-1] + 1]:
+        outer *= d
+    return inner, outer
+
+
+class _SoftmaxBiasFn(torch.autograd.Function):
+    """softmax(scores * scale + bias) over the last dim with the bias
+    broadcast handled inside ONE kernel — the fused gated-attention core
+    (reference fused_gate_attention, protein_folding/attentions.py:126)."""
+
+    @staticmethod
+    def forward(ctx, scores, bias, scale, period):
+        sc = scores.contiguous()
+        bc_ = bias.contiguous()
+        y = hip_ext().softmax_bias_fwd(sc, bc_, period[0], period[1], scale)
+        ctx.save_for_backward(y)
+        ctx.scale = scale
+        ctx.bias_shape = bias.shape
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (y,) = ctx.saved_tensors
+        dy = dy.contiguous()
+        ds = hip_ext().softmax_causal_bwd(dy, y, ctx.scale)  # generic bwd
+        dlogits = ds if ctx.scale == 1.0 else ds / ctx.scale
+        red = [i for i, (a, b) in enumerate(zip(y.shape, ctx.bias_shape))
+               if b == 1 and a != 1]
+        db = dlogits.sum(dim=red, keepdim=True) if red else dlogits
+        return ds, db.to(y.dtype), None, None
+
+
+def fused_softmax_bias(scores, bias, scale: float = 1.0):
+    """Fused softmax(scale*scores + bias) when on GPU and the bias
+    broadcast is a contiguous leading run; torch fallback otherwise."""
+    if use_hip(scores):
+        period = _bias_period(scores.shape, bias.shape)
+        if period is not None:
+            return _SoftmaxBiasFn.apply(scores, bias, scale, period)
+    return torch.softmax(scores.float() * scale + bias.float(),
+                         dim=-1).to(scores.dtype)
 
 
 # ---------------------------------------------------------------------------
