@@ -42,8 +42,13 @@ def get_pairs(word: Tuple[str, ...]):
     return pairs
 
 
-class GPTTokenizer:
-    """Byte-level BPE with the GPT-2 regex pre-tokenizer."""
+from paddlefleetx_amd.data.tokenizers.tokenization_base import (
+    BatchEncoding, TokenizerBase)
+
+
+class GPTTokenizer(TokenizerBase):
+    """Byte-level BPE with the GPT-2 regex pre-tokenizer; batch
+    encode/pad/truncate surface from TokenizerBase."""
 
     # GPT-2 pre-tokenization pattern (reference :158)
     PAT = (r"'s|'t|'re|'ve|'m|'ll|'d| ?\p{L}+| ?\p{N}+|"
@@ -64,11 +69,9 @@ class GPTTokenizer:
             self._re = regex.compile(self.PAT)
         except ImportError:  # plain-re fallback (no \p classes)
             self._re = re.compile(r"\S+|\s+")
-        self.eos_token = eos_token
-        self.pad_token = pad_token or eos_token
-        self.eos_token_id = self.encoder.get(eos_token)
-        self.pad_token_id = self.encoder.get(self.pad_token,
-                                             self.eos_token_id)
+        TokenizerBase.__init__(self, eos_token=eos_token,
+                               pad_token=pad_token or eos_token,
+                               unk_token=eos_token)
 
     # -- constructors -------------------------------------------------------
     @classmethod
@@ -144,16 +147,27 @@ class GPTTokenizer:
             bpe_tokens.extend(self.bpe(token).split(" "))
         return bpe_tokens
 
-    def convert_tokens_to_ids(self, tokens: List[str]) -> List[int]:
-        return [self.encoder[t] for t in tokens]
+    # TokenizerBase hooks
+    def _tokenize(self, text: str) -> List[str]:
+        return self.tokenize(text)
 
-    def encode(self, text: str) -> List[int]:
-        return self.convert_tokens_to_ids(self.tokenize(text))
+    def _convert_token_to_id(self, token: str) -> int:
+        return self.encoder[token]
 
-    def decode(self, ids: List[int]) -> str:
-        text = "".join(self.decoder[int(i)] for i in ids)
-        data = bytearray(self.byte_decoder[c] for c in text)
+    def _convert_id_to_token(self, idx: int) -> str:
+        return self.decoder[int(idx)]
+
+    def convert_tokens_to_string(self, tokens: List[str]) -> str:
+        data = bytearray(self.byte_decoder[c] for c in "".join(tokens))
         return data.decode("utf-8", errors=self.errors)
+
+    def encode(self, text: str, **kwargs) -> List[int]:
+        if not kwargs:
+            return self.convert_tokens_to_ids(self.tokenize(text))
+        return TokenizerBase.encode(self, text, **kwargs)
+
+    def decode(self, ids, skip_special_tokens: bool = False) -> str:
+        return TokenizerBase.decode(self, ids, skip_special_tokens)
 
     def __len__(self) -> int:
         return len(self.encoder)
@@ -162,10 +176,15 @@ class GPTTokenizer:
     def vocab_size(self) -> int:
         return len(self.encoder)
 
-    def pad(self, batch_ids: List[List[int]], max_length: Optional[int] = None,
-            pad_to_multiple_of: Optional[int] = None
-            ) -> Dict[str, List[List[int]]]:
-        """Left-unchanged right-padding (reference :481-520)."""
+    def pad(self, batch_ids, max_length: Optional[int] = None,
+            pad_to_multiple_of: Optional[int] = None, **kw):
+        """Right-padding (reference :481-520). Accepts the legacy
+        list-of-id-lists OR a BatchEncoding (TokenizerBase surface)."""
+        if isinstance(batch_ids, dict):
+            return TokenizerBase.pad(self, batch_ids,
+                                     max_length=max_length,
+                                     pad_to_multiple_of=pad_to_multiple_of,
+                                     **kw)
         longest = max(len(x) for x in batch_ids)
         target = max_length or longest
         if pad_to_multiple_of:
