@@ -401,6 +401,7 @@ class EagerEngine(BasicEngine):
         if not ccfg:
             return
         from paddlefleetx_amd.utils.compression_helper import (prune_model,
+                                                               qat_model,
                                                                quant_model)
         if "Prune" in ccfg:
             p = ccfg["Prune"] or {}
@@ -410,7 +411,13 @@ class EagerEngine(BasicEngine):
                         include=p.get("include"))
         if "Quantization" in ccfg:
             q = ccfg["Quantization"] or {}
-            quant_model(self.module.model, include=q.get("include"))
+            if q.get("enable_qat") or ccfg.get("enable_qat"):
+                # training-time fake quant (reference quant_model
+                # compression_helper.py:210 QAT path)
+                qat_model(self.module.model, include=q.get("include"),
+                          bits=int(q.get("weight_bits", 8)))
+            else:
+                quant_model(self.module.model, include=q.get("include"))
 
     def evaluate(self, valid_data_loader=None, epoch: int = 0):
         self._evaluate_impl(epoch, valid_data_loader)
