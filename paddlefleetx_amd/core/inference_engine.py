@@ -33,24 +33,43 @@ class InferenceEngine:
         mcfg = dict(meta["model"])
         self.device = torch.device("cuda") if torch.cuda.is_available() \
             else torch.device("cpu")
-        from paddlefleetx_amd.models.gpt.generation import GPTForGeneration
-        from paddlefleetx_amd.models.gpt.model import GPTModel
+        dtype = torch.bfloat16 if self.device.type == "cuda" else torch.float32
+        family = mcfg.get("module") or mcfg.get("name") or "GPTModule"
         gen_cfg = dict(meta.get("generation", {}))
         gen_cfg.update(generation_cfg or {})
-        dtype = torch.bfloat16 if self.device.type == "cuda" else torch.float32
-        gpt = GPTModel(dtype=dtype,
-                       **{k: v for k, v in mcfg.items()
-                          if k not in ("name", "module")})
-        self.model = GPTForGeneration(gpt, gen_cfg)
+        if family in ("GPTModule", "GPTGenerationModule") :
+            # decode-optimized runtime: KV-cache + top-p HIP kernel
+            from paddlefleetx_amd.models.gpt.generation import \
+                GPTForGeneration
+            from paddlefleetx_amd.models.gpt.model import GPTModel
+            gpt = GPTModel(dtype=dtype,
+                           **{k: v for k, v in mcfg.items()
+                              if k not in ("name", "module")})
+            self.model = GPTForGeneration(gpt, gen_cfg)
+        else:
+            # model-generic path (reference inference_engine.py:144-271
+            # loads any exported program): rebuild the network from the
+            # exported Model config through the module factory
+            from paddlefleetx_amd.models import build_module
+            module = build_module({"Model": mcfg})
+            self.model = module.model.to(dtype)
         load_inference_model(self.model, model_dir)
         self.model.to(self.device).eval()
         logger.info(f"inference engine ready (model_dir={model_dir}, "
-                    f"mp={mp_degree}, device={self.device})")
+                    f"family={family}, mp={mp_degree}, "
+                    f"device={self.device})")
 
     @torch.no_grad()
-    def predict(self, input_ids) -> torch.Tensor:
-        if not torch.is_tensor(input_ids):
-            input_ids = torch.tensor(input_ids, dtype=torch.long)
-        if input_ids.ndim == 1:
-            input_ids = input_ids.unsqueeze(0)
-        return self.model(input_ids.to(self.device))
+    def predict(self, *inputs) -> torch.Tensor:
+        """Single-tensor convenience (ids [B, S] or [S]) or generic
+        positional inputs for non-LM exported models."""
+        if len(inputs) == 1:
+            x = inputs[0]
+            if not torch.is_tensor(x):
+                x = torch.tensor(x, dtype=torch.long)
+            if x.ndim == 1:
+                x = x.unsqueeze(0)
+            return self.model(x.to(self.device))
+        moved = tuple(t.to(self.device) if torch.is_tensor(t) else t
+                      for t in inputs)
+        return self.model(*moved)

@@ -52,3 +52,30 @@ def test_profiler_guard_disabled_and_enabled(tmp_path):
         (x @ x).sum()
         g2.step()
     g2.stop_and_summary()
+
+
+def test_inference_engine_model_generic_vit(tmp_path):
+    """Non-GPT family export -> model-generic InferenceEngine rebuild
+    (reference inference_engine.py:144-271 loads any exported program)."""
+    from paddlefleetx_amd.parallel.env import set_hcg
+    from paddlefleetx_amd.parallel.topology import HybridTopology
+    import torch.distributed as dist
+    if not dist.is_initialized():
+        set_hcg(HybridTopology())
+    from paddlefleetx_amd.core.inference_engine import InferenceEngine
+    from paddlefleetx_amd.utils.export import export_inference_model
+    from paddlefleetx_amd.models import build_module
+
+    mcfg = {"module": "GeneralClsModule", "name": "GeneralClsModule",
+            "model": {"name": "ViT_tiny_patch16_224", "class_num": 10},
+            "loss": {"train": {"name": "CELoss"}},
+            "metric": {"name": "TopkAcc"}}
+    module = build_module({"Model": mcfg})
+    out_dir = str(tmp_path / "vit_exported")
+    export_inference_model(module.model, mcfg, out_dir)
+    engine = InferenceEngine(out_dir, mp_degree=1)
+    x = torch.randn(2, 3, 224, 224)
+    with torch.no_grad():
+        ref = module.model(x)
+    out = engine.predict(x)
+    assert torch.allclose(ref, out.cpu(), atol=1e-5)
