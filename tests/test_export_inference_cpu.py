@@ -71,6 +71,7 @@ def test_inference_engine_model_generic_vit(tmp_path):
             "loss": {"train": {"name": "CELoss"}},
             "metric": {"name": "TopkAcc"}}
     module = build_module({"Model": mcfg})
+    module.model = module.model.float()
     out_dir = str(tmp_path / "vit_exported")
     export_inference_model(module.model, mcfg, out_dir)
     engine = InferenceEngine(out_dir, mp_degree=1)
