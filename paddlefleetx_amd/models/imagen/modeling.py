@@ -126,8 +126,11 @@ class ImagenModel(nn.Module):
             return self.text_encoder(text_ids, attention_mask=text_mask)
 
     def forward(self, images, text_ids=None, text_mask=None,
-                text_embeds=None):
-        """Training loss for one denoising step."""
+                text_embeds=None, lowres_images=None):
+        """Training loss for one denoising step. SR stages
+        (unet.lowres_cond) take `lowres_images` (or derive them by
+        downsampling the targets) with noise augmentation — reference
+        modeling.py lowres conditioning path."""
         B = images.shape[0]
         device = images.device
         if text_embeds is None and text_ids is not None:
@@ -137,18 +140,41 @@ class ImagenModel(nn.Module):
             keep = (torch.rand(B, device=device) >
                     self.cond_drop_prob)[:, None, None]
             text_embeds = text_embeds * keep
+        lowres_cond_img = None
+        if getattr(self.unet, "lowres_cond", False):
+            if lowres_images is None:
+                lowres_images = F.interpolate(
+                    images, scale_factor=0.25, mode="bilinear",
+                    align_corners=False)
+            lowres_cond_img = F.interpolate(
+                lowres_images, size=images.shape[-2:], mode="bilinear",
+                align_corners=False)
+            if self.training:
+                # lowres noise augmentation (aug time <= 0.5)
+                aug_t = torch.rand(B, device=device) * 0.5
+                lowres_cond_img, _ = self.scheduler.q_sample(
+                    lowres_cond_img, aug_t)
         t = self.scheduler.sample_random_times(B, device)
         x_t, noise = self.scheduler.q_sample(images, t)
         pred = self.unet(x_t.to(images.dtype), t, text_embeds=text_embeds,
-                         text_mask=text_mask)
+                         text_mask=text_mask, lowres_cond_img=lowres_cond_img)
         return self.criterion(pred, noise, log_snr=self.scheduler.log_snr(t))
 
     @torch.no_grad()
     def sample(self, text_ids=None, text_mask=None, batch_size: int = 1,
-               steps: int = 50, device=None):
+               steps: int = 50, device=None, lowres_images=None):
         device = device or next(self.unet.parameters()).device
         text_embeds = self.encode_text(text_ids.to(device), text_mask) \
             if text_ids is not None else None
+        lowres_cond_img = None
+        if getattr(self.unet, "lowres_cond", False):
+            assert lowres_images is not None, \
+                "SR stage sampling needs the previous stage's images"
+            lowres_cond_img = F.interpolate(
+                lowres_images.to(device), size=(self.image_size,
+                                                self.image_size),
+                mode="bilinear", align_corners=False)
+            batch_size = lowres_cond_img.shape[0]
         x = torch.randn(batch_size, self.unet.channels, self.image_size,
                         self.image_size, device=device)
         times = torch.linspace(1.0, 0.0, steps + 1, device=device)
@@ -156,7 +182,8 @@ class ImagenModel(nn.Module):
             t = times[i].repeat(batch_size)
             t_next = times[i + 1].repeat(batch_size)
             eps = self.unet(x, t, text_embeds=text_embeds,
-                            text_mask=text_mask)
+                            text_mask=text_mask,
+                            lowres_cond_img=lowres_cond_img)
             x0 = self.scheduler.predict_start_from_noise(x, t, eps)
             if self.dynamic_thresholding:
                 # per-sample percentile of |x0|, clamp + renormalize
@@ -172,3 +199,58 @@ class ImagenModel(nn.Module):
             noise = torch.randn_like(x) if i < steps - 1 else 0
             x = mean + var.sqrt()[:, None, None, None] * noise
         return x
+
+
+class ImagenCascade(nn.Module):
+    """Cascading DDPM (reference modeling.py:976-1035 Imagen 'base' +
+    SRUnet256 [+ SRUnet1024] chains): one text encoder shared across
+    stages, per-stage schedulers, chained sampling 64 -> 256 (-> 1024).
+    """
+
+    def __init__(self, unets, image_sizes, text_embed_dim: int = 512,
+                 text_encoder_kwargs: Optional[dict] = None,
+                 timesteps: int = 1000, **stage_kwargs):
+        super().__init__()
+        assert len(unets) == len(image_sizes)
+        self.stages = nn.ModuleList()
+        for i, (unet, size) in enumerate(zip(unets, image_sizes)):
+            stage = ImagenModel(unet=unet, image_size=size,
+                                text_embed_dim=text_embed_dim,
+                                text_encoder_kwargs=text_encoder_kwargs,
+                                timesteps=timesteps, **stage_kwargs)
+            if i > 0:
+                # share the (frozen) text encoder with stage 0
+                stage.text_encoder = self.stages[0].text_encoder
+            self.stages.append(stage)
+        self.image_sizes = list(image_sizes)
+
+    def forward(self, images, text_ids=None, text_mask=None,
+                unet_number: int = 0):
+        """Train one stage: the target images are resized to the stage's
+        resolution; SR stages condition on the previous stage's size."""
+        stage = self.stages[unet_number]
+        size = self.image_sizes[unet_number]
+        tgt = F.interpolate(images, size=(size, size), mode="bilinear",
+                            align_corners=False) \
+            if images.shape[-1] != size else images
+        lowres = None
+        if unet_number > 0:
+            prev = self.image_sizes[unet_number - 1]
+            lowres = F.interpolate(images, size=(prev, prev),
+                                   mode="bilinear", align_corners=False)
+        return stage(tgt, text_ids=text_ids, text_mask=text_mask,
+                     lowres_images=lowres)
+
+    @torch.no_grad()
+    def sample(self, text_ids=None, text_mask=None, batch_size: int = 1,
+               steps: int = 25, stop_at_unet_number: Optional[int] = None,
+               device=None):
+        img = None
+        for i, stage in enumerate(self.stages):
+            img = stage.sample(text_ids=text_ids, text_mask=text_mask,
+                               batch_size=batch_size, steps=steps,
+                               device=device, lowres_images=img)
+            if stop_at_unet_number is not None and \
+                    i + 1 >= stop_at_unet_number:
+                break
+        return img

@@ -162,7 +162,8 @@ class Unet(nn.Module):
                  num_resnet_blocks=1, layer_attns=(False, False, True, True),
                  layer_cross_attns=(False, False, True, True),
                  attn_heads: int = 8, attn_dim_head: int = 64,
-                 lowres_cond: bool = False, groups: int = 8, **unused):
+                 lowres_cond: bool = False, groups: int = 8,
+                 memory_efficient: bool = False, **unused):
         super().__init__()
         self.channels = channels
         self.lowres_cond = lowres_cond
@@ -183,20 +184,35 @@ class Unet(nn.Module):
 
         if isinstance(num_resnet_blocks, int):
             num_resnet_blocks = [num_resnet_blocks] * len(dim_mults)
+        if not isinstance(layer_attns, (tuple, list)):
+            layer_attns = [bool(layer_attns)] * len(dim_mults)
 
+        # Efficient U-Net (Imagen paper; reference unet.py memory_efficient
+        # :898, 1139-1148, 1255): each stage DOWNSAMPLES FIRST so its
+        # resnet/attention blocks run at the reduced resolution — the
+        # activation-memory shape the SR stages need at 256/1024 px
+        self.memory_efficient = memory_efficient
         self.downs = nn.ModuleList()
         self.ups = nn.ModuleList()
         n = len(dim_mults)
         for i in range(n):
             d_in, d_out = dims[i], dims[i + 1]
-            blocks = nn.ModuleList([
-                ResnetBlock(d_in if b == 0 else d_out, d_out, time_dim,
-                            groups)
-                for b in range(num_resnet_blocks[i])])
+            if memory_efficient:
+                pre = Downsample(d_in, d_out)
+                blocks = nn.ModuleList([
+                    ResnetBlock(d_out, d_out, time_dim, groups)
+                    for b in range(num_resnet_blocks[i])])
+                down = None
+            else:
+                pre = None
+                blocks = nn.ModuleList([
+                    ResnetBlock(d_in if b == 0 else d_out, d_out, time_dim,
+                                groups)
+                    for b in range(num_resnet_blocks[i])])
+                down = Downsample(d_out, d_out) if i < n - 1 else None
             attn = TransformerBlock(d_out, cond_dim, attn_heads,
                                     attn_dim_head) if layer_attns[i] else None
-            down = Downsample(d_out, d_out) if i < n - 1 else None
-            self.downs.append(nn.ModuleList([blocks, attn, down]))
+            self.downs.append(nn.ModuleList([blocks, attn, down, pre]))
 
         mid = dims[-1]
         self.mid_block1 = ResnetBlock(mid, mid, time_dim, groups)
@@ -212,10 +228,14 @@ class Unet(nn.Module):
                 for b in range(num_resnet_blocks[i])])
             attn = TransformerBlock(d_in, cond_dim, attn_heads,
                                     attn_dim_head) if layer_attns[i] else None
-            up = Upsample(d_in, d_out) if i > 0 else None
+            # memory_efficient pre-downsamples EVERY stage, so every up
+            # stage upsamples (restoring full resolution before the head)
+            up = Upsample(d_in, d_out) if (i > 0 or memory_efficient) \
+                else None
             self.ups.append(nn.ModuleList([blocks, attn, up]))
 
-        self.final_block = ResnetBlock(dims[1], dim, time_dim, groups)
+        final_in = dims[0] if memory_efficient else dims[1]
+        self.final_block = ResnetBlock(final_in, dim, time_dim, groups)
         self.final_conv = nn.Conv2d(dim, channels, 1)
         self._cross = list(layer_cross_attns)
 
@@ -237,7 +257,9 @@ class Unet(nn.Module):
             t = t + self.text_pool_mlp(pooled)
 
         skips = []
-        for blocks, attn, down in self.downs:
+        for blocks, attn, down, pre in self.downs:
+            if pre is not None:
+                x = pre(x)
             for b in blocks:
                 x = b(x, t)
             if attn is not None:
@@ -286,7 +308,7 @@ def SRUnet256(**kw):
     cfg = dict(dim=128, dim_mults=(1, 2, 4, 8), num_resnet_blocks=(2, 4, 8, 8),
                layer_attns=(False, False, False, True),
                layer_cross_attns=(False, False, False, True),
-               lowres_cond=True)
+               lowres_cond=True, memory_efficient=True)
     cfg.update(kw)
     return Unet(**cfg)
 
@@ -294,7 +316,7 @@ def SRUnet256(**kw):
 def SRUnet1024(**kw):
     cfg = dict(dim=128, dim_mults=(1, 2, 4, 8), num_resnet_blocks=(2, 4, 8, 8),
                layer_attns=False, layer_cross_attns=(False, False, False, True),
-               lowres_cond=True)
+               lowres_cond=True, memory_efficient=True)
     if not isinstance(cfg["layer_attns"], (tuple, list)):
         cfg["layer_attns"] = (False, False, False, False)
     cfg.update(kw)

@@ -126,3 +126,67 @@ def test_imagen_module_builds():
     mask = torch.ones(2, 6)
     loss = mod.training_step((imgs, ids, mask))
     assert torch.isfinite(loss)
+
+
+# ---------------------------------------------------------------------------
+# Efficient U-Net + SR cascade (reference unet.py memory_efficient :898,
+# modeling.py SRUnet256 :65 / cascade :976)
+# ---------------------------------------------------------------------------
+
+def test_memory_efficient_unet_shapes():
+    import torch
+    from paddlefleetx_amd.models.imagen.unet import Unet
+    u = Unet(dim=16, dim_mults=(1, 2), num_resnet_blocks=(1, 2),
+             layer_attns=(False, True), layer_cross_attns=(False, True),
+             text_embed_dim=32, memory_efficient=True)
+    x = torch.randn(2, 3, 32, 32)
+    y = u(x, torch.rand(2), text_embeds=torch.randn(2, 5, 32))
+    assert y.shape == x.shape
+    # per the Efficient U-Net design every down stage pre-downsamples
+    assert all(pre is not None for _, _, _, pre in u.downs)
+    assert all(up is not None for _, _, up in u.ups)
+
+
+def test_sr_unet_lowres_conditioning_and_noise_aug():
+    import torch
+    from paddlefleetx_amd.models.imagen.modeling import ImagenModel
+    from paddlefleetx_amd.models.imagen.unet import Unet
+    unet = Unet(dim=16, dim_mults=(1, 2), num_resnet_blocks=1,
+                layer_attns=(False, True), layer_cross_attns=(False, True),
+                text_embed_dim=32, lowres_cond=True, memory_efficient=True)
+    m = ImagenModel(unet=unet, image_size=32, text_embed_dim=32,
+                    text_encoder_kwargs=dict(vocab_size=64, num_layers=1,
+                                             num_heads=2, d_ff=64))
+    m.train()
+    imgs = torch.randn(2, 3, 32, 32)
+    ids = torch.randint(0, 64, (2, 7))
+    loss = m(imgs, text_ids=ids)
+    assert torch.isfinite(loss)
+    loss.backward()
+
+
+def test_imagen_cascade_train_and_sample():
+    import torch
+    from paddlefleetx_amd.models.imagen.modeling import ImagenCascade
+    from paddlefleetx_amd.models.imagen.unet import Unet
+    base = Unet(dim=16, dim_mults=(1, 2), num_resnet_blocks=1,
+                layer_attns=(False, True), layer_cross_attns=(False, True),
+                text_embed_dim=32)
+    sr = Unet(dim=16, dim_mults=(1, 2), num_resnet_blocks=1,
+              layer_attns=(False, True), layer_cross_attns=(False, True),
+              text_embed_dim=32, lowres_cond=True, memory_efficient=True)
+    casc = ImagenCascade([base, sr], image_sizes=[16, 32],
+                         text_embed_dim=32,
+                         text_encoder_kwargs=dict(vocab_size=64,
+                                                  num_layers=1, num_heads=2,
+                                                  d_ff=64))
+    # stage 1 shares the frozen text encoder with stage 0
+    assert casc.stages[1].text_encoder is casc.stages[0].text_encoder
+    imgs = torch.randn(2, 3, 32, 32)
+    ids = torch.randint(0, 64, (2, 6))
+    for stage in (0, 1):
+        loss = casc(imgs, text_ids=ids, unet_number=stage)
+        assert torch.isfinite(loss)
+    casc.eval()
+    out = casc.sample(text_ids=ids, batch_size=2, steps=2)
+    assert out.shape == (2, 3, 32, 32)
