@@ -101,6 +101,11 @@ def _colsum(x2: torch.Tensor) -> torch.Tensor:
 def _fusable(weight: torch.nn.Parameter, bias) -> bool:
     if not _STATE["enabled"] or not torch.is_grad_enabled():
         return False
+    # measured exception (gpurun bench_fused_ops): the fused-QKV wgrad
+    # shape [3h, h] loses hipBLASLt's SplitK at beta=1 (addmm 777us vs
+    # mm 687us + 33us add) — keep plain autograd for that one shape
+    if weight.dim() == 2 and weight.shape[0] == 3 * weight.shape[1]:
+        return False
     if not isinstance(weight, torch.nn.Parameter) or not weight.requires_grad:
         return False
     if _grad_view(weight) is None:
