@@ -60,6 +60,20 @@ def main():
         t_mm = t(lambda: dyw.t().mm(xw))
         print(f"wgrad [{o:5d},{i:5d}]: addmm_ {t_fused:8.1f} us | mm+add {t_sep:8.1f} us | mm alone {t_mm:8.1f} us")
 
+    # MoE dispatch: fused HIP vs torch argsort+index_select
+    T, D2, E, K = 8192, 4096, 64, 2
+    xm = torch.randn(T, D2, device=dev, dtype=bf)
+    ids = torch.randint(0, E, (T * K,), device=dev)
+    def torch_disp():
+        sel = (ids >= 0).nonzero(as_tuple=True)[0]
+        active = ids[sel]
+        perm = torch.argsort(active, stable=True)
+        sel_sorted = sel[perm]
+        torch.bincount(active[perm], minlength=E)
+        return xm.index_select(0, sel_sorted // K)
+    print(f"moe_dispatch fused  (T8192 D4096 E64 k2): {t(lambda: ext.moe_dispatch(xm, ids, E, K)):8.1f} us")
+    print(f"moe_dispatch torch  (argsort+gather):     {t(torch_disp):8.1f} us")
+
     # delta kernel (inside attn bwd) via attn micro shapes
     B, Hh, S, D = 8, 32, 1024, 128
     q = torch.randn(B, Hh, S, D, device=dev, dtype=bf)

@@ -68,6 +68,10 @@ torch::Tensor attn_bwd_packed(torch::Tensor dout, torch::Tensor qkv,
 std::vector<torch::Tensor> topp_select(torch::Tensor sorted_p,
                                        torch::Tensor sorted_idx,
                                        torch::Tensor top_p, torch::Tensor u);
+// moe.hip
+std::vector<torch::Tensor> moe_dispatch(torch::Tensor x,
+                                        torch::Tensor expert_ids,
+                                        long num_experts, long top_k);
 // mfma_probe.hip
 torch::Tensor mfma_gemm16_probe(torch::Tensor a, torch::Tensor b);
 
@@ -96,6 +100,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("row_sumexp", &row_sumexp, "rowwise sum(exp(x-m))");
   m.def("gather_label_logit", &gather_label_logit, "gather label logits");
   m.def("vp_ce_bwd", &vp_ce_bwd, "vocab-parallel CE bwd");
+  m.def("moe_dispatch", &moe_dispatch,
+        "fused MoE token dispatch (histogram+rank+scatter)");
   m.def("attn_fwd", &attn_fwd, "flash attention fwd (MFMA, causal)");
   m.def("attn_bwd", &attn_bwd, "flash attention bwd (MFMA)");
   m.def("attn_fwd_packed", &attn_fwd_packed,

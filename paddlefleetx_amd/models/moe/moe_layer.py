@@ -67,6 +67,44 @@ def _a2a_exchange(x: torch.Tensor, out_splits: List[int],
     return out
 
 
+class _DispatchGather(torch.autograd.Function):
+    """Token dispatch: sort slots by expert and gather their token rows.
+
+    GPU: ONE fused HIP pass (histogram + atomic rank + row scatter,
+    csrc/moe.hip) replacing argsort + index_select (reference
+    global_scatter / _assign_pos, moe/comm_ops.py:28-71). CPU: the
+    stable-argsort equivalent. Backward is the index_add of the row
+    gather (the permutation itself carries no gradient)."""
+
+    @staticmethod
+    def forward(ctx, xf, flat_expert, num_experts, top_k):
+        if xf.is_cuda:
+            from paddlefleetx_amd.ops import hip_ext
+            dispatched, sel_sorted, counts = hip_ext().moe_dispatch(
+                xf, flat_expert, num_experts, top_k)
+        else:
+            sel = (flat_expert >= 0).nonzero(as_tuple=True)[0]
+            active = flat_expert[sel]
+            perm = torch.argsort(active, stable=True)
+            sel_sorted = sel[perm]
+            counts = torch.bincount(active[perm], minlength=num_experts)
+            dispatched = xf.index_select(0, sel_sorted // top_k)
+        ctx.save_for_backward(sel_sorted // top_k)
+        ctx.T = xf.shape[0]
+        return dispatched, sel_sorted, counts
+
+    @staticmethod
+    def backward(ctx, d_disp, _dsel, _dcnt):
+        (token_of_slot,) = ctx.saved_tensors
+        dxf = d_disp.new_zeros(ctx.T, d_disp.shape[-1])
+        dxf.index_add_(0, token_of_slot, d_disp)
+        return dxf, None, None, None
+
+
+def _dispatch_gather(xf, flat_expert, num_experts, top_k):
+    return _DispatchGather.apply(xf, flat_expert, num_experts, top_k)
+
+
 class _AllToAll(torch.autograd.Function):
     """Differentiable uneven all-to-all (comm_ops.py:28-118 global_scatter/
     global_gather collapse into this one primitive + local permutes)."""
@@ -179,14 +217,9 @@ class MoELayer(nn.Module):
             # route dropped slots to a sentinel so they are excluded
             flat_expert = torch.where(keep.reshape(-1), flat_expert,
                                       torch.full_like(flat_expert, -1))
-        sel = (flat_expert >= 0).nonzero(as_tuple=True)[0]
-        active_expert = flat_expert[sel]
-        perm = torch.argsort(active_expert, stable=True)
-        sel_sorted = sel[perm]                   # slot index, sorted by expert
+        dispatched, sel_sorted, counts = _dispatch_gather(
+            xf, flat_expert, self.num_experts, self.top_k)
         token_of_slot = sel_sorted // self.top_k
-        expert_sorted = active_expert[perm]
-
-        counts = torch.bincount(expert_sorted, minlength=self.num_experts)
         # tokens leaving to each EP peer (num_local_experts each)
         send_per_rank = counts.reshape(self.ep_world, -1).sum(dim=1)
 
@@ -209,7 +242,6 @@ class MoELayer(nn.Module):
             recv_matrix = counts.reshape(1, -1)
             in_splits = out_splits = [int(counts.sum())]
 
-        dispatched = xf.index_select(0, token_of_slot)
         dispatched = all_to_all(dispatched, out_splits, in_splits,
                                 self.ep_group_info.group
                                 if self.ep_world > 1 else None)

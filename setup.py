@@ -27,7 +27,8 @@ CSRC = os.path.join(os.path.dirname(os.path.abspath(__file__)), "csrc")
 hip_sources = [
     os.path.join(CSRC, s)
     for s in ("bindings.cpp", "layernorm.hip", "elementwise.hip",
-              "softmax.hip", "attention.hip", "topp.hip", "mfma_probe.hip")
+              "softmax.hip", "attention.hip", "topp.hip", "moe.hip",
+              "mfma_probe.hip")
 ]
 
 ext_modules = [

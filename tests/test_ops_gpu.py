@@ -234,6 +234,33 @@ def test_flash_attention_noncausal(dev, cfg):
             f"{name}: max err {err}"
 
 
+def test_moe_dispatch_fused(dev):
+    """Fused HIP dispatch == the stable-argsort reference up to
+    intra-expert permutation (atomics): same counts, same per-expert row
+    SETS, and the inverse mapping reconstructs every token row."""
+    torch.manual_seed(11)
+    T, D, E, K = 512, 64, 8, 2
+    x = torch.randn(T, D, device=dev, dtype=torch.bfloat16)
+    ids = torch.randint(0, E, (T * K,), device=dev)
+    ids[torch.rand(T * K, device=dev) < 0.1] = -1  # capacity drops
+    disp, sel_sorted, counts = hip_ext().moe_dispatch(x, ids, E, K)
+    # reference
+    sel = (ids >= 0).nonzero(as_tuple=True)[0]
+    active = ids[sel]
+    perm = torch.argsort(active, stable=True)
+    ref_counts = torch.bincount(active[perm], minlength=E)
+    assert torch.equal(counts.cpu(), ref_counts.cpu())
+    assert disp.shape[0] == int(ref_counts.sum())
+    # each slot's row in `disp` equals the row of the token it points to
+    rows_want = x[(sel_sorted // K)]
+    assert torch.equal(disp, rows_want)
+    # slots land inside their expert's segment
+    off = torch.cumsum(ref_counts, 0) - ref_counts
+    seg_of_pos = torch.bucketize(
+        torch.arange(disp.shape[0], device=dev), off, right=True) - 1
+    assert torch.equal(ids[sel_sorted], seg_of_pos)
+
+
 def test_flash_attention_defer_max_spike(dev):
     """T13 defer-max correctness (guide §5.4 rule 26): a spiked K row at
     a late kv tile forces the rescale branch; output must still match the
