@@ -114,8 +114,10 @@ class EagerEngine(BasicEngine):
                     self.module.model, group=sharding_group)
                 ocfg = {k: v for k, v in opt_cfg.items()
                         if k in ("weight_decay", "beta1", "beta2", "epsilon")}
+                sh_cfg = configs.get("Distributed", {}).get("sharding", {})
                 self.optimizer = Stage3AdamW(
-                    self.module.model, lr=self.lr_scheduler.get_lr(), **ocfg)
+                    self.module.model, lr=self.lr_scheduler.get_lr(),
+                    offload=bool(sh_cfg.get("offload", False)), **ocfg)
             else:
                 self.optimizer = build_optimizer(
                     opt_cfg, self.module.model,

@@ -69,6 +69,7 @@ class _Unit:
         self.grad_shard = torch.zeros_like(self.shard)
         self.live = True
         self._grads_pending = 0
+        self._pending_evt = None
 
     # -- storage control ----------------------------------------------------
     def release(self):
@@ -80,6 +81,13 @@ class _Unit:
     def gather(self):
         if self.live:
             return
+        if self._pending_evt is not None:
+            # async prefetch already in flight on the side stream: make
+            # the compute stream wait instead of re-gathering
+            torch.cuda.current_stream().wait_event(self._pending_evt)
+            self._pending_evt = None
+            self.live = True
+            return
         elems = self.numel
         self.flat.untyped_storage().resize_(
             elems * self.flat.element_size())
@@ -87,8 +95,32 @@ class _Unit:
             dist.all_gather_into_tensor(self.flat, self.shard,
                                         group=self.group)
         else:
-            self.flat.copy_(self.shard)
+            # single-process (incl. fake-world sizing rehearsal): replicate
+            self.flat.view(self.world, self.shard_len).copy_(self.shard)
         self.live = True
+
+    def gather_async(self, stream) -> None:
+        """Prefetch the all-gather on a side HIP stream so it overlaps the
+        previous unit's compute (reference group_sharded broadcast_overlap,
+        eager_engine.py:303-307)."""
+        if self.live or self._pending_evt is not None or stream is None:
+            return
+        cur = torch.cuda.current_stream()
+        elems = self.numel
+        self.flat.untyped_storage().resize_(
+            elems * self.flat.element_size())
+        evt0 = torch.cuda.Event()
+        evt0.record(cur)
+        with torch.cuda.stream(stream):
+            stream.wait_event(evt0)
+            if self.world > 1 and dist.is_initialized():
+                dist.all_gather_into_tensor(self.flat, self.shard,
+                                            group=self.group)
+            else:
+                self.flat.view(self.world, self.shard_len).copy_(self.shard)
+            evt = torch.cuda.Event()
+            evt.record(stream)
+        self._pending_evt = evt
 
     def writeback_from_shard(self):
         """After an optimizer step on `shard`, broadcast into flat if live."""
@@ -97,7 +129,7 @@ class _Unit:
                 dist.all_gather_into_tensor(self.flat, self.shard,
                                             group=self.group)
             else:
-                self.flat.copy_(self.shard)
+                self.flat.view(self.world, self.shard_len).copy_(self.shard)
 
     # -- gradients ----------------------------------------------------------
     def reduce_grads(self):
@@ -126,9 +158,12 @@ class GroupShardedStage3(nn.Module):
 
     def __init__(self, model: nn.Module, group=None,
                  unit_classes: Tuple[str, ...] = ("DecoderLayer",
-                                                  "EncoderLayer", "Block")):
+                                                  "EncoderLayer", "Block"),
+                 prefetch: bool = True):
         super().__init__()
         self.model = model
+        self._prefetch = prefetch and torch.cuda.is_available()
+        self._side_stream = torch.cuda.Stream() if self._prefetch else None
         gi = group
         self.group = gi.group if hasattr(gi, "group") else gi
         self.rank = gi.rank if hasattr(gi, "rank") else (
@@ -181,6 +216,13 @@ class GroupShardedStage3(nn.Module):
     def _make_pre(self, u: _Unit):
         def pre(mod, args):
             u.gather()
+            if self._prefetch:
+                # overlap the NEXT unit's all-gather with this unit's
+                # forward compute
+                i = self.units.index(u)
+                if i + 1 < len(self.units) and \
+                        self.units[i + 1].name != "<rest>":
+                    self.units[i + 1].gather_async(self._side_stream)
             return None
         return pre
 
@@ -196,6 +238,12 @@ class GroupShardedStage3(nn.Module):
 
             def regather(_grad):
                 u.gather()
+                if self._prefetch:
+                    # backward walks units in reverse: prefetch the
+                    # PREVIOUS unit's params under this unit's backward
+                    i = self.units.index(u)
+                    if i - 1 >= 0 and self.units[i - 1].name != "<rest>":
+                        self.units[i - 1].gather_async(self._side_stream)
                 return _grad
 
             if torch.is_tensor(out):
@@ -240,15 +288,23 @@ class Stage3AdamW:
     def __init__(self, wrapper: GroupShardedStage3, lr: float = 1e-4,
                  beta1: float = 0.9, beta2: float = 0.95,
                  epsilon: float = 1e-8, weight_decay: float = 0.01,
-                 **unused):
+                 offload: bool = False, **unused):
         self.w = wrapper
         self.lr = lr
         self.beta1, self.beta2, self.eps = beta1, beta2, epsilon
         self.weight_decay = weight_decay
         self._step = 0
+        # CPU offload (reference group_sharded_parallel(offload=True),
+        # eager_engine.py:281-307): fp32 master/m/v live in pinned host
+        # memory; the step runs on CPU and only the bf16 shard returns
+        self.offload = offload
         self.state = []
         for u in wrapper.units:
             master = u.shard.float()
+            if offload:
+                master = master.cpu()
+                if torch.cuda.is_available():
+                    master = master.pin_memory()
             self.state.append({
                 "master": master,
                 "exp_avg": torch.zeros_like(master),
@@ -294,10 +350,20 @@ class Stage3AdamW:
                     u.grad_shard.mul_(scale)
         self._step += 1
         for u, st in zip(self.w.units, self.state):
-            fused_adamw_flat(st["master"], u.grad_shard, st["exp_avg"],
-                             st["exp_avg_sq"], u.shard, self.lr, self.beta1,
-                             self.beta2, self.eps, self.weight_decay,
-                             self._step)
+            if self.offload:
+                g = u.grad_shard.to(st["master"].device,
+                                    non_blocking=False).float()
+                model_out = torch.empty_like(st["master"])
+                fused_adamw_flat(st["master"], g, st["exp_avg"],
+                                 st["exp_avg_sq"], model_out, self.lr,
+                                 self.beta1, self.beta2, self.eps,
+                                 self.weight_decay, self._step)
+                u.shard.copy_(model_out.to(u.shard.dtype))
+            else:
+                fused_adamw_flat(st["master"], u.grad_shard, st["exp_avg"],
+                                 st["exp_avg_sq"], u.shard, self.lr,
+                                 self.beta1, self.beta2, self.eps,
+                                 self.weight_decay, self._step)
             u.writeback_from_shard()
         self.zero_grad()
 
