@@ -159,7 +159,7 @@ class GroupShardedStage3(nn.Module):
     def __init__(self, model: nn.Module, group=None,
                  unit_classes: Tuple[str, ...] = ("DecoderLayer",
                                                   "EncoderLayer", "Block"),
-                 prefetch: bool = True):
+                 prefetch: bool = True, fake_world: int = 0):
         super().__init__()
         self.model = model
         self._prefetch = prefetch and torch.cuda.is_available()
@@ -170,6 +170,12 @@ class GroupShardedStage3(nn.Module):
             dist.get_rank(self.group) if dist.is_initialized() else 0)
         self.world = gi.world_size if hasattr(gi, "world_size") else (
             dist.get_world_size(self.group) if dist.is_initialized() else 1)
+        if fake_world > 1 and self.world == 1:
+            # single-process MEMORY rehearsal of an N-way sharded run:
+            # shards are 1/N sized and gathers replicate — values are not
+            # meaningful across "ranks" but allocation behavior is exact
+            # (the 175B TP8-stage3 sizing check, BASELINE config #3)
+            self.world = fake_world
 
         self.units: List[_Unit] = []
         claimed = set()
