@@ -161,10 +161,11 @@ DEV_INLINE bf8 read_a_frag_lds(const unsigned short* lds, int row_stride,
 // ===========================================================================
 // Forward v2: 8 waves, 128-row Q tile, double-buffered KV, split staging.
 // ===========================================================================
-template <int D, bool CAUSAL>
+template <int D, bool CAUSAL, bool DROP>
 __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
     Strided q, Strided k, Strided v, StridedMut o, float* __restrict__ lse_out,
-    int H, int S, float scale, int kv_total) {
+    int H, int S, float scale, int kv_total, unsigned drop_thresh,
+    float drop_inv_keep, unsigned seed0, unsigned seed1) {
   constexpr int KRS = D + PAD;          // K row-major row stride
   // V^T row stride: 68 shorts (136 B) keeps b64 writes/reads 8B-aligned
   // while avoiding the 0-mod-128B d-stride that made the transpose writes
@@ -387,6 +388,25 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
         psum += __shfl_xor(psum, 32, WAVE);
         l_r[rb][0] += psum;
 
+        if constexpr (DROP) {
+          // attention dropout (reference hybrid_model.py:328 RNG-tracker
+          // dropout): Philox keyed on (q row, kv chunk, bh) -> the SAME
+          // mask regenerates in the backward kernels; the softmax
+          // normalizer l uses the UN-dropped P (dropout acts on the
+          // normalized weights), so only the PV operand is masked
+#pragma unroll
+          for (int ct = 0; ct < 4; ++ct) {
+            unsigned kvc = (unsigned)((kv0 + ct * 16 + crow4) >> 2);
+            philox4 r4 = philox4x32_10((unsigned)my_qrow, kvc,
+                                       (unsigned)bh, seed1, seed0, seed1);
+            const unsigned rr[4] = {r4.x, r4.y, r4.z, r4.w};
+#pragma unroll
+            for (int r = 0; r < 4; ++r)
+              s[rb][ct][r] = (rr[r] >= drop_thresh)
+                  ? s[rb][ct][r] * drop_inv_keep : 0.f;
+          }
+        }
+
         // ---- in-register P hand-off (guide T12): this lane holds ONE
         // q row's 16 P values as 8 packed u32 chunks C[ct][h]
         // (kv = 4g + 16·ct + {2h, 2h+1}, g = lane>>4). The PV A-fragment
@@ -539,11 +559,13 @@ __global__ void attn_bwd_delta_kernel(Strided dout, Strided o,
 // DOUBLE-BUFFERED (158.7 KB of the CU's 160 KB LDS) so each tile costs
 // one barrier instead of two.
 // ===========================================================================
-template <int D, bool CAUSAL>
+template <int D, bool CAUSAL, bool DROP>
 __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dkv_kernel(
     Strided q, Strided k, Strided v, Strided dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
-    StridedMut dk, StridedMut dv, int H, int S, float scale, int q_tiles) {
+    StridedMut dk, StridedMut dv, int H, int S, float scale, int q_tiles,
+    unsigned drop_thresh, float drop_inv_keep, unsigned seed0,
+    unsigned seed1) {
   constexpr int RS = D + PAD;
   constexpr int TS = TILE + PAD;       // P image stride (b128-aligned)
   constexpr int TRS = TILE + 4;        // transposed q/dO stride (see VRS)
@@ -681,6 +703,7 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dkv_kernel(
     f4 pt[4], dpt[4];
     if (active) {
       // ---- S^T = K Q^T; P^T = exp(scale*S^T - lse[q]) ----
+      unsigned dmask[4];  // per-ct keep bits (r=0..3) under dropout
 #pragma unroll
       for (int ct = 0; ct < 4; ++ct) {
         f4 acc = {0.f, 0.f, 0.f, 0.f};
@@ -699,14 +722,32 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dkv_kernel(
             p = __expf(acc[r] * scale - l);
           pt[ct][r] = p;
         }
+        if constexpr (DROP) {
+          // regenerate the fwd mask: counter (q row, kv chunk, bh)
+          philox4 r4 = philox4x32_10((unsigned)qcol,
+                                     (unsigned)(my_kvrow >> 2),
+                                     (unsigned)bh, seed1, seed0, seed1);
+          const unsigned rr[4] = {r4.x, r4.y, r4.z, r4.w};
+          unsigned m = 0;
+#pragma unroll
+          for (int r = 0; r < 4; ++r)
+            if (rr[r] >= drop_thresh) m |= (1u << r);
+          dmask[ct] = m;
+        }
       }
 
-      // ---- stage P^T; dV += P^T dO ----
+      // ---- stage P^T (dropout-masked: dV sees the dropped weights);
+      //      dV += P^T dO ----
 #pragma unroll
       for (int ct = 0; ct < 4; ++ct)
 #pragma unroll
         for (int r = 0; r < 4; r += 2) {
-          unsigned int u = cvt_pk_bf16(pt[ct][r], pt[ct][r + 1]);
+          float p0 = pt[ct][r], p1 = pt[ct][r + 1];
+          if constexpr (DROP) {
+            p0 = (dmask[ct] >> r) & 1 ? p0 * drop_inv_keep : 0.f;
+            p1 = (dmask[ct] >> (r + 1)) & 1 ? p1 * drop_inv_keep : 0.f;
+          }
+          unsigned int u = cvt_pk_bf16(p0, p1);
           unsigned short* base = myp + (crow4 + r) * TS + ct * 16 + ccol;
           base[0] = (unsigned short)u;
           base[TS] = (unsigned short)(u >> 16);
@@ -739,8 +780,12 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dkv_kernel(
         float dlt = lsed_b[TILE + ct * 16 + ccol];
         float ds[4];
 #pragma unroll
-        for (int r = 0; r < 4; ++r)
-          ds[r] = pt[ct][r] * (dpt[ct][r] - dlt) * scale;
+        for (int r = 0; r < 4; ++r) {
+          float dp = dpt[ct][r];
+          if constexpr (DROP)
+            dp = (dmask[ct] >> r) & 1 ? dp * drop_inv_keep : 0.f;
+          ds[r] = pt[ct][r] * (dp - dlt) * scale;
+        }
 #pragma unroll
         for (int r = 0; r < 4; r += 2) {
           unsigned int u = cvt_pk_bf16(ds[r], ds[r + 1]);
@@ -791,11 +836,13 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dkv_kernel(
 // 64-row kv tiles double-buffered in LDS with issue-early staging, one
 // barrier per tile.
 // ===========================================================================
-template <int D, bool CAUSAL>
+template <int D, bool CAUSAL, bool DROP>
 __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dq_kernel(
     Strided q, Strided k, Strided v, Strided dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
-    StridedMut dq, int H, int S, float scale, int kv_total) {
+    StridedMut dq, int H, int S, float scale, int kv_total,
+    unsigned drop_thresh, float drop_inv_keep, unsigned seed0,
+    unsigned seed1) {
   constexpr int RS = D + PAD;
   constexpr int TS = TILE + PAD;   // P image stride
   constexpr int TRS = TILE + 4;    // K^T stride (b64 images, see fwd VRS)
@@ -962,7 +1009,17 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dq_kernel(
             if ((!CAUSAL || kcol <= row_base + r) && kcol < S &&
                 row_base + r < S)
               pv = __expf(sacc[rb][r] * scale - lse_r[rb][r]);
-            ds[r] = pv * (dpacc[rb][r] - dlt_r[rb][r]) * scale;
+            float dp = dpacc[rb][r];
+            if constexpr (DROP) {
+              // q varies with r here, so each value needs its own
+              // counter; only element kcol&3 of the 4 outputs is used
+              philox4 r4 = philox4x32_10((unsigned)(row_base + r),
+                                         (unsigned)(kcol >> 2),
+                                         (unsigned)bh, seed1, seed0, seed1);
+              const unsigned rr[4] = {r4.x, r4.y, r4.z, r4.w};
+              dp = (rr[kcol & 3] >= drop_thresh) ? dp * drop_inv_keep : 0.f;
+            }
+            ds[r] = pv * (dp - dlt_r[rb][r]) * scale;
           }
 #pragma unroll
           for (int r = 0; r < 4; r += 2) {
@@ -1041,23 +1098,33 @@ void check_attn_tensor(const torch::Tensor& t, int d_dim, int D) {
 }
 
 void launch_fwd(Strided q, Strided k, Strided v, StridedMut o, float* lse,
-                int B, int H, int S, int D, float scale, bool causal) {
+                int B, int H, int S, int D, float scale, bool causal,
+                float p_drop = 0.f, unsigned long long seed = 0) {
   int q_blocks = (S + QTILE * 2 - 1) / (QTILE * 2);  // RB=2 home tiles
   int kv_total = (S + TILE - 1) / TILE;
   dim3 grid(B * H, q_blocks);
   auto stream = at::hip::getCurrentHIPStream();
-#define LAUNCH_FWD(DD, CC)                                                  \
-  hipLaunchKernelGGL((attn_fwd_kernel<DD, CC>), grid, dim3(FW_BLOCKT), 0,   \
-                     stream, q, k, v, o, lse, H, S, scale, kv_total)
-  if (D == 128) { if (causal) LAUNCH_FWD(128, true); else LAUNCH_FWD(128, false); }
-  else          { if (causal) LAUNCH_FWD(64, true);  else LAUNCH_FWD(64, false); }
+  unsigned thresh = (unsigned)(p_drop * 4294967296.0);
+  float invk = 1.f / (1.f - p_drop);
+  unsigned s0 = (unsigned)seed, s1 = (unsigned)(seed >> 32);
+  bool drop = p_drop > 0.f;
+#define LAUNCH_FWD(DD, CC, PP)                                              \
+  hipLaunchKernelGGL((attn_fwd_kernel<DD, CC, PP>), grid, dim3(FW_BLOCKT),  \
+                     0, stream, q, k, v, o, lse, H, S, scale, kv_total,     \
+                     thresh, invk, s0, s1)
+#define PICK_FWD(DD, CC) do { if (drop) LAUNCH_FWD(DD, CC, true); \
+                              else LAUNCH_FWD(DD, CC, false); } while (0)
+  if (D == 128) { if (causal) PICK_FWD(128, true); else PICK_FWD(128, false); }
+  else          { if (causal) PICK_FWD(64, true);  else PICK_FWD(64, false); }
+#undef PICK_FWD
 #undef LAUNCH_FWD
 }
 
 void launch_bwd(Strided q, Strided k, Strided v, Strided dout, Strided o,
                 StridedMut dq, StridedMut dk, StridedMut dv, float* lse,
                 float* delta, int B, int H, int S, int D, float scale,
-                bool causal) {
+                bool causal, float p_drop = 0.f,
+                unsigned long long seed = 0) {
   auto stream = at::hip::getCurrentHIPStream();
   long rows = (long)B * H * S;
   hipLaunchKernelGGL(attn_bwd_delta_kernel,
@@ -1068,17 +1135,26 @@ void launch_bwd(Strided q, Strided k, Strided v, Strided dout, Strided o,
   int blocks256 = (S + QTILE * 2 - 1) / (QTILE * 2);  // dq RB=2 home tiles
   dim3 grid(B * H, blocks128);
   dim3 grid_dq(B * H, blocks256);
-#define LAUNCH_BWD(DD, CC)                                                   \
+  unsigned thresh = (unsigned)(p_drop * 4294967296.0);
+  float invk = 1.f / (1.f - p_drop);
+  unsigned s0 = (unsigned)seed, s1 = (unsigned)(seed >> 32);
+  bool drop = p_drop > 0.f;
+#define LAUNCH_BWD(DD, CC, PP)                                               \
   do {                                                                       \
-    hipLaunchKernelGGL((attn_bwd_dkv_kernel<DD, CC>), grid,                  \
+    hipLaunchKernelGGL((attn_bwd_dkv_kernel<DD, CC, PP>), grid,              \
                        dim3(FW_BLOCKT), 0, stream, q, k, v, dout, lse,       \
-                       delta, dk, dv, H, S, scale, tiles64);                 \
-    hipLaunchKernelGGL((attn_bwd_dq_kernel<DD, CC>), grid_dq,                \
+                       delta, dk, dv, H, S, scale, tiles64, thresh, invk,    \
+                       s0, s1);                                              \
+    hipLaunchKernelGGL((attn_bwd_dq_kernel<DD, CC, PP>), grid_dq,            \
                        dim3(FW_BLOCKT), 0, stream, q, k, v, dout, lse,       \
-                       delta, dq, H, S, scale, tiles64);                     \
+                       delta, dq, H, S, scale, tiles64, thresh, invk,        \
+                       s0, s1);                                              \
   } while (0)
-  if (D == 128) { if (causal) LAUNCH_BWD(128, true); else LAUNCH_BWD(128, false); }
-  else          { if (causal) LAUNCH_BWD(64, true);  else LAUNCH_BWD(64, false); }
+#define PICK_BWD(DD, CC) do { if (drop) LAUNCH_BWD(DD, CC, true); \
+                              else LAUNCH_BWD(DD, CC, false); } while (0)
+  if (D == 128) { if (causal) PICK_BWD(128, true); else PICK_BWD(128, false); }
+  else          { if (causal) PICK_BWD(64, true);  else PICK_BWD(64, false); }
+#undef PICK_BWD
 #undef LAUNCH_BWD
 }
 
@@ -1091,7 +1167,8 @@ void launch_bwd(Strided q, Strided k, Strided v, Strided dout, Strided o,
 // q,k,v: [B, H, S, D] contiguous (layout-compat path)
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, bool causal,
-                                    double scale) {
+                                    double scale, double p_drop,
+                                    long seed) {
   TORCH_CHECK(q.dim() == 4 && q.is_contiguous());
   int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
   TORCH_CHECK(k.size(2) == S, "q and k seq length must match");
@@ -1101,14 +1178,16 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
   auto lse = torch::empty({B, H, S}, q.options().dtype(torch::kFloat));
   launch_fwd(strided_of(q, 0, 1, 2), strided_of(k, 0, 1, 2),
              strided_of(v, 0, 1, 2), strided_mut_of(o, 0, 1, 2),
-             lse.data_ptr<float>(), B, H, S, D, (float)scale, causal);
+             lse.data_ptr<float>(), B, H, S, D, (float)scale, causal,
+             (float)p_drop, (unsigned long long)seed);
   return {o, lse};
 }
 
 std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     torch::Tensor k, torch::Tensor v,
                                     torch::Tensor o, torch::Tensor lse,
-                                    bool causal, double scale) {
+                                    bool causal, double scale,
+                                    double p_drop, long seed) {
   int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
   auto dq = torch::empty_like(q);
   auto dk = torch::empty_like(k);
@@ -1120,14 +1199,16 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
              strided_of(o, 0, 1, 2), strided_mut_of(dq, 0, 1, 2),
              strided_mut_of(dk, 0, 1, 2), strided_mut_of(dv, 0, 1, 2),
              lse.data_ptr<float>(), delta.data_ptr<float>(), B, H, S, D,
-             (float)scale, causal);
+             (float)scale, causal, (float)p_drop,
+             (unsigned long long)seed);
   return {dq, dk, dv};
 }
 
 // Packed path: qkv [B, S, Hh, 3, D] (the fused-QKV linear output viewed);
 // returns o [B, S, Hh*D] + lse [B, Hh, S]. Zero layout copies.
 std::vector<torch::Tensor> attn_fwd_packed(torch::Tensor qkv, long num_heads,
-                                           double scale) {
+                                           double scale, double p_drop,
+                                           long seed) {
   TORCH_CHECK(qkv.dim() == 5 && qkv.is_contiguous());
   int B = qkv.size(0), S = qkv.size(1), H = qkv.size(2), D = qkv.size(4);
   TORCH_CHECK(qkv.size(3) == 3 && H == num_heads);
@@ -1142,14 +1223,15 @@ std::vector<torch::Tensor> attn_fwd_packed(torch::Tensor qkv, long num_heads,
   StridedMut om{(__hip_bfloat16*)o.data_ptr(), (long)S * H * D, (long)D,
                 (long)H * D};
   launch_fwd(q, k, v, om, lse.data_ptr<float>(), B, H, S, D, (float)scale,
-             /*causal=*/true);
+             /*causal=*/true, (float)p_drop, (unsigned long long)seed);
   return {o, lse};
 }
 
 // dout [B, S, Hh*D]; returns dqkv [B, S, Hh, 3, D]
 torch::Tensor attn_bwd_packed(torch::Tensor dout, torch::Tensor qkv,
                               torch::Tensor o, torch::Tensor lse,
-                              long num_heads, double scale) {
+                              long num_heads, double scale, double p_drop,
+                              long seed) {
   int B = qkv.size(0), S = qkv.size(1), H = qkv.size(2), D = qkv.size(4);
   auto dqkv = torch::empty_like(qkv);
   auto delta = torch::empty({B, H, S}, qkv.options().dtype(torch::kFloat));
@@ -1168,6 +1250,6 @@ torch::Tensor attn_bwd_packed(torch::Tensor dout, torch::Tensor qkv,
   StridedMut dvm = dqm; dvm.p += 2 * D;
   launch_bwd(q, k, v, dos, os, dqm, dkm, dvm, lse.data_ptr<float>(),
              delta.data_ptr<float>(), B, H, S, D, (float)scale,
-             /*causal=*/true);
+             /*causal=*/true, (float)p_drop, (unsigned long long)seed);
   return dqkv;
 }

@@ -54,16 +54,19 @@ torch::Tensor vp_ce_bwd(torch::Tensor dloss, torch::Tensor logits,
 // attention.hip
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, bool causal,
-                                    double scale);
+                                    double scale, double p_drop, long seed);
 std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     torch::Tensor k, torch::Tensor v,
                                     torch::Tensor o, torch::Tensor lse,
-                                    bool causal, double scale);
+                                    bool causal, double scale,
+                                    double p_drop, long seed);
 std::vector<torch::Tensor> attn_fwd_packed(torch::Tensor qkv, long num_heads,
-                                           double scale);
+                                           double scale, double p_drop,
+                                           long seed);
 torch::Tensor attn_bwd_packed(torch::Tensor dout, torch::Tensor qkv,
                               torch::Tensor o, torch::Tensor lse,
-                              long num_heads, double scale);
+                              long num_heads, double scale, double p_drop,
+                              long seed);
 // topp.hip
 std::vector<torch::Tensor> topp_select(torch::Tensor sorted_p,
                                        torch::Tensor sorted_idx,
@@ -102,12 +105,22 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("vp_ce_bwd", &vp_ce_bwd, "vocab-parallel CE bwd");
   m.def("moe_dispatch", &moe_dispatch,
         "fused MoE token dispatch (histogram+rank+scatter)");
-  m.def("attn_fwd", &attn_fwd, "flash attention fwd (MFMA, causal)");
-  m.def("attn_bwd", &attn_bwd, "flash attention bwd (MFMA)");
+  m.def("attn_fwd", &attn_fwd, "flash attention fwd (MFMA, causal)",
+        py::arg("q"), py::arg("k"), py::arg("v"), py::arg("causal"),
+        py::arg("scale"), py::arg("p_drop") = 0.0, py::arg("seed") = 0);
+  m.def("attn_bwd", &attn_bwd, "flash attention bwd (MFMA)",
+        py::arg("dout"), py::arg("q"), py::arg("k"), py::arg("v"),
+        py::arg("o"), py::arg("lse"), py::arg("causal"), py::arg("scale"),
+        py::arg("p_drop") = 0.0, py::arg("seed") = 0);
   m.def("attn_fwd_packed", &attn_fwd_packed,
-        "flash attention fwd on packed QKV (no layout copies)");
+        "flash attention fwd on packed QKV (no layout copies)",
+        py::arg("qkv"), py::arg("num_heads"), py::arg("scale"),
+        py::arg("p_drop") = 0.0, py::arg("seed") = 0);
   m.def("attn_bwd_packed", &attn_bwd_packed,
-        "flash attention bwd writing packed dQKV");
+        "flash attention bwd writing packed dQKV",
+        py::arg("dout"), py::arg("qkv"), py::arg("o"), py::arg("lse"),
+        py::arg("num_heads"), py::arg("scale"),
+        py::arg("p_drop") = 0.0, py::arg("seed") = 0);
   m.def("topp_select", &topp_select, "top-p nucleus cutoff + draw");
   m.def("mfma_gemm16_probe", &mfma_gemm16_probe,
         "debug: 16x16x32 MFMA fragment-layout probe");

@@ -114,6 +114,27 @@ DEV_INLINE float block_reduce_max(float x, float* lds_scratch) {
   return r;
 }
 
+// ---- Philox4x32-10 counter-based RNG (dropout; torch's generator) ----
+struct philox4 { unsigned x, y, z, w; };
+
+DEV_INLINE philox4 philox4x32_10(unsigned c0, unsigned c1, unsigned c2,
+                                 unsigned c3, unsigned k0, unsigned k1) {
+  const unsigned M0 = 0xD2511F53u, M1 = 0xCD9E8D57u;
+  const unsigned W0 = 0x9E3779B9u, W1 = 0xBB67AE85u;
+#pragma unroll
+  for (int i = 0; i < 10; ++i) {
+    unsigned hi0 = __umulhi(M0, c0), lo0 = M0 * c0;
+    unsigned hi1 = __umulhi(M1, c2), lo1 = M1 * c2;
+    unsigned n0 = hi1 ^ c1 ^ k0;
+    unsigned n1 = lo1;
+    unsigned n2 = hi0 ^ c3 ^ k1;
+    unsigned n3 = lo0;
+    c0 = n0; c1 = n1; c2 = n2; c3 = n3;
+    k0 += W0; k1 += W1;
+  }
+  return {c0, c1, c2, c3};
+}
+
 // ---- width-generic vector load/store (bf16 raw-short or fp32), VEC 4/8 ----
 template <typename T, int VEC>
 DEV_INLINE void vload(const T* p, float* v) {

@@ -113,14 +113,20 @@ class MultiHeadAttention(nn.Module):
                 ca = UlyssesAttention(scale=self.scale, causal=True)
             o = ca(q, k, v).reshape(B, Sl, -1)
             return self.out_proj(o), None
-        if (self.fused_attn and self.attn_dropout_p == 0.0 and cache is None
-                and not use_cache and not self.sequence_parallel):
+        p_att = self.attn_dropout_p if self.training else 0.0
+        if (self.fused_attn and cache is None and not use_cache
+                and not self.sequence_parallel
+                and (p_att == 0.0 or qkv.is_cuda)):
             # packed fast path: kernel reads the fused-QKV linear output
-            # directly and writes [B, S, h*D] — zero layout copies
+            # directly and writes [B, S, h*D] — zero layout copies.
+            # Attention dropout runs IN the kernel (Philox mask keyed by
+            # the per-rank local seed via model_parallel_rng — reference
+            # hybrid_model.py:328 RNG-tracker dropout)
             B, S, _ = qkv.shape
             packed = qkv.view(B, S, self.num_heads_local, 3, self.head_dim)
-            o = flash_attention_packed(packed, self.num_heads_local,
-                                       scale=self.scale)
+            with model_parallel_rng():
+                o = flash_attention_packed(packed, self.num_heads_local,
+                                           scale=self.scale, p_drop=p_att)
             return self.out_proj(o), None
         if self.sequence_parallel:
             # x: [s/mp, B, H]; qkv allgathered the seq dim -> [S, B, 3H/mp]
@@ -144,9 +150,11 @@ class MultiHeadAttention(nn.Module):
             v = torch.cat([cache[1], v], dim=2)
         new_cache = (k, v) if use_cache else None
 
-        if self.fused_attn and self.attn_dropout_p == 0.0 and not (
-                cache is not None and S == 1):
-            o = flash_attention(q, k, v, causal=True, scale=self.scale)
+        if self.fused_attn and not (cache is not None and S == 1) and (
+                p_att == 0.0 or q.is_cuda):
+            with model_parallel_rng():
+                o = flash_attention(q, k, v, causal=True, scale=self.scale,
+                                    p_drop=p_att)
         else:
             scores = torch.matmul(q, k.transpose(-1, -2))
             probs = fused_softmax_causal(scores, self.scale)

@@ -207,15 +207,20 @@ def bias_gelu(x, bias=None):
 
 class _FlashAttnFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, q, k, v, causal, scale):
+    def forward(ctx, q, k, v, causal, scale, p_drop, seed):
         scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
         q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
         if use_hip(q):
-            o, lse = hip_ext().attn_fwd(q, k, v, causal, scale)
+            o, lse = hip_ext().attn_fwd(q, k, v, causal, scale,
+                                        p_drop, seed)
         else:
+            # CPU reference has no Philox twin; dropout handled by the
+            # caller's torch path there
+            assert p_drop == 0.0, "CPU path: dropout not fused"
             o, lse = ref.attention_fwd(q, k, v, causal, scale)
         ctx.save_for_backward(q, k, v, o, lse)
         ctx.causal, ctx.scale = causal, scale
+        ctx.p_drop, ctx.seed = p_drop, seed
         return o
 
     @staticmethod
@@ -224,16 +229,26 @@ class _FlashAttnFn(torch.autograd.Function):
         do = do.contiguous()
         if use_hip(q):
             dq, dk, dv = hip_ext().attn_bwd(do, q, k, v, o, lse,
-                                            ctx.causal, ctx.scale)
+                                            ctx.causal, ctx.scale,
+                                            ctx.p_drop, ctx.seed)
         else:
             dq, dk, dv = ref.attention_bwd(do, q, k, v, o, lse,
                                            ctx.causal, ctx.scale)
-        return dq, dk, dv, None, None
+        return dq, dk, dv, None, None, None, None
 
 
-def flash_attention(q, k, v, causal: bool = True, scale: Optional[float] = None):
-    """q,k,v: [B, H, S, D] -> o: [B, H, S, D]. O(S) memory, online softmax."""
-    return _FlashAttnFn.apply(q, k, v, causal, scale)
+def _drop_seed() -> int:
+    # drawn from the torch CPU generator so the per-rank local_seed
+    # discipline (parallel/rng.py model_parallel_rng) governs the mask
+    return int(torch.randint(0, 2 ** 62, (1,)).item())
+
+
+def flash_attention(q, k, v, causal: bool = True,
+                    scale: Optional[float] = None, p_drop: float = 0.0):
+    """q,k,v: [B, H, S, D] -> o: [B, H, S, D]. O(S) memory, online
+    softmax; in-kernel Philox attention dropout when p_drop > 0."""
+    seed = _drop_seed() if p_drop > 0 else 0
+    return _FlashAttnFn.apply(q, k, v, causal, scale, p_drop, seed)
 
 
 class _FlashAttnPackedFn(torch.autograd.Function):
@@ -241,10 +256,11 @@ class _FlashAttnPackedFn(torch.autograd.Function):
     output viewed), output [B, S, h*D] — no split/transpose/cat copies."""
 
     @staticmethod
-    def forward(ctx, qkv, num_heads, scale):
+    def forward(ctx, qkv, num_heads, scale, p_drop, seed):
         scale = scale if scale is not None else 1.0 / math.sqrt(qkv.shape[-1])
         if use_hip(qkv):
-            o, lse = hip_ext().attn_fwd_packed(qkv, num_heads, scale)
+            o, lse = hip_ext().attn_fwd_packed(qkv, num_heads, scale,
+                                               p_drop, seed)
         else:
             B, S, h, _, D = qkv.shape
             q = qkv[:, :, :, 0].permute(0, 2, 1, 3)
@@ -254,6 +270,7 @@ class _FlashAttnPackedFn(torch.autograd.Function):
             o = o4.permute(0, 2, 1, 3).reshape(B, S, h * D)
         ctx.save_for_backward(qkv, o, lse)
         ctx.num_heads, ctx.scale = num_heads, scale
+        ctx.p_drop, ctx.seed = p_drop, seed
         return o
 
     @staticmethod
@@ -261,7 +278,8 @@ class _FlashAttnPackedFn(torch.autograd.Function):
         qkv, o, lse = ctx.saved_tensors
         if use_hip(qkv):
             dqkv = hip_ext().attn_bwd_packed(do.contiguous(), qkv, o, lse,
-                                             ctx.num_heads, ctx.scale)
+                                             ctx.num_heads, ctx.scale,
+                                             ctx.p_drop, ctx.seed)
         else:
             B, S, h, _, D = qkv.shape
             q = qkv[:, :, :, 0].permute(0, 2, 1, 3)
@@ -274,12 +292,19 @@ class _FlashAttnPackedFn(torch.autograd.Function):
             dqkv = torch.stack(
                 (dq.permute(0, 2, 1, 3), dk.permute(0, 2, 1, 3),
                  dv.permute(0, 2, 1, 3)), dim=3)
-        return dqkv, None, None
+        return dqkv, None, None, None, None
 
 
-def flash_attention_packed(qkv, num_heads: int, scale: Optional[float] = None):
-    """qkv [B, S, h, 3, D] -> o [B, S, h*D]; causal."""
-    return _FlashAttnPackedFn.apply(qkv.contiguous(), num_heads, scale)
+def flash_attention_packed(qkv, num_heads: int,
+                           scale: Optional[float] = None,
+                           p_drop: float = 0.0):
+    """qkv [B, S, h, 3, D] -> o [B, S, h*D]; causal; in-kernel Philox
+    attention dropout when p_drop > 0 (GPU only)."""
+    if p_drop > 0:
+        assert qkv.is_cuda, "fused attention dropout is the GPU path"
+    seed = _drop_seed() if p_drop > 0 else 0
+    return _FlashAttnPackedFn.apply(qkv.contiguous(), num_heads, scale,
+                                    p_drop, seed)
 
 
 # ---------------------------------------------------------------------------

@@ -281,6 +281,60 @@ def test_flash_attention_defer_max_spike(dev):
     assert torch.allclose(lse[m], lse2[m], atol=1e-2, rtol=1e-3)
 
 
+def test_flash_attention_dropout(dev):
+    """In-kernel Philox attention dropout: (a) the mask is deterministic
+    per seed and value-independent, so it can be EXTRACTED with a
+    zeros-Q/K + one-hot-V construction; (b) drop rate ~ p; (c) fwd+bwd
+    on random data match a torch reference using the same mask."""
+    B, H, S, D = 1, 2, 64, 64
+    p, seed = 0.3, 42
+    scale = D ** -0.5
+    # --- mask extraction: scores all equal -> P uniform; one-hot V
+    qz = torch.zeros(B, H, S, D, device=dev, dtype=torch.bfloat16)
+    v_eye = torch.eye(S, device=dev, dtype=torch.bfloat16) \
+        .view(1, 1, S, D).expand(B, H, S, D).contiguous()
+    o_d, _ = hip_ext().attn_fwd(qz, qz, v_eye, True, scale, p, seed)
+    valid = torch.arange(1, S + 1, device=dev, dtype=torch.float32)
+    mask = (o_d.float() * valid[None, None, :, None] * (1 - p))
+    mask = mask.round().clamp(0, 1)  # [B, H, q, kv] in the one-hot basis
+    tril = torch.tril(torch.ones(S, S, device=dev)).view(1, 1, S, S)
+    keep_frac = (mask * tril).sum() / tril.sum() / 1.0
+    assert abs(float(keep_frac) - (1 - p)) < 0.06, float(keep_frac)
+    # determinism per seed / variation across seeds
+    o_d2, _ = hip_ext().attn_fwd(qz, qz, v_eye, True, scale, p, seed)
+    assert torch.equal(o_d, o_d2)
+    o_d3, _ = hip_ext().attn_fwd(qz, qz, v_eye, True, scale, p, seed + 1)
+    assert not torch.equal(o_d, o_d3)
+
+    # --- random data, same (shape, seed) -> same mask; compare to torch
+    torch.manual_seed(0)
+    q = torch.randn(B, H, S, D, device=dev, dtype=torch.bfloat16)
+    k = torch.randn_like(q)
+    v = torch.randn_like(q)
+    do = torch.randn_like(q)
+    o, lse = hip_ext().attn_fwd(q, k, v, True, scale, p, seed)
+    dq, dk, dv = hip_ext().attn_bwd(do, q, k, v, o, lse, True, scale,
+                                    p, seed)
+
+    qf = q.float().requires_grad_(True)
+    kf = k.float().requires_grad_(True)
+    vf = v.float().requires_grad_(True)
+    s = torch.matmul(qf, kf.transpose(-1, -2)) * scale
+    s = s.masked_fill(tril == 0, float("-inf"))
+    a = s.softmax(-1)
+    d_ = a * mask / (1 - p)
+    o_ref = torch.matmul(d_, vf)
+    o_ref.backward(do.float())
+    assert torch.allclose(o.float(), o_ref, atol=5e-2, rtol=5e-2), \
+        (o.float() - o_ref).abs().max()
+    assert torch.allclose(dq.float(), qf.grad, atol=8e-2, rtol=8e-2), \
+        (dq.float() - qf.grad).abs().max()
+    assert torch.allclose(dk.float(), kf.grad, atol=8e-2, rtol=8e-2), \
+        (dk.float() - kf.grad).abs().max()
+    assert torch.allclose(dv.float(), vf.grad, atol=8e-2, rtol=8e-2), \
+        (dv.float() - vf.grad).abs().max()
+
+
 def test_topp_sampling(dev):
     B, V = 8, 50304
     logits = torch.randn(B, V, device=dev)

@@ -74,13 +74,33 @@ def test_zero3_175b_scale_model_memory():
     total_96 = resident_96 + transient
     hbm = torch.cuda.get_device_properties(0).total_memory
     print(f"[sizing] resident/4L={resident/2**30:.1f} GiB, "
-          f"peak={peak/2**30:.1f} GiB, 96L projection="
+          f"peak={peak/2**30:.1f} GiB, 96L on-device projection="
           f"{total_96/2**30:.1f} GiB of {hbm/2**30:.0f} GiB")
-    assert total_96 < hbm * 0.9, (total_96 / 2 ** 30, hbm / 2 ** 30)
+    # MEASURED sizing fact: with fp32 optimizer states ON DEVICE the
+    # 96-layer projection (~340 GiB) exceeds 288 GB HBM — the 175B
+    # single-node config REQUIRES the CPU-offload path below (or a
+    # second sharding axis across nodes)
+    assert total_96 > hbm, "sizing assumption changed — revisit"
     # storage really released after the step
     for u in wrapped.units:
         if u.name != "<rest>":
             assert not u.live
+
+    # ---- offload variant: fp32 master/m/v in pinned host memory ----
+    del opt
+    torch.cuda.empty_cache()
+    torch.cuda.reset_peak_memory_stats()
+    opt2 = Stage3AdamW(wrapped, lr=1e-4, offload=True)
+    resident_off = torch.cuda.memory_allocated()
+    y = wrapped(x)
+    y.float().pow(2).mean().backward()
+    opt2.reduce_and_step()
+    peak_off = torch.cuda.max_memory_allocated()
+    total_96_off = resident_off / L * 96 + (peak_off - resident_off)
+    print(f"[sizing+offload] resident/4L={resident_off/2**30:.1f} GiB, "
+          f"peak={peak_off/2**30:.1f} GiB, 96L projection="
+          f"{total_96_off/2**30:.1f} GiB of {hbm/2**30:.0f} GiB")
+    assert total_96_off < hbm * 0.75,         (total_96_off / 2 ** 30, hbm / 2 ** 30)
 
 
 def test_zero3_prefetch_matches_sync():
