@@ -34,6 +34,10 @@ def parse_args():
     p.add_argument("--micro-batch", type=int, default=None)
     p.add_argument("--acc-steps", type=int, default=None)
     p.add_argument("--seq-len", type=int, default=1024)
+    p.add_argument("--dropout", type=float, default=0.0,
+                   help="hidden+attention dropout (the reference's "
+                        "pretrain configs run 0.1; in-kernel Philox "
+                        "attention dropout on the flash path)")
     return p.parse_args()
 
 
@@ -86,8 +90,8 @@ def main():
         f"Model.num_layers={shape['num_layers']}",
         f"Model.num_attention_heads={shape['num_attention_heads']}",
         f"Model.max_position_embeddings={seq}",
-        "Model.hidden_dropout_prob=0.0",
-        "Model.attention_probs_dropout_prob=0.0",
+        f"Model.hidden_dropout_prob={args.dropout}",
+        f"Model.attention_probs_dropout_prob={args.dropout}",
         f"Global.micro_batch_size={micro}",
         f"Global.local_batch_size={local_bs}",
         "Global.eval_freq=", "Global.save_steps=",
@@ -150,6 +154,7 @@ def main():
             "vs_baseline": None,
             "dtype": "bf16",
             "data": "synthetic",
+            "dropout": args.dropout,
             "config": {
                 "model": args.model, "global_batch": global_batch,
                 "seq_len": seq, "micro_batch": micro,
