@@ -298,7 +298,7 @@ def test_flash_attention_dropout(dev):
     mask = (o_d.float() * valid[None, None, :, None] * (1 - p))
     mask = mask.round().clamp(0, 1)  # [B, H, q, kv] in the one-hot basis
     tril = torch.tril(torch.ones(S, S, device=dev)).view(1, 1, S, S)
-    keep_frac = (mask * tril).sum() / tril.sum() / 1.0
+    keep_frac = (mask * tril).sum() / (tril.sum() * B * H)
     assert abs(float(keep_frac) - (1 - p)) < 0.06, float(keep_frac)
     # determinism per seed / variation across seeds
     o_d2, _ = hip_ext().attn_fwd(qz, qz, v_eye, True, scale, p, seed)
