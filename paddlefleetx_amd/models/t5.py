@@ -272,3 +272,59 @@ class T5Model(nn.Module):
                            attention_mask=decoder_attention_mask,
                            enc_attention_mask=attention_mask)
         return self.lm_head(dec * (self.config.d_model ** -0.5))
+
+
+class T5ForConditionalGeneration(nn.Module):
+    """Seq2seq head with teacher-forced loss + greedy decode
+    (reference t5/modeling.py T5ForConditionalGeneration surface)."""
+
+    def __init__(self, decoder_start_token_id: int = 0, eos_token_id: int = 1,
+                 **kwargs):
+        super().__init__()
+        self.model = T5Model(**kwargs)
+        self.decoder_start_token_id = decoder_start_token_id
+        self.eos_token_id = eos_token_id
+
+    @property
+    def config(self):
+        return self.model.config
+
+    def _shift_right(self, labels):
+        start = torch.full_like(labels[:, :1], self.decoder_start_token_id)
+        return torch.cat([start, labels[:, :-1]], dim=1)
+
+    def forward(self, input_ids, labels=None, decoder_input_ids=None,
+                attention_mask=None):
+        if decoder_input_ids is None:
+            assert labels is not None
+            decoder_input_ids = self._shift_right(labels)
+        logits = self.model(input_ids, decoder_input_ids,
+                            attention_mask=attention_mask)
+        if labels is None:
+            return logits
+        loss = F.cross_entropy(logits.float().flatten(0, 1),
+                               labels.flatten(), ignore_index=-100)
+        return loss, logits
+
+    @torch.no_grad()
+    def generate(self, input_ids, max_length: int = 20,
+                 attention_mask=None):
+        """Greedy decode (the reference's default strategy)."""
+        enc = self.model.encoder(input_ids, attention_mask=attention_mask)
+        B = input_ids.shape[0]
+        dec = torch.full((B, 1), self.decoder_start_token_id,
+                         dtype=torch.long, device=input_ids.device)
+        finished = torch.zeros(B, dtype=torch.bool, device=input_ids.device)
+        for _ in range(max_length):
+            out = self.model.decoder(dec, enc_out=enc,
+                                     enc_attention_mask=attention_mask)
+            logits = self.model.lm_head(
+                out[:, -1] * (self.config.d_model ** -0.5))
+            nxt = logits.argmax(-1, keepdim=True)
+            nxt = torch.where(finished.unsqueeze(1),
+                              torch.full_like(nxt, self.eos_token_id), nxt)
+            dec = torch.cat([dec, nxt], dim=1)
+            finished |= nxt.squeeze(1) == self.eos_token_id
+            if bool(finished.all()):
+                break
+        return dec

@@ -180,3 +180,81 @@ def _ep_worker(rank, world, port):
 @pytest.mark.timeout(300)
 def test_moe_ep2_matches_single_rank():
     _run(_ep_worker, 2)
+
+
+# ---------------------------------------------------------------------------
+# Second (einsum-dispatch, DeepSpeed-style) MoE surface
+# (reference moe_exp/sharded_moe.py:87-379)
+# ---------------------------------------------------------------------------
+
+def test_top1gating_capacity_and_aux():
+    import torch
+    from paddlefleetx_amd.models.moe.sharded_moe import top1gating
+    torch.manual_seed(0)
+    logits = torch.randn(32, 4)
+    aux, combine, dispatch, counts = top1gating(logits, capacity_factor=1.0)
+    assert combine.shape == (32, 4, max(4, 32 // 4))
+    assert torch.isfinite(aux) and aux > 0
+    # each token occupies at most one (expert, slot)
+    assert (dispatch.sum(dim=(1, 2)) <= 1).all()
+    # no slot is double-booked
+    assert (dispatch.sum(dim=0) <= 1).all()
+    # combine weights equal the routed gate prob
+    probs = logits.softmax(-1)
+    routed = combine.sum(dim=(1, 2))
+    kept = dispatch.sum(dim=(1, 2)) > 0
+    assert torch.allclose(routed[kept],
+                          probs.max(-1).values[kept], atol=1e-6)
+
+
+def test_top2gating_normalized_weights():
+    import torch
+    from paddlefleetx_amd.models.moe.sharded_moe import top2gating
+    torch.manual_seed(1)
+    logits = torch.randn(64, 8)
+    aux, combine, dispatch, counts = top2gating(logits, capacity_factor=2.0)
+    routed = combine.sum(dim=(1, 2))
+    two_kept = dispatch.sum(dim=(1, 2)) == 2
+    # tokens keeping both experts have weights normalized to 1
+    assert torch.allclose(routed[two_kept], torch.ones(int(two_kept.sum())),
+                          atol=1e-5)
+
+
+def test_sharded_moe_layer_forward_backward():
+    import torch
+    from paddlefleetx_amd.models.moe.sharded_moe import ShardedMoELayer
+    from paddlefleetx_amd.parallel.env import set_hcg
+    from paddlefleetx_amd.parallel.topology import HybridTopology
+    import torch.distributed as dist
+    if not dist.is_initialized():
+        set_hcg(HybridTopology())
+    torch.manual_seed(2)
+    layer = ShardedMoELayer(16, 32, num_experts=4, k=2, capacity_factor=2.0)
+    x = torch.randn(2, 10, 16, requires_grad=True)
+    y = layer(x)
+    assert y.shape == x.shape
+    (y.sum() + layer.last_aux_loss).backward()
+    assert x.grad is not None and torch.isfinite(x.grad).all()
+    assert layer.gate.wg.weight.grad is not None
+
+
+def test_sharded_moe_matches_dense_expert_mixture():
+    """With capacity ample and k=1, the einsum layer equals routing each
+    token through its argmax expert weighted by its gate prob."""
+    import torch
+    from paddlefleetx_amd.models.moe.sharded_moe import ShardedMoELayer
+    torch.manual_seed(3)
+    layer = ShardedMoELayer(8, 16, num_experts=2, k=1, capacity_factor=8.0)
+    layer.eval()
+    x = torch.randn(1, 12, 8)
+    y = layer(x)
+    xf = x.reshape(-1, 8)
+    logits = layer.gate.wg(xf.float())
+    probs = logits.softmax(-1)
+    idx = logits.argmax(-1)
+    want = torch.zeros_like(xf)
+    for t in range(12):
+        e = int(idx[t])
+        want[t] = layer.experts[e](xf[t:t + 1])[0] * probs[t, e]
+    assert torch.allclose(y.reshape(-1, 8), want, atol=1e-5), \
+        (y.reshape(-1, 8) - want).abs().max()
