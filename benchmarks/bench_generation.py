@@ -74,8 +74,9 @@ def main():
             if dev == "cuda":
                 torch.cuda.synchronize()
         dt = (time.time() - t0) / reps
-        ntok = out.shape[1] - args.prompt_len if out.ndim == 2 \
-            else args.gen_len
+        # GPTForGeneration returns only the GENERATED ids
+        ntok = out.shape[1] if out.shape[1] <= args.gen_len \
+            else out.shape[1] - args.prompt_len
         ms_tok = dt * 1000.0 / max(1, ntok)
         print(f"bs={bs:3d}: {dt*1000:8.1f} ms/gen ({ntok} new tok) = "
               f"{ms_tok:7.2f} ms/token, {bs*ntok/dt:8.0f} tok/s")
