@@ -140,6 +140,21 @@ class AutoEngine:
             module = build_module(self.configs)
         from paddlefleetx_amd.core.engine import EagerEngine
         self._engine = EagerEngine(self.configs, module, mode=mode)
+        # attach shard_tensor annotations matching the planned mesh and
+        # verify them against the constructed layout (the reference's
+        # auto model carries these inline, auto_model.py:92-713)
+        try:
+            from paddlefleetx_amd.parallel.auto_shard import (
+                ProcessMesh, annotate_gpt, validate_against_topology)
+            mesh = ProcessMesh([plan["dp_degree"], plan["mp_degree"],
+                                plan["pp_degree"]], ("dp", "mp", "pp"))
+            annotate_gpt(module.model, mesh)
+            problems = validate_against_topology(module.model)
+            assert not problems, problems
+            self.mesh = mesh
+        except Exception as e:  # non-GPT families: annotations optional
+            logger.warning(f"auto-shard annotations skipped: {e}")
+            self.mesh = None
 
     def __getattr__(self, name):
         return getattr(self._engine, name)
