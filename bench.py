@@ -141,8 +141,11 @@ def main():
     ms_per_step = elapsed / args.steps * 1000.0
 
     if rank == 0:
+        metric = "tokens/sec/GPU GPT-3-6.7B pretrain, DP2×TP2×PP2 at 1/2/4/8 MI355X"
+        if args.model != "GPT-6.7B":
+            metric = metric.replace("GPT-3-6.7B", args.model)
         out = {
-            "metric": "tokens/sec/GPU GPT-3-6.7B pretrain, DP2×TP2×PP2 at 1/2/4/8 MI355X",
+            "metric": metric,
             "value": round(toks_per_s, 1),
             "unit": "tokens/s",
             "n_gpus": n,

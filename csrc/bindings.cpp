@@ -32,6 +32,7 @@ void adamw_flat(torch::Tensor master, torch::Tensor grad, torch::Tensor m,
                 double beta2, double eps, double wd, long step);
 torch::Tensor rope_fwd(torch::Tensor x, torch::Tensor cos, torch::Tensor sin);
 torch::Tensor colsum(torch::Tensor x);
+void grad_sumsq(torch::Tensor x, torch::Tensor out, long slot);
 // softmax.hip
 torch::Tensor softmax_causal_fwd(torch::Tensor s, double scale);
 torch::Tensor softmax_causal_bwd(torch::Tensor dy, torch::Tensor y,
@@ -88,6 +89,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_bwd_residual", &layernorm_bwd_residual,
         "LN bwd with fused residual-grad addend");
   m.def("colsum", &colsum, "column sum [N,H] -> fp32 [H] (dbias)");
+  m.def("grad_sumsq", &grad_sumsq, "fused sumsq (found_inf + grad norm)");
   m.def("bias_gelu_fwd", &bias_gelu_fwd, "fused bias+gelu fwd");
   m.def("bias_gelu_bwd", &bias_gelu_bwd, "fused bias+gelu bwd");
   m.def("adamw_flat", &adamw_flat, "fused AdamW on flat buffers");

@@ -96,6 +96,27 @@ __global__ __launch_bounds__(BLOCK) void colsum_kernel(
   for (int j = 0; j < CV; ++j) atomicAdd(&out[col + j], s[j]);
 }
 
+// ---- fused grad sumsq (found_inf + global-norm in ONE pass) ----
+// out[slot] += sum(x^2) in fp32; NaN/Inf grads propagate into the sum,
+// so !isfinite(out) IS the found_inf flag (reference
+// check_finite_and_unscale, distributed/apis/amp.py:212-216)
+template <typename T>
+__global__ void grad_sumsq_kernel(const T* __restrict__ x, long n,
+                                  float* __restrict__ out, long slot) {
+  __shared__ float sred[BLOCK / WAVE];
+  float s = 0.f;
+  constexpr int V = 4;
+  for (long i = ((long)blockIdx.x * BLOCK + threadIdx.x) * V; i < n;
+       i += (long)gridDim.x * BLOCK * V) {
+    float v[V];
+    vload<T, V>(x + i, v);
+#pragma unroll
+    for (int j = 0; j < V; ++j) s += v[j] * v[j];
+  }
+  s = block_reduce_sum<BLOCK>(s, sred);
+  if (threadIdx.x == 0) atomicAdd(&out[slot], s);
+}
+
 void launch_colsum(const void* x, float* out, long N, int H, bool bf16,
                    hipStream_t stream) {
   constexpr int CV = 4;
@@ -275,6 +296,25 @@ std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
                   x.scalar_type() == torch::kBFloat16, stream);
   }
   return {dx, db};
+}
+
+// accumulate sum(x^2) into out[slot] (fp32). x flat, numel % 4 == 0.
+void grad_sumsq(torch::Tensor x, torch::Tensor out, long slot) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  TORCH_CHECK(out.scalar_type() == torch::kFloat);
+  long n = x.numel();
+  TORCH_CHECK(n % 4 == 0);
+  auto stream = at::hip::getCurrentHIPStream();
+  dim3 grid(grid_for(n, BLOCK * 4));
+  if (x.scalar_type() == torch::kBFloat16)
+    hipLaunchKernelGGL((grad_sumsq_kernel<__hip_bfloat16>), grid,
+                       dim3(BLOCK), 0, stream,
+                       (const __hip_bfloat16*)x.data_ptr(), n,
+                       out.data_ptr<float>(), slot);
+  else
+    hipLaunchKernelGGL((grad_sumsq_kernel<float>), grid, dim3(BLOCK), 0,
+                       stream, (const float*)x.data_ptr(), n,
+                       out.data_ptr<float>(), slot);
 }
 
 // standalone column sum for linear dbias: x [N, H] -> fp32 [H]
