@@ -165,6 +165,7 @@ class FusedAdamW(torch.optim.Optimizer):
 
     # --- gradient plumbing -------------------------------------------------
     def zero_grad(self, set_to_none: bool = False):
+        self._sumsq_cache = None
         for b in self.buckets:
             b.grad_flat.zero_()
             if not self._fp32_main_grad:
@@ -259,14 +260,37 @@ class FusedAdamW(torch.optim.Optimizer):
         for b in self.buckets:
             g = b.grad_shard if self.sharding_group else b.grad_flat
             g.mul_(factor)
+        if getattr(self, "_sumsq_cache", None) is not None:
+            self._sumsq_cache = self._sumsq_cache * (factor * factor)
+
+    def _bucket_sumsq(self) -> torch.Tensor:
+        """One fused pass per bucket accumulating sum(grad^2) fp32 into a
+        [n_buckets] tensor on device (csrc grad_sumsq kernel — the
+        reference's check_finite_and_unscale fusion, amp.py:212-216);
+        the result serves found_inf AND the clip norm without a second
+        read of the 13 GB of gradients. Cached per backward via
+        `_sumsq_cache` (invalidated by zero_grad/scale_grads).
+        """
+        if getattr(self, "_sumsq_cache", None) is not None:
+            return self._sumsq_cache
+        dev = self.buckets[0].grad_flat.device if self.buckets else "cpu"
+        out = torch.zeros(len(self.buckets), dtype=torch.float32, device=dev)
+        use_hip = out.is_cuda
+        if use_hip:
+            from paddlefleetx_amd.ops import hip_ext
+            ext = hip_ext()
+        for i, b in enumerate(self.buckets):
+            g = b.grad_shard if self.sharding_group else b.grad_flat
+            if use_hip:
+                ext.grad_sumsq(g, out, i)
+            else:
+                out[i] = torch.linalg.vector_norm(
+                    g, dtype=torch.float32) ** 2
+        self._sumsq_cache = out
+        return out
 
     def check_finite(self) -> bool:
-        for b in self.buckets:
-            g = b.grad_shard if self.sharding_group else b.grad_flat
-            if not torch.isfinite(
-                    torch.linalg.vector_norm(g, dtype=torch.float32)):
-                return False
-        return True
+        return bool(torch.isfinite(self._bucket_sumsq().sum()))
 
     # --- norm / clip -------------------------------------------------------
     def grad_global_norm(self, mp_group=None, pp_group=None) -> torch.Tensor:
@@ -275,10 +299,11 @@ class FusedAdamW(torch.optim.Optimizer):
         sq = torch.zeros((), dtype=torch.float32, device=device)
         mp_ws = mp_group.world_size if mp_group is not None else 1
         sg = self.sharding_group
+        if mp_ws == 1:
+            sq = self._bucket_sumsq().sum()
         for b in self.buckets:
             if mp_ws == 1:
-                g = b.grad_shard if sg is not None else b.grad_flat
-                sq += torch.linalg.vector_norm(g, dtype=torch.float32) ** 2
+                break
             else:
                 for p, off in zip(b.params, b.offsets):
                     lo, hi = off, off + p.numel()

@@ -12,11 +12,12 @@ REPO = os.path.join(os.path.dirname(__file__), "..")
 
 
 def _run_engine(fused: bool):
-    from paddlefleetx_amd.parallel.env import set_hcg
+    from paddlefleetx_amd.parallel.env import set_hcg, set_seed
     from paddlefleetx_amd.parallel.topology import HybridTopology
     import torch.distributed as dist
     if not dist.is_initialized():
         set_hcg(HybridTopology())
+    set_seed(1234)  # reset the mp RNG tracker between builds
     from paddlefleetx_amd.models import build_module
     from paddlefleetx_amd.core import EagerEngine
     from paddlefleetx_amd.utils.config import get_config
