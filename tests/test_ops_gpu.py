@@ -75,6 +75,24 @@ def test_layernorm_residual_fused(dev, shape):
     assert torch.allclose(db, db2, atol=0.5, rtol=1e-2)
 
 
+def test_grad_sumsq(dev):
+    out = torch.zeros(3, device=dev)
+    xs = [torch.randn(4096 * 13, device=dev, dtype=torch.bfloat16),
+          torch.randn(2048, device=dev, dtype=torch.bfloat16),
+          torch.randn(8192, device=dev)]
+    for i, x in enumerate(xs):
+        hip_ext().grad_sumsq(x, out, i)
+    for i, x in enumerate(xs):
+        want = float(torch.linalg.vector_norm(x, dtype=torch.float32) ** 2)
+        assert abs(float(out[i]) - want) / want < 1e-3, (i, float(out[i]), want)
+    # inf/nan propagate (the found_inf contract)
+    bad = torch.randn(1024, device=dev, dtype=torch.bfloat16)
+    bad[17] = float("inf")
+    out2 = torch.zeros(1, device=dev)
+    hip_ext().grad_sumsq(bad, out2, 0)
+    assert not torch.isfinite(out2[0])
+
+
 def test_colsum(dev):
     x = torch.randn(8192, 4096, device=dev, dtype=torch.bfloat16)
     out = hip_ext().colsum(x)
