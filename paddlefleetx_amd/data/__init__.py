@@ -10,7 +10,8 @@ from torch.utils.data import DataLoader
 
 from paddlefleetx_amd.data.gpt_dataset import GPTDataset, GPTSyntheticDataset
 from paddlefleetx_amd.data.sampler import GPTBatchSampler
-from paddlefleetx_amd.data.ernie_dataset import ErnieSyntheticDataset
+from paddlefleetx_amd.data.ernie_dataset import (ErnieSyntheticDataset,
+                                                 ErnieWWMDataset)
 from paddlefleetx_amd.data.vision_dataset import (ImageFolderDataset,
                                                   SyntheticImageNetDataset)
 from paddlefleetx_amd.parallel.env import (get_data_world_rank,
@@ -24,6 +25,7 @@ _DATASETS = {
     "ImageFolderDataset": ImageFolderDataset,
     "GeneralClsDataset": ImageFolderDataset,
     "ErnieSyntheticDataset": ErnieSyntheticDataset,
+    "ErnieWWMDataset": ErnieWWMDataset,
 }
 
 

@@ -86,3 +86,69 @@ def ernie_collate_fn(samples):
 
 # build_dataloader picks this up as the dataset's collate
 ErnieSyntheticDataset.collate_fn = staticmethod(ernie_collate_fn)
+
+
+class ErnieWWMDataset(Dataset):
+    """ERNIE pretraining over preprocessed Chinese whole-word-mask bins:
+    `<prefix>_ids.npy` + `<prefix>_idx.npz` (+ `<prefix>_wwm.npy`
+    continuation marks) written by tools/preprocess_data.py
+    --model_family ernie --whole_word_mask. Masking selects WHOLE words
+    (a token plus its '##' continuations) per
+    ernie_preprocess.create_masked_lm_predictions_wwm — the reference's
+    create_pretraining_data.py + dataset_utils.py pipeline."""
+
+    CLS, SEP, PAD, MASK = 1, 2, 0, 3
+
+    def __init__(self, input_prefix: str, seq_len: int = 512,
+                 vocab_size: int = 40000, masked_lm_prob: float = 0.15,
+                 mode: str = "Train", seed: int = 1234, **unused):
+        self.ids = np.load(input_prefix + "_ids.npy", mmap_mode="r")
+        lens = np.load(input_prefix + "_idx.npz")["lens"]
+        self.doc_offsets = np.concatenate([[0], np.cumsum(lens)])
+        import os as _os
+        wwm_path = input_prefix + "_wwm.npy"
+        self.wwm = np.load(wwm_path, mmap_mode="r") \
+            if _os.path.exists(wwm_path) else None
+        self.seq_len = int(seq_len)
+        self.vocab_size = int(vocab_size)
+        self.masked_lm_prob = masked_lm_prob
+        self.seed = seed
+
+    def __len__(self):
+        return max(1, len(self.doc_offsets) - 1)
+
+    def __getitem__(self, idx):
+        idx = int(idx)
+        rng = np.random.RandomState(self.seed + idx)
+        lo, hi = int(self.doc_offsets[idx]), int(self.doc_offsets[idx + 1])
+        body = np.asarray(self.ids[lo:hi], dtype=np.int64)
+        cont = np.asarray(self.wwm[lo:hi], dtype=np.int64) \
+            if self.wwm is not None else np.zeros_like(body)
+        max_body = self.seq_len - 2  # [CLS] body [SEP]
+        if len(body) > max_body:
+            start = rng.randint(0, len(body) - max_body + 1)
+            body = body[start:start + max_body]
+            cont = cont[start:start + max_body]
+            if cont[0]:  # don't start mid-word
+                first = np.argmax(cont == 0)
+                body, cont = body[first:], cont[first:]
+        tokens = np.concatenate([[self.CLS], body, [self.SEP]])
+        is_cont = np.concatenate([[0], cont, [0]])
+        from paddlefleetx_amd.data.ernie_preprocess import \
+            create_masked_lm_predictions_wwm
+        masked, labels = create_masked_lm_predictions_wwm(
+            tokens, is_cont, self.vocab_size, rng,
+            masked_lm_prob=self.masked_lm_prob, mask_token_id=self.MASK,
+            special_ids=(self.PAD, self.CLS, self.SEP, self.MASK))
+        pad = self.seq_len - len(masked)
+        masked = np.concatenate([masked, np.full(pad, self.PAD)])
+        labels = np.concatenate([labels, np.full(pad, -1)])
+        token_types = np.zeros(self.seq_len, dtype=np.int64)
+        nsp = 0
+        return (torch.from_numpy(masked.astype(np.int64)),
+                torch.from_numpy(token_types),
+                torch.from_numpy(labels.astype(np.int64)),
+                torch.tensor(nsp))
+
+
+ErnieWWMDataset.collate_fn = staticmethod(ernie_collate_fn)

@@ -111,3 +111,37 @@ def test_preprocess_tool_ernie_wwm_end_to_end():
         lens = np.load(out + "_idx.npz")["lens"]
         assert len(ids) == len(wwm) == int(lens.sum())
         assert wwm.max() == 1 and wwm.min() == 0
+
+
+def test_ernie_wwm_dataset_end_to_end():
+    """preprocess tool output -> ErnieWWMDataset -> whole-word spans
+    masked together."""
+    import numpy as np
+    import torch
+    from paddlefleetx_amd.data.ernie_dataset import ErnieWWMDataset
+    with tempfile.TemporaryDirectory() as td:
+        prefix = os.path.join(td, "c")
+        # two docs; continuation marks pair tokens into words
+        ids = np.array([10, 11, 12, 13, 14, 15,  20, 21, 22, 23],
+                       dtype=np.int32)
+        wwm = np.array([0, 1, 0, 1, 0, 1,  0, 0, 1, 1], dtype=np.int8)
+        np.save(prefix + "_ids.npy", ids)
+        np.savez(prefix + "_idx.npz", lens=np.array([6, 4]))
+        np.save(prefix + "_wwm.npy", wwm)
+        ds = ErnieWWMDataset(prefix, seq_len=12, vocab_size=100,
+                             masked_lm_prob=0.5)
+        assert len(ds) == 2
+        found_span_mask = False
+        for seed in range(8):
+            ds.seed = seed * 100
+            masked, tt, labels, nsp = ds[0]
+            assert masked.shape == (12,)
+            pred = (labels != -1).nonzero().flatten().tolist()
+            # predictions at a continuation position imply its head too
+            body = [10, 11, 12, 13, 14, 15]
+            for i in pred:
+                tok = body[i - 1]  # offset by [CLS]
+                if tok in (11, 13, 15):  # continuation tokens
+                    assert (i - 1) in pred
+                    found_span_mask = True
+        assert found_span_mask
