@@ -34,6 +34,9 @@ def parse_args():
     p.add_argument("--micro-batch", type=int, default=None)
     p.add_argument("--acc-steps", type=int, default=None)
     p.add_argument("--seq-len", type=int, default=1024)
+    p.add_argument("--hip-graph", action="store_true",
+                   help="capture the micro-step in a hipGraph (launch-"
+                        "bound small-model configs)")
     p.add_argument("--dropout", type=float, default=0.0,
                    help="hidden+attention dropout (the reference's "
                         "pretrain configs run 0.1; in-kernel Philox "
@@ -98,6 +101,8 @@ def main():
         f"Distributed.dp_degree={dp}", f"Distributed.mp_degree={tp}",
         f"Distributed.pp_degree={pp}",
     ]
+    if args.hip_graph:
+        overrides.append("Engine.hip_graph=True")
     cfg = get_config(cfg_path, overrides=overrides)
     hcg = init_dist_env(cfg)
     rank = hcg.global_rank

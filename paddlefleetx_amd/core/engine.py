@@ -151,6 +151,18 @@ class EagerEngine(BasicEngine):
         if ckpt_dir:
             self.load(ckpt_dir)
 
+        # hipGraph capture of the micro-step (Engine.hip_graph: true):
+        # records one forward+backward after warmup and replays it with
+        # static input buffers — removes per-kernel launch overhead on
+        # launch-bound (small-model) configs. Constraints enforced in
+        # _maybe_capture_graph (single process, no overlap hooks, no
+        # dropout/dynamic loss scale: Philox seeds and the scale are
+        # baked into the capture).
+        self._graph = None
+        self._graph_want = bool(configs.get("Engine", {})
+                                .get("hip_graph", False))
+        self._graph_steps_seen = 0
+
         from paddlefleetx_amd.utils.profiler import ProfilerGuard
         self.profiler = ProfilerGuard(configs.get("Profiler"))
 
@@ -254,21 +266,84 @@ class EagerEngine(BasicEngine):
                 batch, self.module.loss_fn, self.accumulate_steps,
                 scale=self.loss_scale)
         else:
+            if self._graph_want and self._graph is None:
+                self._maybe_capture_graph(batch)
             loss = self._model_forward_backward(batch)
         self._optim_update_params()
         return loss
 
     def _model_forward_backward(self, batch) -> torch.Tensor:
         micros = _split_micro(batch, self.accumulate_steps)
-        total = 0.0
+        total = None
         for i, mb in enumerate(micros):
-            loss = self.module.training_step(mb)
-            scaled = loss * (self.loss_scale / self.accumulate_steps)
-            if self._overlap_reduce and i == len(micros) - 1:
-                self.optimizer.begin_overlap_reduce()
-            self.module.backward(scaled)
-            total += float(loss.detach())
-        return torch.tensor(total / self.accumulate_steps)
+            loss = self._graph_step(mb) if self._graph is not None \
+                else None
+            if loss is None:
+                loss = self.module.training_step(mb)
+                scaled = loss * (self.loss_scale / self.accumulate_steps)
+                if self._overlap_reduce and i == len(micros) - 1:
+                    self.optimizer.begin_overlap_reduce()
+                self.module.backward(scaled)
+            # accumulate ON DEVICE: a float() here would host-sync every
+            # micro-batch and stall the launch pipeline. clone() because
+            # under graph replay `loss` is the static capture tensor that
+            # the NEXT replay overwrites
+            d = loss.detach().clone()
+            total = d if total is None else total + d
+        return total / self.accumulate_steps
+
+    def _maybe_capture_graph(self, batch):
+        """Capture ONE micro-batch forward+backward into a hipGraph after
+        two warmup steps. Grad accumulation works across replays because
+        every parameter gradient lives in a pre-allocated bucket view
+        (optims/optimizer.py) that the recorded kernels add into."""
+        import torch.distributed as tdist
+        if self.device.type != "cuda" or self.is_pipeline:
+            return
+        if tdist.is_initialized() and tdist.get_world_size() > 1:
+            return  # collectives inside capture not supported here
+        if getattr(self, "_overlap_reduce", False) or self.loss_scale != 1.0:
+            return
+        mcfg = self.configs.get("Model", {})
+        if float(mcfg.get("hidden_dropout_prob", 0) or 0) > 0 or                 float(mcfg.get("attention_probs_dropout_prob", 0) or 0) > 0:
+            return  # RNG seeds would be frozen into the capture
+        self._graph_steps_seen += 1
+        if self._graph_steps_seen <= 2:
+            return  # allocator/TunableOp warmup on the eager path
+        micros = _split_micro(batch, self.accumulate_steps)
+        static = tuple(t.clone() if torch.is_tensor(t) else t
+                       for t in micros[0])
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                self.optimizer.zero_grad()
+                loss = self.module.training_step(static)
+                (loss * (1.0 / self.accumulate_steps)).backward()
+        torch.cuda.current_stream().wait_stream(s)
+        self.optimizer.zero_grad()
+        g = torch.cuda.CUDAGraph()
+        try:
+            with torch.cuda.graph(g):
+                loss = self.module.training_step(static)
+                (loss * (1.0 / self.accumulate_steps)).backward()
+        except Exception as e:
+            logger.warning(f"hipGraph capture failed, staying eager: {e}")
+            self.optimizer.zero_grad()
+            self._graph_want = False
+            return
+        self.optimizer.zero_grad()
+        self._graph = {"obj": g, "static": static, "loss": loss}
+        logger.info("hipGraph captured: replaying the micro-step "
+                    f"(acc={self.accumulate_steps})")
+
+    def _graph_step(self, mb):
+        st = self._graph["static"]
+        for dst, srct in zip(st, mb):
+            if torch.is_tensor(dst):
+                dst.copy_(srct, non_blocking=True)
+        self._graph["obj"].replay()
+        return self._graph["loss"]
 
     def _optim_update_params(self):
         model = self.module.model

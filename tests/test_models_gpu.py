@@ -128,3 +128,57 @@ def test_topp_generation_gpu():
                                  "eos_token_id": 511})
     out = gen(torch.randint(0, 511, (2, 8), device="cuda"))
     assert out.shape[0] == 2 and out.shape[1] <= 6
+
+
+def test_engine_hipgraph_matches_eager():
+    dev = "cuda"
+    """Engine.hip_graph: captured micro-step replay == eager execution
+    (same losses and updated params over several steps)."""
+    from paddlefleetx_amd.parallel.env import set_hcg, set_seed
+    from paddlefleetx_amd.parallel.topology import HybridTopology
+    import torch.distributed as dist
+    if not dist.is_initialized():
+        set_hcg(HybridTopology())
+    from paddlefleetx_amd.models import build_module
+    from paddlefleetx_amd.core import EagerEngine
+    from paddlefleetx_amd.utils.config import get_config
+    import os
+    repo = os.path.join(os.path.dirname(__file__), "..")
+
+    def run(graph):
+        set_seed(1234)
+        cfg = get_config(os.path.join(
+            repo, "paddlefleetx_amd/configs/nlp/gpt/"
+            "pretrain_gpt_345M_single_card.yaml"),
+            overrides=["Model.hidden_size=256", "Model.num_layers=2",
+                       "Model.num_attention_heads=4",
+                       "Model.vocab_size=1024",
+                       "Model.max_position_embeddings=128",
+                       "Model.hidden_dropout_prob=0.0",
+                       "Model.attention_probs_dropout_prob=0.0",
+                       "Global.micro_batch_size=2",
+                       "Global.local_batch_size=4",
+                       f"Engine.hip_graph={graph}"])
+        torch.manual_seed(7)
+        module = build_module(cfg)
+        engine = EagerEngine(cfg, module)
+        losses = []
+        for s in range(6):
+            torch.manual_seed(100 + s)
+            batch = (torch.randint(0, 1024, (4, 128), device=dev),
+                     torch.arange(128, device=dev).repeat(4, 1),
+                     torch.randint(0, 1024, (4, 128), device=dev),
+                     torch.ones(4, 128, device=dev))
+            losses.append(float(engine._fit_impl(batch)))
+        if graph:
+            assert engine._graph is not None, "graph never captured"
+        params = [b.model_flat.clone() for b in engine.optimizer.buckets]
+        return losses, params
+
+    l0, p0 = run(False)
+    l1, p1 = run(True)
+    for a, b in zip(l0, l1):
+        assert abs(a - b) < 5e-3, (l0, l1)
+    for a, b in zip(p0, p1):
+        assert torch.allclose(a.float(), b.float(), atol=1e-2), \
+            (a - b).float().abs().max()
