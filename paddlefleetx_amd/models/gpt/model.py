@@ -124,9 +124,17 @@ class MultiHeadAttention(nn.Module):
             # hybrid_model.py:328 RNG-tracker dropout)
             B, S, _ = qkv.shape
             packed = qkv.view(B, S, self.num_heads_local, 3, self.head_dim)
-            with model_parallel_rng():
+            if p_att > 0.0:
+                # mp-rank-local Philox seed for the in-kernel mask; the
+                # context seeds the generator, so enter it ONLY when a
+                # seed is drawn (hipGraph capture forbids reseeding)
+                with model_parallel_rng():
+                    o = flash_attention_packed(packed, self.num_heads_local,
+                                               scale=self.scale,
+                                               p_drop=p_att)
+            else:
                 o = flash_attention_packed(packed, self.num_heads_local,
-                                           scale=self.scale, p_drop=p_att)
+                                           scale=self.scale)
             return self.out_proj(o), None
         if self.sequence_parallel:
             # x: [s/mp, B, H]; qkv allgathered the seq dim -> [S, B, 3H/mp]
@@ -152,9 +160,12 @@ class MultiHeadAttention(nn.Module):
 
         if self.fused_attn and not (cache is not None and S == 1) and (
                 p_att == 0.0 or q.is_cuda):
-            with model_parallel_rng():
-                o = flash_attention(q, k, v, causal=True, scale=self.scale,
-                                    p_drop=p_att)
+            if p_att > 0.0:
+                with model_parallel_rng():
+                    o = flash_attention(q, k, v, causal=True,
+                                        scale=self.scale, p_drop=p_att)
+            else:
+                o = flash_attention(q, k, v, causal=True, scale=self.scale)
         else:
             scores = torch.matmul(q, k.transpose(-1, -2))
             probs = fused_softmax_causal(scores, self.scale)
