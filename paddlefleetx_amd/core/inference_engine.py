@@ -55,6 +55,12 @@ class InferenceEngine:
             self.model = module.model.to(dtype)
         load_inference_model(self.model, model_dir)
         self.model.to(self.device).eval()
+        # opt-in fp8 serving path (ops/fp8.py): the TRT-optimized-runtime
+        # analogue of the reference (inference_engine.py:227-242) — fp8
+        # MFMA GEMMs at 2x the bf16 rate on gfx950
+        if (generation_cfg or {}).get("fp8") or meta.get("fp8"):
+            from paddlefleetx_amd.ops.fp8 import convert_fp8_linears
+            convert_fp8_linears(self.model)
         logger.info(f"inference engine ready (model_dir={model_dir}, "
                     f"family={family}, mp={mp_degree}, "
                     f"device={self.device})")
