@@ -164,8 +164,9 @@ DEV_INLINE bf8 read_a_frag_lds(const unsigned short* lds, int row_stride,
 template <int D, bool CAUSAL, bool DROP>
 __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
     Strided q, Strided k, Strided v, StridedMut o, float* __restrict__ lse_out,
-    int H, int S, float scale, int kv_total, unsigned drop_thresh,
-    float drop_inv_keep, unsigned seed0, unsigned seed1) {
+    int H, int S, float scale, int kv_total, int n_home,
+    unsigned drop_thresh, float drop_inv_keep, unsigned seed0,
+    unsigned seed1) {
   constexpr int KRS = D + PAD;          // K row-major row stride
   // V^T row stride: 68 shorts (136 B) keeps b64 writes/reads 8B-aligned
   // while avoiding the 0-mod-128B d-stride that made the transpose writes
@@ -184,9 +185,12 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
   unsigned short* vt_lds = smem + 2 * KSZ;       // [2][VSZ]
 
   // bh on blockIdx.x: the linear dispatch then gives every CU one block
-  // of each qt, balancing the causal tile-count imbalance (B*H >= 256
-  // keeps all CUs fed within one qt wave)
-  const int qt = blockIdx.y;
+  // of each qt, balancing the causal tile-count imbalance when
+  // B*H >= 256. For SMALL bh (pipeline-stage micro shapes), n_home < 0
+  // marks PAIRED dispatch: gridDim.y is halved and each block runs BOTH
+  // qt = blockIdx.y and its complement (-n_home-1) - blockIdx.y, whose
+  // causal tile counts sum to a constant — equal work per block without
+  // needing >= 256 blocks per tile position.
   const int b = blockIdx.x / H, hh = blockIdx.x % H;
   const long bh = blockIdx.x;
   const int wid = threadIdx.x / WAVE;
@@ -195,6 +199,8 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
   const __hip_bfloat16* qp = q.at(b, hh);
   const __hip_bfloat16* kp = k.at(b, hh);
   const __hip_bfloat16* vp = v.at(b, hh);
+
+  auto run_qtile = [&](const int qt) {
 
   // RB 16-row fragments per wave (32 q rows/wave, 256/block): every K/V
   // B-fragment LDS read feeds RB MFMAs, and barriers amortize over 2x rows
@@ -517,6 +523,17 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_fwd_kernel(
       }
     }
   }
+  };  // run_qtile
+
+  const int n_home_abs = n_home < 0 ? -n_home : n_home;
+  run_qtile(blockIdx.y);
+  if (n_home < 0) {
+    const int qt2 = (n_home_abs - 1) - (int)blockIdx.y;
+    if (qt2 != (int)blockIdx.y && qt2 >= 0) {
+      __syncthreads();
+      run_qtile(qt2);
+    }
+  }
 }
 
 // ===========================================================================
@@ -564,7 +581,7 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dkv_kernel(
     Strided q, Strided k, Strided v, Strided dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
     StridedMut dk, StridedMut dv, int H, int S, float scale, int q_tiles,
-    unsigned drop_thresh, float drop_inv_keep, unsigned seed0,
+    int n_home, unsigned drop_thresh, float drop_inv_keep, unsigned seed0,
     unsigned seed1) {
   constexpr int RS = D + PAD;
   constexpr int TS = TILE + PAD;       // P image stride (b128-aligned)
@@ -580,7 +597,6 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dkv_kernel(
   __shared__ unsigned short p_lds[FW_WAVES * 16 * TS];
   __shared__ float lsed_lds[2][2 * TILE];
 
-  const int kt = blockIdx.y;            // 128-row kv block
   const int b = blockIdx.x / H, hh = blockIdx.x % H;
   const long bh = blockIdx.x;
   const int wid = threadIdx.x / WAVE;
@@ -591,6 +607,7 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dkv_kernel(
   const __hip_bfloat16* vp = v.at(b, hh);
   const __hip_bfloat16* dop = dout.at(b, hh);
 
+  auto run_kvtile = [&](const int kt) {
   const int kvrow0 = kt * QTILE + wid * 16;
   bf8 kfrag[D / 32], vfrag[D / 32];
   load_a_frags<D>(kp, k.rs, kvrow0, S, lane, kfrag);
@@ -828,6 +845,19 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dkv_kernel(
       vrow[(dt + 1) * 16 + ccol] = (unsigned short)(uv >> 16);
     }
   }
+  };  // run_kvtile
+
+  // paired dispatch for small bh: block kt=0 streams ALL q tiles while
+  // the last kt streams few — (kt, n-1-kt) pairs even the totals
+  const int n_home_abs = n_home < 0 ? -n_home : n_home;
+  run_kvtile(blockIdx.y);
+  if (n_home < 0) {
+    const int kt2 = (n_home_abs - 1) - (int)blockIdx.y;
+    if (kt2 != (int)blockIdx.y && kt2 >= 0) {
+      __syncthreads();
+      run_kvtile(kt2);
+    }
+  }
 }
 
 // ===========================================================================
@@ -840,7 +870,7 @@ template <int D, bool CAUSAL, bool DROP>
 __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dq_kernel(
     Strided q, Strided k, Strided v, Strided dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
-    StridedMut dq, int H, int S, float scale, int kv_total,
+    StridedMut dq, int H, int S, float scale, int kv_total, int n_home,
     unsigned drop_thresh, float drop_inv_keep, unsigned seed0,
     unsigned seed1) {
   constexpr int RS = D + PAD;
@@ -858,7 +888,6 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dq_kernel(
   unsigned short* v_lds = smem + 2 * (KSZ + TSZ);
   unsigned short* p_lds = smem + 2 * (KSZ + TSZ + VSZ);
 
-  const int qt = blockIdx.y;   // 256-row q block
   const int b = blockIdx.x / H, hh = blockIdx.x % H;
   const long bh = blockIdx.x;
   const int wid = threadIdx.x / WAVE;
@@ -869,6 +898,7 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dq_kernel(
   const __hip_bfloat16* vp = v.at(b, hh);
   const __hip_bfloat16* dop = dout.at(b, hh);
 
+  auto run_qtile = [&](const int qt) {
   const int qrow0 = qt * (QTILE * RB) + wid * (16 * RB);
   bf8 qfrag[RB][D / 32], dofrag[RB][D / 32];
 #pragma unroll
@@ -1072,6 +1102,17 @@ __global__ __launch_bounds__(FW_BLOCKT) void attn_bwd_dq_kernel(
         qrow_p[(dt + 1) * 16 + ccol] = (unsigned short)(u >> 16);
       }
     }
+  };  // run_qtile
+
+  const int n_home_abs = n_home < 0 ? -n_home : n_home;
+  run_qtile(blockIdx.y);
+  if (n_home < 0) {
+    const int qt2 = (n_home_abs - 1) - (int)blockIdx.y;
+    if (qt2 != (int)blockIdx.y && qt2 >= 0) {
+      __syncthreads();
+      run_qtile(qt2);
+    }
+  }
 }
 
 Strided strided_of(const torch::Tensor& t, int b_dim, int h_dim, int s_dim) {
@@ -1102,7 +1143,12 @@ void launch_fwd(Strided q, Strided k, Strided v, StridedMut o, float* lse,
                 float p_drop = 0.f, unsigned long long seed = 0) {
   int q_blocks = (S + QTILE * 2 - 1) / (QTILE * 2);  // RB=2 home tiles
   int kv_total = (S + TILE - 1) / TILE;
-  dim3 grid(B * H, q_blocks);
+  // small-bh causal balance: pair (qt, n-1-qt) per block so per-block
+  // work is constant without needing B*H >= 256 (ROADMAP r1 weak #6)
+  bool pair = causal && B * H < 256 && q_blocks > 1;
+  int grid_y = pair ? (q_blocks + 1) / 2 : q_blocks;
+  int n_home = pair ? -q_blocks : q_blocks;
+  dim3 grid(B * H, grid_y);
   auto stream = at::hip::getCurrentHIPStream();
   unsigned thresh = (unsigned)(p_drop * 4294967296.0);
   float invk = 1.f / (1.f - p_drop);
@@ -1111,7 +1157,7 @@ void launch_fwd(Strided q, Strided k, Strided v, StridedMut o, float* lse,
 #define LAUNCH_FWD(DD, CC, PP)                                              \
   hipLaunchKernelGGL((attn_fwd_kernel<DD, CC, PP>), grid, dim3(FW_BLOCKT),  \
                      0, stream, q, k, v, o, lse, H, S, scale, kv_total,     \
-                     thresh, invk, s0, s1)
+                     n_home, thresh, invk, s0, s1)
 #define PICK_FWD(DD, CC) do { if (drop) LAUNCH_FWD(DD, CC, true); \
                               else LAUNCH_FWD(DD, CC, false); } while (0)
   if (D == 128) { if (causal) PICK_FWD(128, true); else PICK_FWD(128, false); }
@@ -1133,8 +1179,11 @@ void launch_bwd(Strided q, Strided k, Strided v, Strided dout, Strided o,
   int tiles64 = (S + TILE - 1) / TILE;       // inner streamed tiles
   int blocks128 = (S + QTILE - 1) / QTILE;   // dkv per-block home tile
   int blocks256 = (S + QTILE * 2 - 1) / (QTILE * 2);  // dq RB=2 home tiles
-  dim3 grid(B * H, blocks128);
-  dim3 grid_dq(B * H, blocks256);
+  bool pair = causal && B * H < 256;
+  int nh_dkv = (pair && blocks128 > 1) ? -blocks128 : blocks128;
+  int nh_dq = (pair && blocks256 > 1) ? -blocks256 : blocks256;
+  dim3 grid(B * H, nh_dkv < 0 ? (blocks128 + 1) / 2 : blocks128);
+  dim3 grid_dq(B * H, nh_dq < 0 ? (blocks256 + 1) / 2 : blocks256);
   unsigned thresh = (unsigned)(p_drop * 4294967296.0);
   float invk = 1.f / (1.f - p_drop);
   unsigned s0 = (unsigned)seed, s1 = (unsigned)(seed >> 32);
@@ -1143,12 +1192,12 @@ void launch_bwd(Strided q, Strided k, Strided v, Strided dout, Strided o,
   do {                                                                       \
     hipLaunchKernelGGL((attn_bwd_dkv_kernel<DD, CC, PP>), grid,              \
                        dim3(FW_BLOCKT), 0, stream, q, k, v, dout, lse,       \
-                       delta, dk, dv, H, S, scale, tiles64, thresh, invk,    \
-                       s0, s1);                                              \
+                       delta, dk, dv, H, S, scale, tiles64, nh_dkv, thresh,  \
+                       invk, s0, s1);                                        \
     hipLaunchKernelGGL((attn_bwd_dq_kernel<DD, CC, PP>), grid_dq,            \
                        dim3(FW_BLOCKT), 0, stream, q, k, v, dout, lse,       \
-                       delta, dq, H, S, scale, tiles64, thresh, invk,        \
-                       s0, s1);                                              \
+                       delta, dq, H, S, scale, tiles64, nh_dq, thresh,       \
+                       invk, s0, s1);                                        \
   } while (0)
 #define PICK_BWD(DD, CC) do { if (drop) LAUNCH_BWD(DD, CC, true); \
                               else LAUNCH_BWD(DD, CC, false); } while (0)

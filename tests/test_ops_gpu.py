@@ -211,6 +211,7 @@ def test_flash_attention_fwd(dev, cfg):
     dict(B=1, H=2, S=512, D=64),
     dict(B=2, H=2, S=192, D=64),   # ragged S (tail masking in dkv/dq)
     dict(B=1, H=2, S=320, D=128),  # ragged S, D=128
+    dict(B=4, H=16, S=512, D=128),  # pp-stage shape: PAIRED dispatch
 ])
 def test_flash_attention_bwd(dev, cfg):
     B, H, S, D = cfg["B"], cfg["H"], cfg["S"], cfg["D"]
