@@ -43,6 +43,8 @@ def main():
             repo, "paddlefleetx_amd/configs/nlp/moe/"
             "pretrain_moe_345M_64experts_ep8.yaml"),
             overrides=["Distributed.expert_parallel_degree=1",
+                       "Distributed.dp_degree=1",
+                       "Distributed.world_size=1",
                        "Global.micro_batch_size=8",
                        "Global.local_batch_size=8"])
         module = build_module(cfg)
@@ -60,6 +62,7 @@ def main():
             repo, "paddlefleetx_amd/configs/vis/vit/"
             "ViT_huge_patch14_224_pretrain_dp8.yaml"),
             overrides=["Distributed.dp_degree=1",
+                       "Distributed.world_size=1",
                        "Global.micro_batch_size=32",
                        "Global.local_batch_size=32",
                        "Global.global_batch_size=32"])
