@@ -140,3 +140,55 @@ def test_stage3_vs_plain_adamw_equivalence():
     for (n1, p1), (n2, p2) in zip(net1.named_parameters(),
                                   net2.named_parameters()):
         assert torch.allclose(p1, p2, atol=1e-4), (n1, (p1 - p2).abs().max())
+
+
+def test_zero3_cpu_offload_step():
+    """Stage3AdamW(offload=True): fp32 state on host; a step matches the
+    on-device-state step numerically."""
+    import torch
+    from paddlefleetx_amd.parallel.env import set_hcg
+    from paddlefleetx_amd.parallel.topology import HybridTopology
+    import torch.distributed as dist
+    if not dist.is_initialized():
+        set_hcg(HybridTopology())
+    from paddlefleetx_amd.parallel.zero3 import (GroupShardedStage3,
+                                                 Stage3AdamW)
+    import torch.nn as nn
+
+    class Blk(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.l = nn.Linear(16, 16)
+
+        def forward(self, x):
+            return torch.relu(self.l(x))
+
+    class Net(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.blocks = nn.ModuleList([Blk() for _ in range(2)])
+
+        def forward(self, x):
+            for b in self.blocks:
+                x = b(x)
+            return x
+
+    def run(offload):
+        torch.manual_seed(5)
+        net = Net()
+        w = GroupShardedStage3(net, group=None, unit_classes=("Blk",),
+                               prefetch=False)
+        opt = Stage3AdamW(w, lr=1e-2, offload=offload)
+        torch.manual_seed(1)
+        x = torch.randn(4, 16)
+        for _ in range(3):
+            y = w(x)
+            y.pow(2).mean().backward()
+            opt.reduce_and_step()
+        w.gather_full_params()
+        return [p.detach().clone() for p in net.parameters()]
+
+    p0 = run(False)
+    p1 = run(True)
+    for a, b in zip(p0, p1):
+        assert torch.allclose(a, b, atol=1e-6), (a - b).abs().max()
