@@ -79,13 +79,6 @@ def _rotate_wait(reqs):
         r.wait()
 
 
-def _rotate(tensors, g):
-    """Blocking ring exchange (used where no compute can overlap)."""
-    reqs, outs, _sends = _rotate_start(tensors, g)
-    _rotate_wait(reqs)
-    return outs
-
-
 def _halves(rank: int, cp: int):
     """Zigzag layout: rank r's local sequence = global half-chunks
     (r, 2cp-1-r) concatenated — balances the causal triangle so every
@@ -176,19 +169,22 @@ class _RingAttnFn(torch.autograd.Function):
         cur_k, cur_v = k, v
         cur_dk = torch.zeros_like(k, dtype=torch.float32)
         cur_dv = torch.zeros_like(v, dtype=torch.float32)
+        dkv_pend = None  # in-flight (dk, dv) hop from the previous step
         h = q.shape[2] // 2
         for j in range(cp):
             src = (r - j) % cp
-            # k/v for the NEXT step can travel during this step's block
-            # backward; the dk/dv accumulators move after the update
+            # k/v for the NEXT step travel during this step's block
+            # backward. The (dk, dv) accumulators posted at the end of
+            # the PREVIOUS step are also still in flight here: the block
+            # backward doesn't read them, so the wait is deferred until
+            # just before the += below and the hop overlaps compute too.
             kv_pend = _rotate_start([cur_k, cur_v], g) if cp > 1 else None
+            blocks = []  # (dq_slice, dk_slice, dq_j, dk_j, dv_j)
             if not ctx.zigzag:
                 if src <= r:
-                    dq_j, dk_j, dv_j = _block_bwd(do, q, cur_k, cur_v, o,
-                                                  lse, src == r, ctx.scale)
-                    dq_acc += dq_j.float()
-                    cur_dk += dk_j.float()
-                    cur_dv += dv_j.float()
+                    blocks.append((slice(None), slice(None),
+                                   *_block_bwd(do, q, cur_k, cur_v, o,
+                                               lse, src == r, ctx.scale)))
             else:
                 for qi, gq in enumerate(_halves(r, cp)):
                     qs = slice(qi * h, (qi + 1) * h)
@@ -197,23 +193,31 @@ class _RingAttnFn(torch.autograd.Function):
                         if mode is None:
                             continue
                         ks = slice(ki * h, (ki + 1) * h)
-                        dq_j, dk_j, dv_j = _block_bwd(
+                        blocks.append((qs, ks, *_block_bwd(
                             do[:, :, qs].contiguous(),
                             q[:, :, qs].contiguous(),
                             cur_k[:, :, ks].contiguous(),
                             cur_v[:, :, ks].contiguous(),
                             o[:, :, qs].contiguous(),
                             lse[:, :, qs].contiguous(),
-                            mode == "causal", ctx.scale)
-                        dq_acc[:, :, qs] += dq_j.float()
-                        cur_dk[:, :, ks] += dk_j.float()
-                        cur_dv[:, :, ks] += dv_j.float()
+                            mode == "causal", ctx.scale)))
+            if dkv_pend is not None:
+                _rotate_wait(dkv_pend[0])
+                cur_dk, cur_dv = dkv_pend[1]
+                dkv_pend = None
+            for qs, ks, dq_j, dk_j, dv_j in blocks:
+                dq_acc[:, :, qs] += dq_j.float()
+                cur_dk[:, :, ks] += dk_j.float()
+                cur_dv[:, :, ks] += dv_j.float()
             if cp > 1:
                 # after cp hops each (k, v, dk, dv) quartet is back at
-                # its owning rank
+                # its owning rank; the last dk/dv hop is awaited below
                 _rotate_wait(kv_pend[0])
                 cur_k, cur_v = kv_pend[1]
-                cur_dk, cur_dv = _rotate([cur_dk, cur_dv], g)
+                dkv_pend = _rotate_start([cur_dk, cur_dv], g)
+        if dkv_pend is not None:
+            _rotate_wait(dkv_pend[0])
+            cur_dk, cur_dv = dkv_pend[1]
         return (dq_acc.to(q.dtype), cur_dk.to(k.dtype), cur_dv.to(v.dtype),
                 None, None)
 
