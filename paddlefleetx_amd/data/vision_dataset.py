@@ -53,9 +53,13 @@ class ImageFolderDataset(Dataset):
     collate_fn = None
 
     def __init__(self, root: str, image_size: int = 224, mode: str = "Train",
-                 **unused):
+                 rand_augment: bool = False, ra_num_ops: int = 2,
+                 ra_magnitude: int = 9, **unused):
         self.root = root
         self.image_size = image_size
+        self.rand_augment = None
+        if rand_augment and mode == "Train":
+            self.rand_augment = RandAugment(ra_num_ops, ra_magnitude)
         classes = sorted(d for d in os.listdir(root)
                          if os.path.isdir(os.path.join(root, d)))
         self.class_to_idx = {c: i for i, c in enumerate(classes)}
@@ -83,6 +87,8 @@ class ImageFolderDataset(Dataset):
                     .float() / 255.0
             except ImportError as e:
                 raise RuntimeError(f"cannot decode {path}: PIL missing") from e
+        if self.rand_augment is not None:
+            img = self.rand_augment(img)
         return img, label
 
 
@@ -131,3 +137,89 @@ def cutmix_batch(images: torch.Tensor, labels: torch.Tensor,
     y = one_hot(labels, num_classes, smoothing)
     y = lam_adj * y + (1.0 - lam_adj) * y[perm]
     return out, y
+
+
+# ---------------------------------------------------------------------------
+# RandAugment (reference data/transforms/preprocess.py rand-aug), done
+# natively on CHW float tensors in [0, 1] — no PIL dependency.
+# ---------------------------------------------------------------------------
+
+def _affine(img: torch.Tensor, theta_2x3: torch.Tensor) -> torch.Tensor:
+    c, h, w = img.shape
+    grid = torch.nn.functional.affine_grid(
+        theta_2x3.unsqueeze(0), (1, c, h, w), align_corners=False)
+    return torch.nn.functional.grid_sample(
+        img.unsqueeze(0), grid, padding_mode="zeros",
+        align_corners=False).squeeze(0)
+
+
+class RandAugment:
+    """Pick `num_ops` random ops at strength `magnitude`/`num_bins` per
+    image. Ops: brightness, contrast, sharpness, solarize, posterize,
+    autocontrast, rotate, shear-x/y, translate-x/y (the affine ones via
+    grid_sample). Callable on a CHW float tensor in [0, 1]."""
+
+    OPS = ("identity", "brightness", "contrast", "sharpness", "solarize",
+           "posterize", "autocontrast", "rotate", "shear_x", "shear_y",
+           "translate_x", "translate_y")
+
+    def __init__(self, num_ops: int = 2, magnitude: int = 9,
+                 num_bins: int = 31, generator: "torch.Generator" = None):
+        self.num_ops = int(num_ops)
+        self.m = float(magnitude) / float(num_bins - 1)
+        self.gen = generator
+
+    def _rand(self, n):
+        return torch.rand(n, generator=self.gen)
+
+    def _apply(self, img: torch.Tensor, op: str, sign: float) -> torch.Tensor:
+        m = self.m
+        if op == "identity":
+            return img
+        if op == "brightness":
+            return (img * (1.0 + sign * 0.9 * m)).clamp(0, 1)
+        if op == "contrast":
+            mean = img.mean()
+            return (mean + (img - mean) * (1.0 + sign * 0.9 * m)).clamp(0, 1)
+        if op == "sharpness":
+            k = torch.full((img.shape[0], 1, 3, 3), 1.0 / 9.0)
+            blur = torch.nn.functional.conv2d(
+                img.unsqueeze(0), k, padding=1,
+                groups=img.shape[0]).squeeze(0)
+            return (img + sign * 0.9 * m * (img - blur)).clamp(0, 1)
+        if op == "solarize":
+            thr = 1.0 - m  # magnitude lowers the inversion threshold
+            return torch.where(img >= thr, 1.0 - img, img)
+        if op == "posterize":
+            bits = max(1, 8 - int(round(4 * m)))
+            q = float(1 << (8 - bits))
+            return torch.floor(img * 255.0 / q) * q / 255.0
+        if op == "autocontrast":
+            lo = img.amin(dim=(1, 2), keepdim=True)
+            hi = img.amax(dim=(1, 2), keepdim=True)
+            scale = torch.where(hi > lo, 1.0 / (hi - lo).clamp_min(1e-6),
+                                torch.ones_like(hi))
+            return ((img - lo) * scale).clamp(0, 1)
+        # affine ops
+        t = torch.eye(2, 3)
+        if op == "rotate":
+            a = sign * m * (30.0 * 3.141592653589793 / 180.0)
+            ca, sa = float(torch.cos(torch.tensor(a))), \
+                float(torch.sin(torch.tensor(a)))
+            t[0, 0], t[0, 1], t[1, 0], t[1, 1] = ca, -sa, sa, ca
+        elif op == "shear_x":
+            t[0, 1] = sign * 0.3 * m
+        elif op == "shear_y":
+            t[1, 0] = sign * 0.3 * m
+        elif op == "translate_x":
+            t[0, 2] = sign * 0.45 * m
+        elif op == "translate_y":
+            t[1, 2] = sign * 0.45 * m
+        return _affine(img, t)
+
+    def __call__(self, img: torch.Tensor) -> torch.Tensor:
+        for _ in range(self.num_ops):
+            op = self.OPS[int(self._rand(1) * len(self.OPS))]
+            sign = 1.0 if float(self._rand(1)) < 0.5 else -1.0
+            img = self._apply(img, op, sign)
+        return img
