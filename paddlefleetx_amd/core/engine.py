@@ -442,7 +442,14 @@ class EagerEngine(BasicEngine):
                 break
             batch = tuple(t.to(self.device) if torch.is_tensor(t) else t
                           for t in batch)
-            loss = self.module.validation_step(batch)
+            if self.is_pipeline:
+                # stage-local validation_step can't see the whole model;
+                # run the forward-only pipeline schedule instead
+                # (reference eager_engine.py:655 eval_batch)
+                loss = self.module.model.eval_pipeline(
+                    batch, self.module.loss_fn, self.accumulate_steps)
+            else:
+                loss = self.module.validation_step(batch)
             self.module.validation_step_end({
                 "epoch": epoch, "batch": i, "loss": float(loss),
                 "eval_cost": (time.time() - t0) / (i + 1)})
