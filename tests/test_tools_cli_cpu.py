@@ -70,3 +70,22 @@ def test_export_cli(tmp_path):
     assert any(f.endswith(".json") for f in files), files
     assert any("rank" in f or f.endswith((".safetensors", ".pdparams",
                                           ".pt")) for f in files), files
+
+
+@pytest.mark.timeout(300)
+def test_export_then_inference_cli(tmp_path):
+    # export a tiny model, then run tools/inference.py over the artifact
+    cmd = [sys.executable, os.path.join(REPO, "tools", "export.py"),
+           "-c", GEN_CFG, "--output-dir", str(tmp_path)]
+    for o in TINY_MODEL + ["Engine.mix_precision.enable=False",
+                           "Generation.eos_token_id=255",
+                           "Generation.max_dec_len=6"]:
+        cmd += ["-o", o]
+    _run(cmd)
+    cmd = [sys.executable, os.path.join(REPO, "tools", "inference.py"),
+           "-c", GEN_CFG, "--model-dir", str(tmp_path),
+           "--input-ids", "5,17,101"]
+    for o in ["Generation.eos_token_id=255", "Generation.max_dec_len=6"]:
+        cmd += ["-o", o]
+    out = _run(cmd)
+    assert "generated ids" in out
