@@ -51,6 +51,9 @@ MODELS = {
     "GPT-345M": dict(hidden_size=1024, num_layers=24, num_attention_heads=16),
     "GPT-1.3B": dict(hidden_size=2048, num_layers=24, num_attention_heads=16),
     "GPT-6.7B": dict(hidden_size=4096, num_layers=32, num_attention_heads=32),
+    # fits in ONE MI355X's 288 GB with full Adam state resident
+    # (params 26 + master 52 + m/v 104 + grads 26 GB) — no sharding
+    "GPT-13B": dict(hidden_size=5120, num_layers=40, num_attention_heads=40),
 }
 
 
