@@ -89,3 +89,21 @@ def test_export_then_inference_cli(tmp_path):
         cmd += ["-o", o]
     out = _run(cmd)
     assert "generated ids" in out
+
+
+@pytest.mark.timeout(300)
+def test_auto_cli():
+    cmd = [sys.executable, os.path.join(REPO, "tools", "auto.py"),
+           "-c", EVAL_CFG]
+    for o in TINY_MODEL + [
+            "Data.Train.dataset.seq_len=64",
+            "Data.Train.dataset.vocab_size=256",
+            "Data.Train.dataset.num_samples=32",
+            "Data.Train.loader.num_workers=0",
+            "Global.micro_batch_size=2", "Global.local_batch_size=4",
+            "Global.global_batch_size=4", "Global.max_steps=2",
+            "Global.logging_freq=1", "Global.eval_freq=",
+            "Global.save_steps=", "Engine.mix_precision.enable=False"]:
+        cmd += ["-o", o]
+    out = _run(cmd)
+    assert "ips_total" in out
