@@ -30,7 +30,8 @@ def parse_args():
     p.add_argument("--steps", type=int, default=10)
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--model", type=str, default="GPT-6.7B",
-                   choices=["GPT-TEST", "GPT-345M", "GPT-1.3B", "GPT-6.7B"])
+                   choices=["GPT-TEST", "GPT-345M", "GPT-1.3B", "GPT-6.7B",
+                            "GPT-13B"])
     p.add_argument("--micro-batch", type=int, default=None)
     p.add_argument("--acc-steps", type=int, default=None)
     p.add_argument("--seq-len", type=int, default=1024)
