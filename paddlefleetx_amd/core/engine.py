@@ -561,7 +561,8 @@ class EagerEngine(BasicEngine):
                 for p, off in zip(b.params, b.offsets):
                     b.model_flat[off:off + p.numel()].copy_(p.data.reshape(-1))
                     p.data = b.model_flat[off:off + p.numel()].view(p.shape)
-                b.master.copy_(b.model_flat.float())
+                b.master.copy_(
+                    b.model_flat[b.shard_lo:b.shard_hi].to(torch.float32))
         opt_path = os.path.join(path, "model_state.pdopt")
         if self.optimizer is not None and os.path.exists(opt_path):
             self.optimizer.load_state_dict(

@@ -75,7 +75,10 @@ class _Bucket:
         self.shard_len = total // shard_world
         self.shard_lo = shard_rank * self.shard_len
         self.shard_hi = self.shard_lo + self.shard_len
-        self.master = self.model_flat[self.shard_lo:self.shard_hi].float()
+        # copy=True: .float() on an fp32 flat would return the VIEW, and
+        # an aliased master breaks the checkpoint copy_ paths
+        self.master = self.model_flat[self.shard_lo:self.shard_hi].to(
+            torch.float32, copy=True)
         self.exp_avg = torch.zeros(self.shard_len, dtype=torch.float32,
                                    device=device)
         self.exp_avg_sq = torch.zeros(self.shard_len, dtype=torch.float32,
@@ -370,7 +373,9 @@ class FusedAdamW(torch.optim.Optimizer):
             b.master.copy_(s["master"])
             b.exp_avg.copy_(s["exp_avg"])
             b.exp_avg_sq.copy_(s["exp_avg_sq"])
-            b.model_flat.copy_(b.master.to(b.dtype))
+            # master covers only this rank's shard; the other slices of
+            # model_flat were already repacked from the model checkpoint
+            b.model_flat[b.shard_lo:b.shard_hi].copy_(b.master.to(b.dtype))
 
 
 class _MainGradHook:
