@@ -8,7 +8,8 @@ import numpy as np
 import torch
 from torch.utils.data import DataLoader
 
-from paddlefleetx_amd.data.gpt_dataset import GPTDataset, GPTSyntheticDataset
+from paddlefleetx_amd.data.gpt_dataset import (BlendedGPTDataset, GPTDataset,
+                                               GPTSyntheticDataset)
 from paddlefleetx_amd.data.sampler import GPTBatchSampler
 from paddlefleetx_amd.data.ernie_dataset import (ErnieSyntheticDataset,
                                                  ErnieWWMDataset)
@@ -20,6 +21,7 @@ from paddlefleetx_amd.utils.log import logger
 
 _DATASETS = {
     "GPTDataset": GPTDataset,
+    "BlendedGPTDataset": BlendedGPTDataset,
     "GPTSyntheticDataset": GPTSyntheticDataset,
     "SyntheticImageNetDataset": SyntheticImageNetDataset,
     "ImageFolderDataset": ImageFolderDataset,

@@ -68,10 +68,12 @@ class GPTDataset(Dataset):
 
     def __init__(self, input_dir: str, split=(949, 50, 1), mode: str = "Train",
                  max_seq_len: int = 1024, num_samples: int = 10000,
-                 seed: int = 1234, **unused):
-        files = sorted(f for f in os.listdir(input_dir) if f.endswith("_ids.npy"))
-        assert files, f"no *_ids.npy token bins under {input_dir}"
-        prefix = os.path.join(input_dir, files[0][:-len("_ids.npy")])
+                 seed: int = 1234, prefix: str = None, **unused):
+        if prefix is None:
+            files = sorted(f for f in os.listdir(input_dir)
+                           if f.endswith("_ids.npy"))
+            assert files, f"no *_ids.npy token bins under {input_dir}"
+            prefix = os.path.join(input_dir, files[0][:-len("_ids.npy")])
         self.token_ids = np.load(prefix + "_ids.npy", mmap_mode="r")
         idx = np.load(prefix + "_idx.npz")
         # lens[i] = tokens in doc i; docs[i] = start offset of doc i
@@ -133,3 +135,48 @@ class GPTDataset(Dataset):
         position_ids = torch.arange(self.seq_len, dtype=torch.int64)
         loss_mask = torch.ones(self.seq_len, dtype=torch.float32)
         return tokens, position_ids, labels, loss_mask
+
+
+class BlendedGPTDataset(Dataset):
+    """Weighted blend over every token-bin prefix under `input_dir`
+    (reference multi-dataset blending, fast_index_map_helpers.cpp:32-90
+    via data_tools blending): sample i comes from dataset
+    `dataset_index[i]`, drawn in proportion to `weights`."""
+
+    def __init__(self, input_dir: str, weights=None, split=(949, 50, 1),
+                 mode: str = "Train", max_seq_len: int = 1024,
+                 num_samples: int = 10000, seed: int = 1234, **unused):
+        from paddlefleetx_amd.data.index_builder import build_blending_indices
+        files = sorted(f for f in os.listdir(input_dir)
+                       if f.endswith("_ids.npy"))
+        assert files, f"no *_ids.npy token bins under {input_dir}"
+        prefixes = [os.path.join(input_dir, f[:-len("_ids.npy")])
+                    for f in files]
+        if weights is None:
+            weights = [1.0] * len(prefixes)
+        assert len(weights) == len(prefixes), \
+            f"{len(weights)} weights for {len(prefixes)} bin prefixes"
+        w = np.asarray(weights, dtype=np.float64)
+        w = w / w.sum()
+        self.children = []
+        for p, wi in zip(prefixes, w):
+            # oversample each child slightly (reference 1.005 margin) so
+            # the blend never runs a child dry
+            n_i = int(np.ceil(num_samples * float(wi) * 1.005)) + 1
+            self.children.append(GPTDataset(
+                input_dir, split=split, mode=mode, max_seq_len=max_seq_len,
+                num_samples=n_i, seed=seed, prefix=p))
+        self.dataset_index, self.dataset_sample_index = \
+            build_blending_indices(w, int(num_samples))
+        self.num_samples = int(num_samples)
+        self.seq_len = max_seq_len
+        logger.info(f"BlendedGPTDataset[{mode}]: {len(prefixes)} bins, "
+                    f"weights {np.round(w, 3).tolist()}, "
+                    f"{num_samples} samples")
+
+    def __len__(self):
+        return self.num_samples
+
+    def __getitem__(self, idx):
+        child = self.children[int(self.dataset_index[idx])]
+        return child[int(self.dataset_sample_index[idx]) % len(child)]

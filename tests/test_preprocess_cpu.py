@@ -85,3 +85,29 @@ def test_build_pair_mapping_cpp_matches_python():
     for s, e, t in got:
         assert e - s >= 2 and 4 <= t <= 128
         assert any(s >= lo and e <= hi for lo, hi in bounds)
+
+
+def test_blended_dataset_ratio_and_provenance(tmp_path):
+    """BlendedGPTDataset draws from every bin prefix in proportion to
+    the weights (reference multi-dataset blending)."""
+    d = tmp_path / "blend"
+    d.mkdir()
+    # two bins with distinguishable token values
+    for name, val in (("aaa", 1), ("bbb", 2)):
+        ids = np.full(2000, val, dtype=np.uint16)
+        np.save(d / f"{name}_ids.npy", ids)
+        np.savez(d / f"{name}_idx.npz",
+                 lens=np.full(20, 100, dtype=np.int64))
+    from paddlefleetx_amd.data.gpt_dataset import BlendedGPTDataset
+    ds = BlendedGPTDataset(str(d), weights=[3.0, 1.0], mode="Train",
+                           max_seq_len=16, num_samples=40)
+    assert len(ds) == 40
+    src = []
+    for i in range(40):
+        tokens, pos, labels, mask = ds[i]
+        assert tokens.shape == (16,)
+        vals = set(tokens.tolist())
+        assert vals in ({1}, {2}), vals  # never mixes bins inside a sample
+        src.append(tokens[0].item())
+    # 3:1 weighting -> 30/10 exactly (deterministic greedy blending)
+    assert src.count(1) == 30 and src.count(2) == 10
