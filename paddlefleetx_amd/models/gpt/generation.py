@@ -69,6 +69,7 @@ class GPTForGeneration(nn.Module):
         self.use_topp_sampling = bool(cfg.get("use_topp_sampling", False))
         self.eos_token_id = int(cfg.get("eos_token_id", 50256))
         self.pad_token_id = int(cfg.get("pad_token_id", 0))
+        self.num_return_sequences = int(cfg.get("num_return_sequences", 1))
 
     # -- one forward --------------------------------------------------------
     def _logits(self, input_ids, position_ids, caches):
@@ -132,6 +133,13 @@ class GPTForGeneration(nn.Module):
         was_training = self.gpt.training
         self.gpt.eval()
         with torch.no_grad():
+            if self.decode_strategy == "sampling" and \
+                    self.num_return_sequences > 1:
+                # expand_inputs_for_generation (single_model.py:1408-1412):
+                # each prompt row is repeated n times, so the output is
+                # [B*n, max_dec_len] grouped by prompt
+                input_ids = input_ids.repeat_interleave(
+                    self.num_return_sequences, dim=0)
             out = self.sample(input_ids)
         if was_training:
             self.gpt.train()

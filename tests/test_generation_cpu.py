@@ -135,3 +135,21 @@ def test_tokenizer_roundtrip_synthetic_vocab():
     assert padded["input_ids"][1] == [3, tok.pad_token_id, tok.pad_token_id,
                                       tok.pad_token_id]
     assert padded["attention_mask"][0] == [1, 1, 0, 0]
+
+
+def test_num_return_sequences_expansion():
+    """sampling + num_return_sequences=n returns [B*n, len] grouped by
+    prompt (reference expand_inputs_for_generation)."""
+    torch.manual_seed(5)
+    gpt = tiny_gpt().eval()
+    gen = GPTForGeneration(gpt, {"max_dec_len": 5, "top_k": 8,
+                                 "num_return_sequences": 3,
+                                 "eos_token_id": 127, "pad_token_id": 0})
+    out = gen(torch.randint(0, 127, (2, 4)))
+    assert out.shape[0] == 6 and out.shape[1] <= 5
+    # greedy ignores it (reference expands only the sampling branch)
+    gen2 = GPTForGeneration(gpt, {"max_dec_len": 5,
+                                  "decoding_strategy": "greedy_search",
+                                  "num_return_sequences": 3,
+                                  "eos_token_id": 127})
+    assert gen2(torch.randint(0, 127, (2, 4))).shape[0] == 2
