@@ -112,7 +112,7 @@ class SyntheticGLUEDataset(Dataset):
         return self.num_samples
 
     def __getitem__(self, idx):
-        g = torch.Generator().manual_seed(idx)
+        g = torch.Generator().manual_seed(int(idx))
         n = int(torch.randint(4, self.max_length, (1,), generator=g))
         ids = torch.randint(0, self.vocab_size, (self.max_length,),
                             generator=g)
