@@ -12,6 +12,7 @@ bench regressed e2e (ROADMAP) — per-shape long tuning + best-of merge is
 the safe form.
 """
 
+import argparse
 import os
 import sys
 
@@ -20,7 +21,11 @@ sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 import torch
 import torch.nn.functional as F
 
-M = 8192  # micro 8 x seq 1024
+_ap = argparse.ArgumentParser()
+_ap.add_argument("--m", type=int, default=8192,
+                 help="GEMM M = micro_batch x seq (8192=micro8, "
+                      "16384=micro16, 32768=micro32)")
+M = _ap.parse_args().m
 H = 4096
 SHAPES = [  # (out_features, in_features) of the 6.7B layer GEMMs
     (3 * H, H),        # qkv
