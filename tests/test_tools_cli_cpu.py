@@ -107,3 +107,13 @@ def test_auto_cli():
         cmd += ["-o", o]
     out = _run(cmd)
     assert "ips_total" in out
+
+
+@pytest.mark.timeout(300)
+def test_auto_export_cli(tmp_path):
+    cmd = [sys.executable, os.path.join(REPO, "tools", "auto_export.py"),
+           "-c", EVAL_CFG, "--output-dir", str(tmp_path)]
+    for o in TINY_MODEL + ["Engine.mix_precision.enable=False"]:
+        cmd += ["-o", o]
+    _run(cmd)
+    assert any(f.endswith(".json") for f in os.listdir(tmp_path))
