@@ -77,16 +77,18 @@ def main():
     from paddlefleetx_amd.models import build_module
     from paddlefleetx_amd.core import EagerEngine
 
-    # defaults from the measured 1-GPU sweep (profiles/): micro=8 keeps the
-    # hipBLASLt GEMMs at M=8192 where MFMA efficiency is ~20% higher than
-    # M=2048. Under pipeline parallel trade a little GEMM width for more
-    # micro-batches (1F1B bubble = (pp-1)/(acc+pp-1)).
+    # defaults from the measured 1-GPU sweep (profiles/): micro=16 puts
+    # the hipBLASLt GEMMs at M=16384 (same-box A/B vs micro8: 26234 vs
+    # 25904 tok/s; micro32 gains nothing more and doubles activation
+    # memory — gpurun_out/r2_m16_ab.txt, r2_m32.txt). Under pipeline
+    # parallel trade GEMM width for more micro-batches
+    # (1F1B bubble = (pp-1)/(acc+pp-1)).
     if pp > 1:
         micro = args.micro_batch or 4
         acc = args.acc_steps or 8
     else:
-        micro = args.micro_batch or 8
-        acc = args.acc_steps or 4
+        micro = args.micro_batch or 16
+        acc = args.acc_steps or 2
     local_bs = micro * acc
     seq = args.seq_len
     cfg_path = os.path.join(os.path.dirname(os.path.abspath(__file__)),
