@@ -86,6 +86,11 @@ def main():
     if pp > 1:
         micro = args.micro_batch or 4
         acc = args.acc_steps or 8
+    elif args.model == "GPT-13B":
+        # measured sweet spot (profiles/r02_13b_single_gpu.txt): micro4
+        # saturates the GEMMs; micro16 would overrun 288 GB
+        micro = args.micro_batch or 4
+        acc = args.acc_steps or 4
     else:
         micro = args.micro_batch or 16
         acc = args.acc_steps or 2
