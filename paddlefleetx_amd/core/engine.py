@@ -539,6 +539,8 @@ class EagerEngine(BasicEngine):
         meta = {"epoch": epoch, "step": step,
                 "global_step": self.module.global_step,
                 "lr_scheduler": self.lr_scheduler.state_dict(),
+                "loss_scale": self.loss_scale,
+                "good_steps": getattr(self, "_good_steps", 0),
                 "cpu_rng_state": torch.get_rng_state()}
         if torch.cuda.is_available():
             meta["cuda_rng_state"] = torch.cuda.get_rng_state()
@@ -575,6 +577,11 @@ class EagerEngine(BasicEngine):
             self.module.global_step = meta.get("global_step", meta["step"])
             if "lr_scheduler" in meta:
                 self.lr_scheduler.load_state_dict(meta["lr_scheduler"])
+            if "loss_scale" in meta and self.loss_scale != 1.0:
+                # resume the dynamic fp16 scale where it left off (a
+                # reset to the 32768 default would burn found_inf skips)
+                self.loss_scale = float(meta["loss_scale"])
+                self._good_steps = int(meta.get("good_steps", 0))
             if "cpu_rng_state" in meta:
                 torch.set_rng_state(meta["cpu_rng_state"])
             if "cuda_rng_state" in meta and torch.cuda.is_available():

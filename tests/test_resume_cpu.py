@@ -81,3 +81,29 @@ def test_resume_matches_continuous_run():
     got = c.module.model.state_dict()
     for k, v in ref.items():
         assert torch.allclose(got[k].float(), v.float(), atol=1e-6), k
+
+
+def test_resume_restores_loss_scale():
+    """Dynamic fp16 loss scale survives save/load (a reset to the 32768
+    default would burn found_inf skip steps after resume)."""
+    set_hcg(HybridTopology())
+    cfg = _cfg({"max_steps": 2,
+                "mix_precision": {"enable": True, "dtype": "float16",
+                                  "scale_loss": 4096.0}})
+    with tempfile.TemporaryDirectory() as td:
+        cfg["Engine"]["save_load"] = {"output_dir": td}
+        a = _build(cfg)
+        a.loss_scale = 512.0   # pretend the dynamic scaler backed off
+        a._good_steps = 7
+        a.fit(_batches(2))
+        a.save(0, 2)
+
+        cfg2 = _cfg({"max_steps": 4,
+                     "mix_precision": {"enable": True, "dtype": "float16",
+                                       "scale_loss": 4096.0},
+                     "save_load": {"output_dir": td,
+                                   "ckpt_dir": os.path.join(
+                                       td, "epoch_0_step_2")}})
+        b = _build(cfg2)
+        assert b.loss_scale == a.loss_scale
+        assert b._good_steps == a._good_steps
