@@ -565,6 +565,16 @@ class EagerEngine(BasicEngine):
                     p.data = b.model_flat[off:off + p.numel()].view(p.shape)
                 b.master.copy_(
                     b.model_flat[b.shard_lo:b.shard_hi].to(torch.float32))
+        else:
+            from paddlefleetx_amd.parallel.zero3 import Stage3AdamW
+            if isinstance(self.optimizer, Stage3AdamW):
+                # shards were refreshed by the wrapper's load_state_dict;
+                # re-derive the fp32 masters (the optimizer checkpoint, if
+                # present, overwrites them exactly below)
+                for u, st in zip(self.optimizer.w.units,
+                                 self.optimizer.state):
+                    st["master"].copy_(u.shard.to(st["master"].device)
+                                       .float())
         opt_path = os.path.join(path, "model_state.pdopt")
         if self.optimizer is not None and os.path.exists(opt_path):
             self.optimizer.load_state_dict(

@@ -287,6 +287,17 @@ class GroupShardedStage3(nn.Module):
         self.gather_full_params()
         return self.model.state_dict(*a, **k)
 
+    def load_state_dict(self, sd, strict: bool = True):
+        """Gather, copy the full state in-place (params view the flat
+        buffers, so the copy lands there), then refresh every unit's
+        persistent shard from its slice of the loaded flat."""
+        self.gather_full_params()
+        ret = self.model.load_state_dict(sd, strict=strict)
+        for u in self.units:
+            u.shard.copy_(u.flat[u.rank * u.shard_len:
+                                 (u.rank + 1) * u.shard_len])
+        return ret
+
 
 class Stage3AdamW:
     """AdamW on the per-unit fp32 master shards (fused HIP kernel)."""
