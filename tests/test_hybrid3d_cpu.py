@@ -87,7 +87,7 @@ def test_dp2_tp2_pp2_vpp2_world8():
         assert p.exitcode == 0, f"worker failed with {p.exitcode}"
 
 
-def _ckpt_worker(rank, world, port, tmpdir):
+def _ckpt_worker(rank, world, port, tmpdir, vpp=1):
     import sys
     sys.path.insert(0, REPO)
     os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
@@ -113,7 +113,8 @@ def _ckpt_worker(rank, world, port, tmpdir):
                   "attention_probs_dropout_prob": 0.0, "fused_attn": False},
         "Optimizer": {"name": "FusedAdamW", "weight_decay": 0.0,
                       "lr": {"name": "ConstantLR", "learning_rate": 1e-3}},
-        "Distributed": {"pp_degree": 2},
+        "Distributed": {"pp_degree": 2,
+                        "pipeline": {"virtual_pp_degree": vpp}},
     }
     module = build_module(cfg)
     engine = EagerEngine(cfg, module)
@@ -218,3 +219,21 @@ def test_fp16_scaler_with_pipeline():
     for p in procs:
         p.join(300)
         assert p.exitcode == 0, f"worker failed with {p.exitcode}"
+
+
+@pytest.mark.timeout(600)
+def test_pp2_vpp2_checkpoint_roundtrip():
+    """Interleaved virtual stages save/load through the same shard-dir
+    layout (each rank holds two non-adjacent chunks)."""
+    import tempfile
+    from port_util import free_port
+    port = free_port()
+    ctx = mp.get_context("spawn")
+    with tempfile.TemporaryDirectory() as tmpdir:
+        procs = [ctx.Process(target=_ckpt_worker, args=(r, 2, port, tmpdir, 2))
+                 for r in range(2)]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(300)
+            assert p.exitcode == 0, f"worker failed with {p.exitcode}"
