@@ -60,6 +60,7 @@ class _FusedWgradLinear(torch.autograd.Function):
         return F.linear(x, weight, bias)
 
     @staticmethod
+    @torch.autograd.function.once_differentiable
     def backward(ctx, dy):
         x, w = ctx.saved_tensors
         dyc = dy.contiguous()
@@ -134,6 +135,7 @@ class _FusedBiasAdd(torch.autograd.Function):
         return y + bias
 
     @staticmethod
+    @torch.autograd.function.once_differentiable
     def backward(ctx, dy):
         bref = ctx.bias_ref
         bg = _grad_view(bref)
