@@ -111,3 +111,28 @@ def test_blended_dataset_ratio_and_provenance(tmp_path):
         src.append(tokens[0].item())
     # 3:1 weighting -> 30/10 exactly (deterministic greedy blending)
     assert src.count(1) == 30 and src.count(2) == 10
+
+
+def test_blended_dataset_via_build_dataloader(tmp_path):
+    """The blended dataset is addressable from the YAML Data section."""
+    d = tmp_path / "blend2"
+    d.mkdir()
+    for name, val in (("x", 3), ("y", 4)):
+        np.save(d / f"{name}_ids.npy", np.full(2000, val, dtype=np.uint16))
+        np.savez(d / f"{name}_idx.npz",
+                 lens=np.full(20, 100, dtype=np.int64))
+    from paddlefleetx_amd.data import build_dataloader
+    from paddlefleetx_amd.parallel.env import set_hcg
+    from paddlefleetx_amd.parallel.topology import HybridTopology
+    set_hcg(HybridTopology())
+    cfg = {"Train": {"dataset": {"name": "BlendedGPTDataset",
+                                 "input_dir": str(d),
+                                 "weights": [1.0, 1.0],
+                                 "max_seq_len": 16,
+                                 "num_samples": 16},
+                     "sampler": {"shuffle": False, "drop_last": True},
+                     "loader": {"num_workers": 0}}}
+    loader = build_dataloader(cfg, "Train", batch_size=4)
+    tokens, pos, labels, mask = next(iter(loader))
+    assert tokens.shape == (4, 16)
+    assert set(tokens.unique().tolist()) <= {3, 4}
