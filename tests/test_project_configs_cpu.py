@@ -84,6 +84,7 @@ def test_moco_config_one_step():
     "nlp/gpt/qat_gpt_345M_single_card.yaml",
     "nlp/gpt/prune_gpt_345M_single_card.yaml",
     "nlp/gpt/pretrain_gpt_1.3B_single_card.yaml",
+    "nlp/gpt/pretrain_gpt_13B_single_card.yaml",
 ])
 def test_gpt_variant_configs_load(rel):
     cfg = _cfg(rel, ["Model.hidden_size=64", "Model.num_layers=2",
