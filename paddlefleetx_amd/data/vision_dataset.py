@@ -54,12 +54,16 @@ class ImageFolderDataset(Dataset):
 
     def __init__(self, root: str, image_size: int = 224, mode: str = "Train",
                  rand_augment: bool = False, ra_num_ops: int = 2,
-                 ra_magnitude: int = 9, **unused):
+                 ra_magnitude: int = 9, transforms=None, **unused):
         self.root = root
         self.image_size = image_size
         self.rand_augment = None
         if rand_augment and mode == "Train":
             self.rand_augment = RandAugment(ra_num_ops, ra_magnitude)
+        self.transforms = None
+        if transforms:
+            from paddlefleetx_amd.data.transforms import build_transforms
+            self.transforms = build_transforms(transforms)
         classes = sorted(d for d in os.listdir(root)
                          if os.path.isdir(os.path.join(root, d)))
         self.class_to_idx = {c: i for i, c in enumerate(classes)}
@@ -87,6 +91,8 @@ class ImageFolderDataset(Dataset):
                     .float() / 255.0
             except ImportError as e:
                 raise RuntimeError(f"cannot decode {path}: PIL missing") from e
+        if self.transforms is not None:
+            img = self.transforms(img)
         if self.rand_augment is not None:
             img = self.rand_augment(img)
         return img, label
