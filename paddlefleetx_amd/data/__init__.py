@@ -13,8 +13,8 @@ from paddlefleetx_amd.data.gpt_dataset import (BlendedGPTDataset, GPTDataset,
 from paddlefleetx_amd.data.sampler import GPTBatchSampler
 from paddlefleetx_amd.data.ernie_dataset import (ErnieSyntheticDataset,
                                                  ErnieWWMDataset)
-from paddlefleetx_amd.data.vision_dataset import (ImageFolderDataset,
-                                                  SyntheticImageNetDataset)
+from paddlefleetx_amd.data.vision_dataset import (
+    ContrativeLearningDataset, ImageFolderDataset, SyntheticImageNetDataset)
 from paddlefleetx_amd.parallel.env import (get_data_world_rank,
                                            get_data_world_size)
 from paddlefleetx_amd.utils.log import logger
@@ -26,6 +26,7 @@ _DATASETS = {
     "SyntheticImageNetDataset": SyntheticImageNetDataset,
     "ImageFolderDataset": ImageFolderDataset,
     "GeneralClsDataset": ImageFolderDataset,
+    "ContrativeLearningDataset": ContrativeLearningDataset,
     "ErnieSyntheticDataset": ErnieSyntheticDataset,
     "ErnieWWMDataset": ErnieWWMDataset,
 }

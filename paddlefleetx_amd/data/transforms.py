@@ -70,7 +70,7 @@ class ResizeImage:
 
 
 class CenterCropImage:
-    def __init__(self, size):
+    def __init__(self, size, **unused):
         self.size = (size, size) if isinstance(size, int) else size
 
     def __call__(self, img: torch.Tensor) -> torch.Tensor:
@@ -119,7 +119,7 @@ class RandCropImage:
 
 
 class RandFlipImage:
-    def __init__(self, flip_code: int = 1, generator=None):
+    def __init__(self, flip_code: int = 1, generator=None, **unused):
         self.flip_code = flip_code  # 1: horizontal (reference :219)
         self.gen = generator
 
@@ -163,8 +163,9 @@ class ColorJitter:
     via a luma-preserving channel rotation approximation."""
 
     def __init__(self, brightness=0.0, contrast=0.0, saturation=0.0,
-                 hue=0.0, generator=None):
+                 hue=0.0, p: float = 1.0, generator=None, **unused):
         self.b, self.c, self.s, self.h = brightness, contrast, saturation, hue
+        self.p = float(p)  # RandomApply probability (reference moco v2)
         self.gen = generator
 
     def _f(self, mag):
@@ -173,6 +174,8 @@ class ColorJitter:
 
     def __call__(self, img: torch.Tensor) -> torch.Tensor:
         img = _chw(img)
+        if self.p < 1.0 and float(torch.rand((), generator=self.gen)) >= self.p:
+            return img
         if self.b:
             img = (img * self._f(self.b)).clamp(0, 1)
         if self.c:
@@ -194,14 +197,17 @@ class ColorJitter:
 
 class GaussianBlur:
     def __init__(self, sigma=(0.1, 2.0), kernel_size: int = 9,
-                 generator=None):
+                 p: float = 1.0, generator=None, **unused):
         self.sigma = sigma if isinstance(sigma, (tuple, list)) \
             else (sigma, sigma)
         self.k = kernel_size | 1  # odd
+        self.p = float(p)
         self.gen = generator
 
     def __call__(self, img: torch.Tensor) -> torch.Tensor:
         img = _chw(img)
+        if self.p < 1.0 and float(torch.rand((), generator=self.gen)) >= self.p:
+            return img
         lo, hi = self.sigma
         s = lo + (hi - lo) * float(torch.rand((), generator=self.gen))
         half = self.k // 2

@@ -54,13 +54,15 @@ class ImageFolderDataset(Dataset):
 
     def __init__(self, root: str, image_size: int = 224, mode: str = "Train",
                  rand_augment: bool = False, ra_num_ops: int = 2,
-                 ra_magnitude: int = 9, transforms=None, **unused):
+                 ra_magnitude: int = 9, transforms=None,
+                 transform_ops=None, **unused):
         self.root = root
         self.image_size = image_size
         self.rand_augment = None
         if rand_augment and mode == "Train":
             self.rand_augment = RandAugment(ra_num_ops, ra_magnitude)
         self.transforms = None
+        transforms = transforms or transform_ops  # reference key name
         if transforms:
             from paddlefleetx_amd.data.transforms import build_transforms
             self.transforms = build_transforms(transforms)
@@ -229,3 +231,26 @@ class RandAugment:
             sign = 1.0 if float(self._rand(1)) < 0.5 else -1.0
             img = self._apply(img, op, sign)
         return img
+
+
+class ContrativeLearningDataset(ImageFolderDataset):
+    """Two-view contrastive dataset (reference vision_dataset.py:379,
+    keeping its spelling): each sample is ((view_q, view_k), label) where
+    both views are independent draws of the transform pipeline."""
+
+    def __getitem__(self, idx):
+        path, label = self.samples[idx]
+        base = super().__getitem__(idx)[0] if self.transforms is None \
+            else None
+        if self.transforms is None:
+            # no pipeline configured: jitter the second view slightly
+            return (base, base + 0.05 * torch.randn_like(base)), label
+        if path.endswith(".npy"):
+            img = torch.from_numpy(np.load(path)).float()
+        else:
+            from PIL import Image
+            im = Image.open(path).convert("RGB").resize(
+                (self.image_size, self.image_size))
+            img = torch.from_numpy(np.asarray(im)).permute(2, 0, 1) \
+                .float() / 255.0
+        return (self.transforms(img), self.transforms(img)), label

@@ -86,3 +86,18 @@ def test_build_transforms_from_config():
         raise AssertionError("expected ValueError")
     except ValueError:
         pass
+
+
+def test_contrastive_dataset_two_views(tmp_path):
+    import os
+    from paddlefleetx_amd.data.vision_dataset import ContrativeLearningDataset
+    os.makedirs(tmp_path / "cls0")
+    np.save(tmp_path / "cls0" / "a.npy",
+            np.random.rand(3, 32, 32).astype("float32"))
+    ds = ContrativeLearningDataset(str(tmp_path), transform_ops=[
+        {"RandCropImage": {"size": 24}},
+        {"ColorJitter": {"brightness": 0.4, "p": 0.8}},
+    ])
+    (q, k), label = ds[0]
+    assert q.shape == (3, 24, 24) and k.shape == (3, 24, 24)
+    assert label == 0
