@@ -14,7 +14,8 @@ from paddlefleetx_amd.data.sampler import GPTBatchSampler
 from paddlefleetx_amd.data.ernie_dataset import (ErnieSyntheticDataset,
                                                  ErnieWWMDataset)
 from paddlefleetx_amd.data.vision_dataset import (
-    ContrativeLearningDataset, ImageFolderDataset, SyntheticImageNetDataset)
+    CIFAR10Dataset, ContrativeLearningDataset, ImageFolderDataset,
+    SyntheticImageNetDataset)
 from paddlefleetx_amd.parallel.env import (get_data_world_rank,
                                            get_data_world_size)
 from paddlefleetx_amd.utils.log import logger
@@ -27,6 +28,8 @@ _DATASETS = {
     "ImageFolderDataset": ImageFolderDataset,
     "GeneralClsDataset": ImageFolderDataset,
     "ContrativeLearningDataset": ContrativeLearningDataset,
+    "CIFAR10": CIFAR10Dataset,
+    "CIFAR10Dataset": CIFAR10Dataset,
     "ErnieSyntheticDataset": ErnieSyntheticDataset,
     "ErnieWWMDataset": ErnieWWMDataset,
 }

@@ -254,3 +254,44 @@ class ContrativeLearningDataset(ImageFolderDataset):
             img = torch.from_numpy(np.asarray(im)).permute(2, 0, 1) \
                 .float() / 255.0
         return (self.transforms(img), self.transforms(img)), label
+
+
+class CIFAR10Dataset(Dataset):
+    """CIFAR-10 from the standard python-pickle batches on local disk
+    (reference vision_dataset.py:302; no download here — point `root` at
+    an extracted cifar-10-batches-py directory)."""
+
+    collate_fn = None
+
+    def __init__(self, root: str, mode: str = "Train", transforms=None,
+                 transform_ops=None, **unused):
+        import pickle
+        names = [f"data_batch_{i}" for i in range(1, 6)] \
+            if mode == "Train" else ["test_batch"]
+        datas, labels = [], []
+        for n in names:
+            path = os.path.join(root, n)
+            if not os.path.exists(path):
+                continue
+            with open(path, "rb") as f:
+                d = pickle.load(f, encoding="bytes")
+            datas.append(np.asarray(d[b"data"] if b"data" in d
+                                    else d["data"]))
+            labels.extend(d[b"labels"] if b"labels" in d else d["labels"])
+        assert datas, f"no CIFAR-10 batches under {root}"
+        self.images = np.concatenate(datas).reshape(-1, 3, 32, 32)
+        self.labels = np.asarray(labels, dtype=np.int64)
+        self.transforms = None
+        ops = transforms or transform_ops
+        if ops:
+            from paddlefleetx_amd.data.transforms import build_transforms
+            self.transforms = build_transforms(ops)
+
+    def __len__(self):
+        return len(self.labels)
+
+    def __getitem__(self, idx):
+        img = torch.from_numpy(self.images[idx].astype(np.float32) / 255.0)
+        if self.transforms is not None:
+            img = self.transforms(img)
+        return img, int(self.labels[idx])

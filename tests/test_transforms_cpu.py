@@ -101,3 +101,17 @@ def test_contrastive_dataset_two_views(tmp_path):
     (q, k), label = ds[0]
     assert q.shape == (3, 24, 24) and k.shape == (3, 24, 24)
     assert label == 0
+
+
+def test_cifar10_dataset(tmp_path):
+    import pickle
+    from paddlefleetx_amd.data.vision_dataset import CIFAR10Dataset
+    data = (np.random.rand(20, 3072) * 255).astype(np.uint8)
+    with open(tmp_path / "data_batch_1", "wb") as f:
+        pickle.dump({b"data": data, b"labels": list(range(10)) * 2}, f)
+    ds = CIFAR10Dataset(str(tmp_path), mode="Train",
+                        transform_ops=[{"RandFlipImage": {}}])
+    assert len(ds) == 20
+    img, label = ds[3]
+    assert img.shape == (3, 32, 32) and 0 <= label < 10
+    assert img.max() <= 1.0
