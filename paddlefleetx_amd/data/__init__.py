@@ -47,6 +47,9 @@ def _register_lazy():
     from paddlefleetx_amd.data.multimodal_dataset import \
         SyntheticImagenDataset
     _DATASETS.setdefault("SyntheticImagenDataset", SyntheticImagenDataset)
+    from paddlefleetx_amd.data.multimodal_dataset import ImagenFileDataset
+    _DATASETS.setdefault("ImagenDataset", ImagenFileDataset)
+    _DATASETS.setdefault("ImagenFileDataset", ImagenFileDataset)
     from paddlefleetx_amd.data.folding_dataset import SyntheticFoldingDataset
     _DATASETS.setdefault("SyntheticFoldingDataset", SyntheticFoldingDataset)
 

@@ -190,3 +190,23 @@ def test_imagen_cascade_train_and_sample():
     casc.eval()
     out = casc.sample(text_ids=ids, batch_size=2, steps=2)
     assert out.shape == (2, 3, 32, 32)
+
+
+def test_imagen_file_dataset(tmp_path):
+    """Filelist-sharded Imagen reader (reference ImagenDataset surface)."""
+    import numpy as np
+    import os
+    from paddlefleetx_amd.data.multimodal_dataset import (ImagenFileDataset,
+                                                          get_keys)
+    np.save(tmp_path / "im.npy", np.random.rand(3, 64, 64).astype("float32"))
+    (tmp_path / "shard0.tsv").write_text("im.npy\tan example caption\n")
+    (tmp_path / "filelist.txt").write_text("shard0.tsv\n")
+    ds = ImagenFileDataset(str(tmp_path / "filelist.txt"), text_max_len=8,
+                           rank=0)
+    img, ids, mask = ds[0]
+    assert img.shape == (3, 64, 64)
+    assert float(img.min()) >= -1.0 and float(img.max()) <= 1.0
+    assert int(mask.sum()) == 3  # three caption words
+    # rank sharding: 1 shard over 2 ranks pads, each rank sees 1 file
+    assert len(get_keys(str(tmp_path / "filelist.txt"), 2, rank=0)) == 1
+    assert len(get_keys(str(tmp_path / "filelist.txt"), 2, rank=1)) == 1
