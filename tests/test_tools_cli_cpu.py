@@ -117,3 +117,19 @@ def test_auto_export_cli(tmp_path):
         cmd += ["-o", o]
     _run(cmd)
     assert any(f.endswith(".json") for f in os.listdir(tmp_path))
+
+
+@pytest.mark.timeout(300)
+def test_inference_benchmark_cli(tmp_path):
+    cmd = [sys.executable, os.path.join(REPO, "tools", "export.py"),
+           "-c", GEN_CFG, "--output-dir", str(tmp_path)]
+    for o in TINY_MODEL + ["Engine.mix_precision.enable=False",
+                           "Generation.eos_token_id=255",
+                           "Generation.max_dec_len=4"]:
+        cmd += ["-o", o]
+    _run(cmd)
+    out = _run([sys.executable,
+                os.path.join(REPO, "projects/gpt/benchmark.py"),
+                "--model-dir", str(tmp_path), "--seq-len", "8",
+                "--iter", "2"])
+    assert "run time" in out and "ms/token" in out
