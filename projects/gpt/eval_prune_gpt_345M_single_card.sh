@@ -1,0 +1,4 @@
+#!/usr/bin/env bash
+# reference projects/gpt/eval_prune_gpt_345M_single_card.sh
+cd "$(dirname "$0")/../.."
+python tools/eval.py -c paddlefleetx_amd/configs/nlp/gpt/prune_gpt_345M_single_card.yaml "$@"

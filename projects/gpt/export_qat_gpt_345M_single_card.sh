@@ -1,0 +1,4 @@
+#!/usr/bin/env bash
+# reference projects/gpt/export_qat_gpt_345M_single_card.sh
+cd "$(dirname "$0")/../.."
+python tools/export.py -c paddlefleetx_amd/configs/nlp/gpt/qat_gpt_345M_single_card.yaml "$@"
