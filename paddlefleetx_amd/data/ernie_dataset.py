@@ -50,12 +50,17 @@ class ErnieSyntheticDataset(Dataset):
 
     def __init__(self, num_samples: int = 10000, seq_len: int = 512,
                  vocab_size: int = 40000, masked_lm_prob: float = 0.15,
-                 mode: str = "Train", seed: int = 1234, **unused):
+                 mode: str = "Train", seed: int = 1234,
+                 seq_cls: bool = False, num_classes: int = 2, **unused):
         self.num_samples = int(num_samples)
         self.seq_len = int(seq_len)
         self.vocab_size = int(vocab_size)
         self.masked_lm_prob = masked_lm_prob
         self.seed = seed
+        # finetune form: (input_ids, token_type_ids, label) for
+        # ErnieSeqClsModule (reference finetune_ernie yamls)
+        self.seq_cls = bool(seq_cls)
+        self.num_classes = int(num_classes)
 
     def __len__(self):
         return self.num_samples
@@ -73,6 +78,12 @@ class ErnieSyntheticDataset(Dataset):
         token_types = np.concatenate([
             np.zeros(len_a + 2, dtype=np.int64),
             np.ones(len_b + 1, dtype=np.int64)])
+        if self.seq_cls:
+            # separable synthetic task: the label is carried by the
+            # parity of the first real token
+            label = int(tokens[1] % self.num_classes)
+            return (torch.from_numpy(tokens),
+                    torch.from_numpy(token_types), torch.tensor(label))
         masked, labels = create_masked_lm_predictions(
             tokens, self.vocab_size, rng,
             masked_lm_prob=self.masked_lm_prob, mask_token_id=self.MASK,
@@ -81,7 +92,8 @@ class ErnieSyntheticDataset(Dataset):
                 torch.from_numpy(labels), torch.tensor(nsp))
 
 def ernie_collate_fn(samples):
-    return tuple(torch.stack([s[i] for s in samples]) for i in range(4))
+    n = len(samples[0])
+    return tuple(torch.stack([s[i] for s in samples]) for i in range(n))
 
 
 # build_dataloader picks this up as the dataset's collate
