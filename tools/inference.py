@@ -31,9 +31,16 @@ def main():
     args = parse_args()
     cfg = get_config(args.config, overrides=args.override, show=True)
     init_dist_env(cfg)
-    mp = int(cfg.get("Distributed", {}).get("mp_degree", 1) or 1)
-    engine = InferenceEngine(args.model_dir, mp_degree=mp,
-                             generation_cfg=cfg.get("Generation"))
+    inf = cfg.get("Inference", {}) or {}
+    model_dir = args.model_dir if args.model_dir != "./exported_model" \
+        else inf.get("model_dir", args.model_dir)
+    mp = int(inf.get("mp_degree") or
+             cfg.get("Distributed", {}).get("mp_degree", 1) or 1)
+    gen = dict(cfg.get("Generation") or {})
+    if inf.get("fp8"):
+        gen["fp8"] = True  # opt-in serving fp8 (InferenceEngine)
+    engine = InferenceEngine(model_dir, mp_degree=mp,
+                             generation_cfg=gen or None)
     ids = [int(t) for t in args.input_ids.split(",")]
     out = engine.predict(ids)
     logger.info(f"input ids: {ids}")
