@@ -34,7 +34,8 @@ def main():
     valid_loader = build_dataloader(engine.configs, "Eval") \
         if "Eval" in engine.configs.get("Data", {}) else None
     if args.tune and train_loader is not None:
-        engine.tune(train_loader)
+        cands = (cfg.get("Tuning", {}) or {}).get("candidates")
+        engine.tune(train_loader, candidates=cands)
     engine.fit(train_loader, valid_loader)
 
 
