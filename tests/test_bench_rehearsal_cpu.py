@@ -52,3 +52,27 @@ def test_bench_world8_dp2tp2pp2():
     rec = _run_bench(8)
     assert rec["config"]["parallelism"] == "dp2tp2pp2"
     assert rec["config"]["loss"] is not None
+
+
+def test_bench_defaults_contract():
+    """Driver-contract guard: model table, topology mapping and the
+    measured micro/acc defaults must not drift."""
+    import importlib.util
+    spec = importlib.util.spec_from_file_location(
+        "bench", os.path.join(REPO, "bench.py"))
+    bench = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(bench)
+    assert bench.topology_for(1) == (1, 1, 1)
+    assert bench.topology_for(2) == (2, 1, 1)
+    assert bench.topology_for(4) == (4, 1, 1)
+    assert bench.topology_for(8) == (2, 2, 2)
+    assert bench.MODELS["GPT-6.7B"] == dict(hidden_size=4096, num_layers=32,
+                                            num_attention_heads=32)
+    assert bench.MODELS["GPT-13B"] == dict(hidden_size=5120, num_layers=40,
+                                           num_attention_heads=40)
+    src = open(os.path.join(REPO, "bench.py")).read()
+    # measured defaults (profiles/r02_micro16_ab.txt): micro16/acc2 at
+    # pp==1, micro4/acc8 under pipeline, 13B capped at micro4
+    assert "args.micro_batch or 16" in src
+    assert "args.acc_steps or 2" in src
+    assert "args.micro_batch or 4" in src
