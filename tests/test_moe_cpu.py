@@ -258,3 +258,37 @@ def test_sharded_moe_matches_dense_expert_mixture():
         want[t] = layer.experts[e](xf[t:t + 1])[0] * probs[t, e]
     assert torch.allclose(y.reshape(-1, 8), want, atol=1e-5), \
         (y.reshape(-1, 8) - want).abs().max()
+
+
+def test_moe_random_routing_under_recompute():
+    """Activation recompute must REPLAY the gshard gate's random
+    second-expert sampling (and dropout): grads with recompute equal
+    grads without, at identical seeds — checkpoint's preserve-RNG path
+    is what makes MoE + recompute sound."""
+    from paddlefleetx_amd.models.gpt.model import (GPTForPretraining,
+                                                   GPTModel,
+                                                   GPTPretrainingCriterion)
+    from paddlefleetx_amd.parallel.env import set_seed
+    moe_cfg = {"expert_mode": True, "num_experts": 4, "gate": "gshard",
+               "top_k": 2}
+    grads = {}
+    for rec in (False, True):
+        set_seed(1234)
+        torch.manual_seed(3)
+        m = GPTForPretraining(GPTModel(
+            vocab_size=128, hidden_size=32, num_layers=2,
+            num_attention_heads=4, max_position_embeddings=32,
+            fused_attn=False, hidden_dropout_prob=0.1,
+            attention_probs_dropout_prob=0.0,  # attn dropout uses the
+            # mp-rng tracker; hidden dropout + routing use the global
+            # stream that checkpoint preserves
+            use_recompute=rec, moe_configs=moe_cfg))
+        m.train()
+        torch.manual_seed(7)
+        tokens = torch.randint(0, 128, (2, 16))
+        labels = torch.randint(0, 128, (2, 16))
+        loss = GPTPretrainingCriterion()(m(tokens), labels,
+                                         torch.ones(2, 16))
+        loss.backward()
+        grads[rec] = m.gpt.layers[0].attn.qkv.weight.grad.clone()
+    assert torch.allclose(grads[False], grads[True], atol=1e-6)
