@@ -26,7 +26,8 @@ from paddlefleetx_amd.ops import (FusedLayerNorm, bias_gelu, flash_attention,
                                   flash_attention_packed,
                                   fused_softmax_causal)
 from paddlefleetx_amd.parallel.env import get_hcg
-from paddlefleetx_amd.parallel.rng import model_parallel_rng
+from paddlefleetx_amd.parallel.rng import (checkpoint_rng_context,
+                                            model_parallel_rng)
 from paddlefleetx_amd.parallel.tp import (ColumnParallelLinear,
                                           ParallelCrossEntropy,
                                           RowParallelLinear,
@@ -267,7 +268,8 @@ class TransformerDecoderLayer(nn.Module):
             # recompute only the attention core (QK^T/softmax/PV + projs);
             # reference granularity "core_attn" (hybrid_model.py:303-346)
             a = checkpoint(lambda t: self.attn(t)[0], h,
-                           use_reentrant=False)
+                           use_reentrant=False,
+                           context_fn=checkpoint_rng_context)
             return self._dropout(a), None
         a, new_cache = self.attn(h, cache=cache, use_cache=use_cache)
         return self._dropout(a), new_cache
@@ -278,7 +280,9 @@ class TransformerDecoderLayer(nn.Module):
     def forward(self, x, cache: Optional[KVCache] = None, use_cache: bool = False):
         if self.use_recompute and self.recompute_granularity == "full_attn" \
                 and self.training and not use_cache and torch.is_grad_enabled():
-            a = checkpoint(self._attn_branch_nocache, x, use_reentrant=False)
+            a = checkpoint(self._attn_branch_nocache, x,
+                           use_reentrant=False,
+                           context_fn=checkpoint_rng_context)
             new_cache = None
         else:
             a, new_cache = self._attn_branch(x, cache, use_cache)
@@ -393,7 +397,8 @@ class GPTModel(nn.Module):
             if (self.use_recompute and self.recompute_granularity == "full"
                     and self.training and not use_cache
                     and torch.is_grad_enabled()):
-                x = checkpoint(layer, x, use_reentrant=False)
+                x = checkpoint(layer, x, use_reentrant=False,
+                               context_fn=checkpoint_rng_context)
             elif use_cache:
                 x, c = layer(x, cache=cache_i, use_cache=True)
                 new_caches.append(c)

@@ -24,6 +24,25 @@ class EmbeddingPipe(GPTEmbeddings):
         return super().forward(input_ids, position_ids)
 
 
+class TransformerDecoderLayerPipe(TransformerDecoderLayer):
+    """Layer-level ("full") recompute inside the pipeline (reference
+    PipelineLayer recompute_interval, hybrid_model.py:1182-1188): the
+    single-path model checkpoints layers in GPTModel's loop, which the
+    pipeline never runs — so the pipe layer checkpoints ITSELF. The
+    rng context replays the mp-tracker dropout streams exactly."""
+
+    def forward(self, x, cache=None, use_cache: bool = False):
+        if self.use_recompute and self.recompute_granularity == "full" \
+                and self.training and not use_cache \
+                and torch.is_grad_enabled():
+            from torch.utils.checkpoint import checkpoint
+
+            from paddlefleetx_amd.parallel.rng import checkpoint_rng_context
+            return checkpoint(super().forward, x, use_reentrant=False,
+                              context_fn=checkpoint_rng_context)
+        return super().forward(x, cache=cache, use_cache=use_cache)
+
+
 class LayerNormPipe(FusedLayerNorm):
     def __init__(self, *args, sequence_parallel: bool = False, **kwargs):
         super().__init__(*args, **kwargs)
@@ -88,7 +107,7 @@ class GPTForPretrainingPipe(PipelineModule):
         ]
         for _ in range(num_layers):
             descs.append(LayerDesc(
-                TransformerDecoderLayer, hidden_size, num_attention_heads,
+                TransformerDecoderLayerPipe, hidden_size, num_attention_heads,
                 ffn_hidden_size, hidden_dropout=hidden_dropout_prob,
                 attn_dropout=attention_probs_dropout_prob,
                 fused_attn=fused_attn, dtype=dtype,

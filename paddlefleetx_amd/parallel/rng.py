@@ -74,3 +74,40 @@ def get_rng_tracker() -> RNGStateTracker:
 def model_parallel_rng():
     with _TRACKER.fork("local_seed"):
         yield
+
+
+def checkpoint_rng_context():
+    """`context_fn` for torch.utils.checkpoint(use_reentrant=False):
+    checkpoint preserves only the GLOBAL torch/cuda RNG, so dropout (and
+    the flash-dropout seed draw) inside a `model_parallel_rng()` fork
+    would consume a FRESH tracker stream on the recompute re-forward —
+    different masks forward vs backward, silently wrong gradients. The
+    forward context snapshots the tracker; the recompute context rewinds
+    to that snapshot for the replay and then puts the advanced states
+    back, so the net stream position matches a no-recompute run."""
+    snapshot: Dict[str, tuple] = {}
+
+    def _clone(states):
+        return {k: (c.clone(), None if g is None else g.clone())
+                for k, (c, g) in states.items()}
+
+    class _Forward(contextlib.AbstractContextManager):
+        def __enter__(self):
+            snapshot.clear()
+            snapshot.update(_clone(_TRACKER._states))
+            return self
+
+        def __exit__(self, *exc):
+            return False
+
+    class _Recompute(contextlib.AbstractContextManager):
+        def __enter__(self):
+            self._advanced = _TRACKER._states
+            _TRACKER._states = _clone(snapshot)
+            return self
+
+        def __exit__(self, *exc):
+            _TRACKER._states = self._advanced
+            return False
+
+    return _Forward(), _Recompute()

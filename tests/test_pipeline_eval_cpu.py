@@ -84,3 +84,65 @@ def test_pipeline_engine_evaluate():
     for p in procs:
         p.join(240)
         assert p.exitcode == 0, f"worker failed with {p.exitcode}"
+
+
+def _recompute_worker(rank, world, port):
+    import sys
+    sys.path.insert(0, REPO)
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank))
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from paddlefleetx_amd.models import build_module
+    from paddlefleetx_amd.parallel.env import set_hcg, set_seed
+    from paddlefleetx_amd.parallel.topology import HybridTopology
+    set_hcg(HybridTopology(pp=2))
+
+    def build(rec):
+        set_seed(1234)
+        cfg = {
+            "Global": {"global_batch_size": 4},
+            "Engine": {"mix_precision": {"enable": False},
+                       "accumulate_steps": 2},
+            "Model": {"name": "GPTModule", "vocab_size": 128,
+                      "hidden_size": 32, "num_layers": 4,
+                      "num_attention_heads": 2,
+                      "max_position_embeddings": 16,
+                      "hidden_dropout_prob": 0.1,
+                      "attention_probs_dropout_prob": 0.1,
+                      "fused_attn": False, "use_recompute": rec,
+                      "recompute_granularity": "full"},
+            "Distributed": {"pp_degree": 2},
+        }
+        return build_module(cfg)
+
+    torch.manual_seed(7)
+    batch = (torch.randint(0, 128, (4, 16)),
+             torch.arange(16).repeat(4, 1),
+             torch.randint(0, 128, (4, 16)), torch.ones(4, 16))
+    losses = {}
+    for rec in (False, True):
+        m = build(rec)
+        torch.manual_seed(21)  # dropout stream identical across variants
+        from paddlefleetx_amd.parallel.env import set_seed as _ss
+        _ss(1234)
+        losses[rec] = float(m.model.forward_backward_pipeline(
+            batch, m.loss_fn, 2))
+    # pipe-layer "full" recompute replays dropout exactly -> same loss
+    assert abs(losses[False] - losses[True]) < 1e-6, losses
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_pipeline_full_recompute_with_dropout():
+    from port_util import free_port
+    port = free_port()
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_recompute_worker, args=(r, 2, port))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(240)
+        assert p.exitcode == 0, f"worker failed with {p.exitcode}"

@@ -114,3 +114,37 @@ def test_recompute_granularities_equivalent():
         grads[gran] = m.gpt.layers[0].attn.qkv.weight.grad.clone()
     for gran in ("full", "full_attn", "core_attn"):
         assert torch.allclose(grads[None], grads[gran], atol=1e-6), gran
+
+
+def test_recompute_replays_dropout_streams_exactly():
+    """Attention dropout draws from the mp-RNG tracker, which torch's
+    checkpoint does NOT preserve by itself — the checkpoint_rng_context
+    snapshot/rewind (parallel/rng.py) makes the re-forward replay the
+    same masks. Grads must be EXACT vs no recompute with BOTH dropouts
+    active, for every granularity."""
+    import torch
+    from paddlefleetx_amd.models.gpt.model import (GPTForPretraining,
+                                                   GPTModel,
+                                                   GPTPretrainingCriterion)
+    from paddlefleetx_amd.parallel.env import set_seed
+    grads = {}
+    for gran in (None, "full", "full_attn", "core_attn"):
+        set_seed(1234)
+        torch.manual_seed(3)
+        m = GPTForPretraining(GPTModel(
+            vocab_size=128, hidden_size=32, num_layers=2,
+            num_attention_heads=4, max_position_embeddings=32,
+            fused_attn=False, hidden_dropout_prob=0.1,
+            attention_probs_dropout_prob=0.1,
+            use_recompute=gran is not None,
+            recompute_granularity=gran or "full"))
+        m.train()
+        torch.manual_seed(7)
+        tokens = torch.randint(0, 128, (2, 16))
+        labels = torch.randint(0, 128, (2, 16))
+        loss = GPTPretrainingCriterion()(m(tokens), labels,
+                                         torch.ones(2, 16))
+        loss.backward()
+        grads[gran] = m.gpt.layers[0].attn.qkv.weight.grad.clone()
+    for gran in ("full", "full_attn", "core_attn"):
+        assert torch.equal(grads[None], grads[gran]), gran
