@@ -542,6 +542,8 @@ class EagerEngine(BasicEngine):
                 "loss_scale": self.loss_scale,
                 "good_steps": getattr(self, "_good_steps", 0),
                 "cpu_rng_state": torch.get_rng_state()}
+        from paddlefleetx_amd.parallel.rng import get_rng_tracker
+        meta["rng_tracker"] = get_rng_tracker().state_dict()
         if torch.cuda.is_available():
             meta["cuda_rng_state"] = torch.cuda.get_rng_state()
         torch.save(meta, os.path.join(out, "meta_state.pdopt"))
@@ -594,6 +596,11 @@ class EagerEngine(BasicEngine):
                 self._good_steps = int(meta.get("good_steps", 0))
             if "cpu_rng_state" in meta:
                 torch.set_rng_state(meta["cpu_rng_state"])
+            if "rng_tracker" in meta:
+                # resume the mp-dropout streams where they left off
+                # (the reference restores only the global CUDA state)
+                from paddlefleetx_amd.parallel.rng import get_rng_tracker
+                get_rng_tracker().load_state_dict(meta["rng_tracker"])
             if "cuda_rng_state" in meta and torch.cuda.is_available():
                 torch.cuda.set_rng_state(meta["cuda_rng_state"])
         logger.info(f"loaded checkpoint from {path}")

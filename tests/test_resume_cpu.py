@@ -20,8 +20,11 @@ def _cfg(extra_engine=None):
         "Model": {"name": "GPTModule", "vocab_size": 128, "hidden_size": 32,
                   "num_layers": 2, "num_attention_heads": 2,
                   "max_position_embeddings": 16,
-                  "hidden_dropout_prob": 0.0,
-                  "attention_probs_dropout_prob": 0.0, "fused_attn": False},
+                  # dropout ON: resume must also restore the global AND
+                  # mp-tracker RNG streams to match the continuous run
+                  "hidden_dropout_prob": 0.1,
+                  "attention_probs_dropout_prob": 0.1,
+                  "fused_attn": False},
         "Optimizer": {"name": "FusedAdamW", "weight_decay": 0.01,
                       "lr": {"name": "ConstantLR", "learning_rate": 1e-3}},
         "Distributed": {},
