@@ -182,3 +182,34 @@ def test_engine_hipgraph_matches_eager():
     for a, b in zip(p0, p1):
         assert torch.allclose(a.float(), b.float(), atol=1e-2), \
             (a - b).float().abs().max()
+
+
+def test_recompute_dropout_exact_on_flash_path():
+    """GPU analogue of the checkpoint_rng_context fix: the flash
+    attention dropout SEED is drawn from the mp-tracker stream, so the
+    recompute re-forward must replay the same seed (and the same
+    in-kernel Philox masks). Grads exact vs no recompute."""
+    from paddlefleetx_amd.models.gpt.model import (GPTForPretraining,
+                                                   GPTModel,
+                                                   GPTPretrainingCriterion)
+    from paddlefleetx_amd.parallel.env import set_seed
+    grads = {}
+    for rec in (False, True):
+        set_seed(1234)
+        torch.manual_seed(3)
+        m = GPTForPretraining(GPTModel(
+            vocab_size=512, hidden_size=256, num_layers=2,
+            num_attention_heads=2, max_position_embeddings=128,
+            fused_attn=True, hidden_dropout_prob=0.1,
+            attention_probs_dropout_prob=0.1,
+            use_recompute=rec, recompute_granularity="full",
+            dtype=torch.bfloat16)).cuda()
+        m.train()
+        torch.manual_seed(7)
+        tokens = torch.randint(0, 512, (2, 128), device="cuda")
+        labels = torch.randint(0, 512, (2, 128), device="cuda")
+        loss = GPTPretrainingCriterion()(m(tokens), labels,
+                                         torch.ones(2, 128, device="cuda"))
+        loss.backward()
+        grads[rec] = m.gpt.layers[0].attn.qkv.weight.grad.float().clone()
+    assert torch.equal(grads[False], grads[True])
